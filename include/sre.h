@@ -1,0 +1,131 @@
+/* sre.h — C-ABI drop-in boundary of the MI355X-native state-root engine.
+ *
+ * This is the extern-C surface a reth maintainer would bind over FFI to
+ * replace the Merkle-stage state-root computation. Each entry point cites the
+ * reference interface it replaces (paths relative to /root/reference):
+ *
+ *   - sre_upload_accounts / sre_upload_storage + sre_root mirror
+ *     `DatabaseStateRoot::overlay_root(tx, HashedPostStateSorted)`
+ *     (crates/trie/db/src/state.rs:119-138) and the object-safe
+ *     `StateRootProvider::state_root(HashedPostState) -> B256`
+ *     (crates/storage/storage-api/src/trie.rs:13-41): sorted hashed entries
+ *     in, 32-byte root out.
+ *   - sre_storage_roots mirrors `StorageRootProvider::storage_root`
+ *     (crates/storage/storage-api/src/trie.rs:45-58) /
+ *     `StorageRoot::calculate_with_cursors` (crates/trie/trie/src/trie.rs:750-876)
+ *     batched over every account.
+ *   - sre_subtree_roots / sre_finish_top are the multi-GPU decomposition of
+ *     `StateRoot::calculate`'s account-trie walk (crates/trie/trie/src/trie.rs:171-374):
+ *     per-top-nibble subtrie digests exchanged over RCCL, root finished from
+ *     the 16 child references.
+ *
+ * Input contract (matches the reference's cursor semantics):
+ *   - account entries sorted ascending by `key` (the keccak256 of the
+ *     address — reth's HashedAccounts table order,
+ *     crates/trie/common/src/hashed_state.rs:331 `into_sorted`), no
+ *     duplicates, no "empty" accounts (reth never stores them:
+ *     crates/trie/trie/src/hashed_cursor/post_state.rs:70-82).
+ *   - storage entries sorted ascending by (acct_key, slot_key), no
+ *     duplicates, values nonzero (zero-valued slots are absent, not stored:
+ *     crates/trie/trie/src/trie.rs:819-825), and every acct_key present in
+ *     the accounts upload.
+ *
+ * Thread model: all calls on one host thread per ctx; the engine is
+ * internally multi-stream. The engine COPIES input buffers at upload; the
+ * caller owns them afterwards. Errors: nonzero int return +
+ * sre_last_error(). This library is GPU-only by design: sre_create fails
+ * loudly when no HIP device is present — there is no CPU fallback.
+ */
+#ifndef SRE_H
+#define SRE_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* Hashed account entry — reth `Account` (external reth-primitives-traits) +
+ * hashed address key; bytecode_hash None is represented as KECCAK_EMPTY, the
+ * same normalization `into_trie_account` applies
+ * (crates/stages/stages/src/stages/merkle.rs:297). balance is big-endian
+ * 32-byte U256. 104 bytes, no padding. */
+typedef struct {
+    uint8_t  key[32];       /* keccak256(address) */
+    uint64_t nonce;
+    uint8_t  balance[32];   /* big-endian U256 */
+    uint8_t  code_hash[32]; /* KECCAK_EMPTY when no code */
+} sre_account_entry;
+
+/* Hashed storage entry — one row of reth's HashedStorages dup-sorted table
+ * (crates/storage/db-api/src/tables/mod.rs:481-507). value big-endian U256,
+ * nonzero. 96 bytes, no padding. */
+typedef struct {
+    uint8_t acct_key[32];   /* keccak256(address) */
+    uint8_t slot_key[32];   /* keccak256(slot) */
+    uint8_t value[32];      /* big-endian U256, != 0 */
+} sre_storage_entry;
+
+typedef struct sre_ctx sre_ctx;
+
+/* Per-call timing/throughput stats (HIP-event measured, for bench reporting;
+ * mirrors the spirit of crates/trie/trie/src/metrics.rs:7-60). Times in ms. */
+typedef struct {
+    double   total_ms;          /* whole sre_root device pipeline */
+    double   leaf_hash_ms;      /* storage+account leaf RLP+keccak kernels */
+    uint64_t leaf_count;        /* leaves hashed (storage + account) */
+    uint64_t leaf_blocks;       /* keccak-f[1600] invocations in leaf kernels */
+    double   branch_hash_ms;    /* branch/extension assemble+keccak kernels */
+    uint64_t branch_count;      /* branch nodes built */
+    uint64_t branch_blocks;     /* keccak-f invocations in branch kernels */
+    double   sort_ms;           /* lcp/bucket/merge machinery */
+    uint64_t levels;            /* depth levels processed */
+} sre_stats;
+
+/* Create a context bound to HIP device `device`. Returns NULL on failure
+ * (use sre_last_error(NULL) for the reason). */
+sre_ctx *sre_create(int device);
+void     sre_destroy(sre_ctx *ctx);
+
+/* Upload sorted entries to device memory (H2D copy happens here, outside any
+ * timed region). May be called again to replace the state. */
+int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entries, uint64_t n);
+int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries, uint64_t n);
+
+/* Compute the state root over the uploaded entries.
+ * Equivalent surface: StateRoot::root() (crates/trie/trie/src/trie.rs:154). */
+int sre_root(sre_ctx *ctx, uint8_t out_root[32]);
+
+/* Per-account storage roots (n must equal the uploaded account count; out is
+ * n*32 bytes, account order). Runs the storage pass only. */
+int sre_storage_roots(sre_ctx *ctx, uint8_t *out, uint64_t n);
+
+/* Shard mode: compute, for each account top nibble owned by this rank's
+ * upload, (a) the child reference the root branch would embed (len in
+ * out_child_lens[i]; 0 = nibble absent) and (b) the 32-byte root hash the
+ * subtree would have if it were the entire trie, plus leaf counts. The
+ * caller all-gathers these across ranks and calls sre_finish_top. */
+int sre_subtree_roots(sre_ctx *ctx,
+                      uint8_t  out_child_refs[16][33],
+                      uint8_t  out_child_lens[16],
+                      uint8_t  out_root_hash[16][32],
+                      uint64_t out_counts[16]);
+
+/* Finish the account-trie top from gathered per-nibble results.
+ * counts[i] = total accounts under nibble i across all ranks. */
+int sre_finish_top(sre_ctx *ctx,
+                   const uint8_t  child_refs[16][33],
+                   const uint8_t  child_lens[16],
+                   const uint8_t  root_hash[16][32],
+                   const uint64_t counts[16],
+                   uint8_t out_root[32]);
+
+int sre_get_stats(sre_ctx *ctx, sre_stats *out);
+
+/* Last error string for ctx (or the global creation error when ctx==NULL). */
+const char *sre_last_error(const sre_ctx *ctx);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* SRE_H */
