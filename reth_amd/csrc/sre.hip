@@ -1,0 +1,1782 @@
+// sre.hip — MI355X-native state-root engine (gfx950 / CDNA4).
+//
+// From-scratch GPU implementation of the Merkle Patricia Trie commitment
+// path behind include/sre.h. NOT a port: the reference
+// (/root/reference/crates/trie, Rust, CPU-only) streams leaves through a
+// sequential HashBuilder (crates/trie/trie/src/trie.rs:171-374); this engine
+// instead builds each trie bottom-up from the sorted leaf stream with
+// data-parallel LCP levels:
+//
+//   1. lcp[i] between adjacent keys (segment-aware; -1 sentinels at segment
+//      boundaries = per-account storage tries, matching the per-account
+//      independence of StorageRoot::calculate_with_cursors, trie.rs:750-876).
+//   2. every leaf's parent-branch depth D = max(lcp[i], lcp[i+1]); leaf RLP
+//      (hex-prefix short key from D+1 + RLP value) assembled directly in a
+//      per-lane LDS slot, hashed by a per-lane Keccak-f[1600] sponge with the
+//      5x5 state in VGPRs (all state indices compile-time constant).
+//   3. levels d = maxD..0: nodes with parent depth == d grouped into branch
+//      nodes (adjacent runs with junction lcp == d), 17-item branch RLP
+//      assembled in LDS + hashed; extension wrap when the depth jumps; the
+//      new node re-enters the level machinery at its own parent depth.
+//   4. nodes reaching parent depth -1 are (per-segment) roots: root hash is
+//      always keccak(RLP) (proof_v2/mod.rs:1628 compute_root_hash rule).
+//
+// Node semantics restated from the reference:
+//   leaf  = RLP([HP(short,1), RLP_value])        proof_v2/node.rs:40-66
+//   ext   = RLP([HP(shared,0), child_ref])
+//   branch= RLP([c0..c15, ""])  (value slot always empty in state tries)
+//   ref   = RLP if len<32 else 0xa0||keccak      RlpNode::from_rlp rule
+//   storage value = minimal big-endian RLP of U256 (trie.rs:819-825)
+//   account value = RLP([nonce,balance,storage_root,code_hash]) <= 110 B
+//                   (trie.rs:472-476, root.rs:9-31)
+//   empty storage -> EMPTY_ROOT_HASH (trie.rs:771-781)
+//
+// All hashing and node construction runs on-device; there is no CPU
+// fallback. Parity: bit-exact vs oracle/ (tests/test_gpu_parity.py).
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <cstdio>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "../../include/sre.h"
+
+#define BLOCK 256u
+
+enum {
+    E_OK = 0,
+    E_UNSORTED_ACCT = 1,
+    E_UNSORTED_STORAGE = 2,
+    E_ORPHAN_STORAGE = 3,
+    E_ZERO_VALUE = 4,
+    E_INTERNAL = 5,
+};
+
+// ---------------------------------------------------------------------------
+// device keccak-256 (one sponge per lane, state in registers)
+// ---------------------------------------------------------------------------
+
+__constant__ uint64_t KRC[24] = {
+    0x0000000000000001ULL, 0x0000000000008082ULL, 0x800000000000808aULL,
+    0x8000000080008000ULL, 0x000000000000808bULL, 0x0000000080000001ULL,
+    0x8000000080008081ULL, 0x8000000000008009ULL, 0x000000000000008aULL,
+    0x0000000000000088ULL, 0x0000000080008009ULL, 0x000000008000000aULL,
+    0x000000008000808bULL, 0x800000000000008bULL, 0x8000000000008089ULL,
+    0x8000000000008003ULL, 0x8000000000008002ULL, 0x8000000000000080ULL,
+    0x000000000000800aULL, 0x800000008000000aULL, 0x8000000080008081ULL,
+    0x8000000000008080ULL, 0x0000000080000001ULL, 0x8000000080008008ULL,
+};
+
+__constant__ uint8_t D_EMPTY_ROOT[32] = {
+    // keccak(rlp("")) — trie.rs:771-781 empty-storage shortcut
+    0x56, 0xe8, 0x1f, 0x17, 0x1b, 0xcc, 0x55, 0xa6, 0xff, 0x83, 0x45, 0xe6,
+    0x92, 0xc0, 0xf8, 0x6e, 0x5b, 0x48, 0xe0, 0x1b, 0x99, 0x6c, 0xad, 0xc0,
+    0x01, 0x62, 0x2f, 0xb5, 0xe3, 0x63, 0xb4, 0x21,
+};
+
+__device__ __forceinline__ uint64_t rotl64(uint64_t x, int n)
+{
+    return (x << n) | (x >> (64 - n));
+}
+
+// Keccak-f[1600]. Inner loops unrolled so every s[]/b[] index is a
+// compile-time constant (register-resident; dynamic indexing would spill to
+// scratch — cdna_hip_programming.md §5.4 rule 20).
+__device__ void keccak_f(uint64_t s[25])
+{
+    constexpr int ROT[25] = {
+        0, 1, 62, 28, 27, 36, 44, 6, 55, 20, 3, 10, 43, 25, 39,
+        41, 45, 15, 21, 8, 18, 2, 61, 56, 14,
+    };
+    for (int r = 0; r < 24; ++r) {
+        uint64_t c[5], d[5], b[25];
+#pragma unroll
+        for (int x = 0; x < 5; ++x)
+            c[x] = s[x] ^ s[x + 5] ^ s[x + 10] ^ s[x + 15] ^ s[x + 20];
+#pragma unroll
+        for (int x = 0; x < 5; ++x)
+            d[x] = c[(x + 4) % 5] ^ rotl64(c[(x + 1) % 5], 1);
+#pragma unroll
+        for (int i = 0; i < 25; ++i)
+            s[i] ^= d[i % 5];
+#pragma unroll
+        for (int x = 0; x < 5; ++x) {
+#pragma unroll
+            for (int y = 0; y < 5; ++y) {
+                const int src = x + 5 * y;
+                const int dst = y + 5 * ((2 * x + 3 * y) % 5);
+                b[dst] = ROT[src] ? rotl64(s[src], ROT[src]) : s[src];
+            }
+        }
+#pragma unroll
+        for (int y = 0; y < 5; ++y) {
+#pragma unroll
+            for (int x = 0; x < 5; ++x)
+                s[x + 5 * y] =
+                    b[x + 5 * y] ^ ((~b[(x + 1) % 5 + 5 * y]) & b[(x + 2) % 5 + 5 * y]);
+        }
+        s[0] ^= KRC[r];
+    }
+}
+
+// Hash a message in an LDS slot already zero-padded to nblocks*136 bytes
+// with the 0x01 / 0x80 domain bytes applied.
+__device__ __forceinline__ void keccak_lds(const uint64_t *slot, int nblocks,
+                                           uint64_t out[4])
+{
+    uint64_t s[25];
+#pragma unroll
+    for (int i = 0; i < 25; ++i)
+        s[i] = 0;
+    for (int blk = 0; blk < nblocks; ++blk) {
+#pragma unroll
+        for (int i = 0; i < 17; ++i)
+            s[i] ^= slot[blk * 17 + i];
+        keccak_f(s);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+        out[i] = s[i];
+}
+
+// keccak pad10*1 (domain 0x01) over an LDS slot holding len message bytes
+// (slot zeroed beyond len). Returns block count.
+__device__ __forceinline__ int keccak_pad(uint8_t *slot, int len)
+{
+    int nblocks = len / 136 + 1;
+    slot[len] = 0x01;
+    slot[nblocks * 136 - 1] |= 0x80;
+    return nblocks;
+}
+
+// ---------------------------------------------------------------------------
+// device RLP / hex-prefix / key helpers (assembly goes straight into LDS;
+// no dynamically-indexed private arrays -> no scratch spills)
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ uint8_t nib_of(const uint8_t *key, int i)
+{
+    return (key[i >> 1] >> ((i & 1) ? 0 : 4)) & 0x0f;
+}
+
+__device__ __forceinline__ int rlp_list_hdr_len(int payload)
+{
+    return payload < 56 ? 1 : (payload <= 255 ? 2 : 3);
+}
+
+__device__ __forceinline__ int rlp_list_hdr_write(uint8_t *p, int payload)
+{
+    if (payload < 56) {
+        p[0] = (uint8_t)(0xc0 + payload);
+        return 1;
+    }
+    if (payload <= 255) {
+        p[0] = 0xf8;
+        p[1] = (uint8_t)payload;
+        return 2;
+    }
+    p[0] = 0xf9;
+    p[1] = (uint8_t)(payload >> 8);
+    p[2] = (uint8_t)payload;
+    return 3;
+}
+
+// hex-prefix item length for key nibbles [from,to): HP bytes = 1+floor(n/2);
+// the RLP string wrapper adds 1 prefix byte unless it is a single byte <0x80
+// (HP first byte is <= 0x3f, so the 1-byte case is always unprefixed).
+__device__ __forceinline__ int hp_bytes(int from, int to)
+{
+    return 1 + ((to - from) >> 1);
+}
+
+__device__ __forceinline__ int hp_item_len(int from, int to)
+{
+    int hb = hp_bytes(from, to);
+    return hb == 1 ? 1 : 1 + hb;
+}
+
+// write the RLP string item of HP(key[from..to), leaf) at p; returns bytes.
+__device__ __forceinline__ int hp_item_write(uint8_t *p, const uint8_t *key,
+                                             int from, int to, int leaf)
+{
+    int n = to - from;
+    int odd = n & 1;
+    int hb = hp_bytes(from, to);
+    int w = 0;
+    if (hb > 1)
+        p[w++] = (uint8_t)(0x80 + hb);
+    uint8_t first = (uint8_t)((leaf ? 0x20 : 0x00) | (odd ? 0x10 : 0x00));
+    int i = from;
+    if (odd) {
+        first |= nib_of(key, i);
+        i++;
+    }
+    p[w++] = first;
+    for (; i < to; i += 2)
+        p[w++] = (uint8_t)((nib_of(key, i) << 4) | nib_of(key, i + 1));
+    return w;
+}
+
+__device__ __forceinline__ uint64_t be64_at(const uint8_t *p)
+{
+    uint64_t v;
+    memcpy(&v, p, 8);
+    return __builtin_bswap64(v);
+}
+
+// lcp in nibbles of two 32-byte keys; sets *gt if a > b.
+__device__ __forceinline__ int key_lcp(const uint8_t *a, const uint8_t *b, bool *gt)
+{
+    *gt = false;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+        uint64_t wa = be64_at(a + 8 * k);
+        uint64_t wb = be64_at(b + 8 * k);
+        if (wa != wb) {
+            *gt = wa > wb;
+            return k * 16 + (__clzll(wa ^ wb) >> 2);
+        }
+    }
+    return 64;
+}
+
+// minimal big-endian length of a 32-byte big-endian value (0 for zero)
+__device__ __forceinline__ int min_be_len(const uint8_t *v)
+{
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+        uint64_t w = be64_at(v + 8 * k);
+        if (w)
+            return 32 - 8 * k - (__clzll(w) >> 3);
+    }
+    return 0;
+}
+
+// active trie node record, 48 bytes
+struct node_rec {
+    uint32_t s, e;    // covered leaf interval [s, e)
+    uint32_t seg;     // segment id (storage: account segment; account: 0/top nibble)
+    int8_t depth;     // parent branch depth this node waits for (-1 = root)
+    uint8_t ref_len;  // 1..33 (0 marks a consumed root record)
+    uint8_t ref[33];
+    uint8_t pad_;
+};
+static_assert(sizeof(node_rec) == 48, "node_rec must be 48 bytes");
+
+__device__ __forceinline__ void copy_rec(node_rec *dst, const node_rec *src)
+{
+    const uint32_t *s32 = (const uint32_t *)src;
+    uint32_t *d32 = (uint32_t *)dst;
+#pragma unroll
+    for (int i = 0; i < 12; ++i)
+        d32[i] = s32[i];
+}
+
+// ref from an LDS-resident rlp + its precomputed hash
+__device__ __forceinline__ void make_ref(const uint8_t *rlp, int len,
+                                         const uint64_t hash[4], uint8_t *ref,
+                                         uint8_t *ref_len)
+{
+    if (len < 32) {
+        *ref_len = (uint8_t)len;
+        for (int i = 0; i < len; ++i)
+            ref[i] = rlp[i];
+    } else {
+        *ref_len = 33;
+        ref[0] = 0xa0;
+        memcpy(ref + 1, hash, 32);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// batch keccak (input generation / hashing-stage offload)
+// ---------------------------------------------------------------------------
+
+#define SLOT_KB 136
+__global__ void __launch_bounds__(BLOCK) k_keccak_batch(
+    const uint8_t *__restrict__ in, uint64_t stride, uint32_t len, uint64_t n,
+    uint8_t *__restrict__ out)
+{
+    __shared__ uint8_t lds[BLOCK * SLOT_KB];
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n)
+        return;
+    uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_KB;
+    uint64_t *slot64 = (uint64_t *)slot;
+#pragma unroll
+    for (int k = 0; k < SLOT_KB / 8; ++k)
+        slot64[k] = 0;
+    const uint8_t *msg = in + i * stride;
+    for (uint32_t b = 0; b < len; ++b)
+        slot[b] = msg[b];
+    int nblocks = keccak_pad(slot, (int)len);
+    uint64_t hash[4];
+    keccak_lds(slot64, nblocks, hash);
+    uint64_t *o = (uint64_t *)(out + 32 * i);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+        o[k] = hash[k];
+}
+
+// ---------------------------------------------------------------------------
+// segmentation + lcp kernels
+// ---------------------------------------------------------------------------
+
+__global__ void k_seg_flags(const sre_storage_entry *__restrict__ st, uint64_t ns,
+                            uint32_t *__restrict__ flags, uint32_t *__restrict__ err)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= ns)
+        return;
+    if (i == 0) {
+        flags[0] = 1;
+        return;
+    }
+    bool gt;
+    int l = key_lcp(st[i - 1].acct_key, st[i].acct_key, &gt);
+    if (gt)
+        atomicOr(err, 1u << E_UNSORTED_STORAGE);
+    flags[i] = (l < 64) ? 1u : 0u;
+}
+
+__global__ void k_seg_starts(const uint32_t *__restrict__ flags,
+                             const uint32_t *__restrict__ seg_id, uint64_t ns,
+                             uint32_t *__restrict__ seg_start)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= ns)
+        return;
+    if (flags[i])
+        seg_start[seg_id[i]] = (uint32_t)i;
+}
+
+__global__ void k_seg_acct(const sre_storage_entry *__restrict__ st,
+                           const uint32_t *__restrict__ seg_start, uint32_t n_seg,
+                           const sre_account_entry *__restrict__ acct, uint64_t na,
+                           uint32_t *__restrict__ seg_acct, uint32_t *__restrict__ err)
+{
+    uint32_t s = blockIdx.x * blockDim.x + threadIdx.x;
+    if (s >= n_seg)
+        return;
+    const uint8_t *key = st[seg_start[s]].acct_key;
+    uint64_t lo = 0, hi = na;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2;
+        bool gt;
+        int l = key_lcp(acct[mid].key, key, &gt);
+        if (l == 64) {
+            seg_acct[s] = (uint32_t)mid;
+            return;
+        }
+        if (gt)
+            hi = mid;
+        else
+            lo = mid + 1;
+    }
+    atomicOr(err, 1u << E_ORPHAN_STORAGE);
+}
+
+__global__ void k_lcp_storage(const sre_storage_entry *__restrict__ st, uint64_t ns,
+                              const uint32_t *__restrict__ seg_id,
+                              int8_t *__restrict__ lcp, uint32_t *__restrict__ err)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i > ns)
+        return;
+    if (i == 0 || i == ns || seg_id[i] != seg_id[i - 1]) {
+        lcp[i] = -1;
+        return;
+    }
+    bool gt;
+    int l = key_lcp(st[i - 1].slot_key, st[i].slot_key, &gt);
+    if (gt || l == 64)
+        atomicOr(err, 1u << E_UNSORTED_STORAGE);
+    lcp[i] = (int8_t)l;
+}
+
+__global__ void k_lcp_account(const sre_account_entry *__restrict__ acct, uint64_t na,
+                              int subtree, int8_t *__restrict__ lcp,
+                              uint32_t *__restrict__ err)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i > na)
+        return;
+    if (i == 0 || i == na) {
+        lcp[i] = -1;
+        return;
+    }
+    bool gt;
+    int l = key_lcp(acct[i - 1].key, acct[i].key, &gt);
+    if (gt || l == 64)
+        atomicOr(err, 1u << E_UNSORTED_ACCT);
+    if (subtree && l == 0)
+        l = -1;
+    lcp[i] = (int8_t)l;
+}
+
+// ---------------------------------------------------------------------------
+// leaf kernels
+// ---------------------------------------------------------------------------
+
+// Storage leaf: rlp <= 2 + 34 + 34 = 70 B -> always one keccak block.
+#define SLOT_STO 136
+__global__ void __launch_bounds__(BLOCK) k_leaf_storage(
+    const sre_storage_entry *__restrict__ st, uint64_t ns,
+    const int8_t *__restrict__ lcp, const uint32_t *__restrict__ seg_id,
+    node_rec *__restrict__ recs, uint8_t *__restrict__ depths,
+    uint32_t *__restrict__ hist, uint8_t *__restrict__ seg_roots,
+    uint32_t *__restrict__ err)
+{
+    __shared__ uint8_t lds[BLOCK * SLOT_STO + 66 * 4];
+    uint32_t *hist_l = (uint32_t *)(lds + BLOCK * SLOT_STO);
+    if (threadIdx.x < 66)
+        hist_l[threadIdx.x] = 0;
+    __syncthreads();
+
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < ns) {
+        uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_STO;
+        uint64_t *slot64 = (uint64_t *)slot;
+#pragma unroll
+        for (int k = 0; k < SLOT_STO / 8; ++k)
+            slot64[k] = 0;
+
+        int8_t l0 = lcp[i], l1 = lcp[i + 1];
+        int D = l0 > l1 ? l0 : l1;
+        int from = D + 1;
+
+        const uint8_t *key = st[i].slot_key;
+        const uint8_t *val = st[i].value;
+        int vlen = min_be_len(val);
+        if (vlen == 0) {
+            atomicOr(err, 1u << E_ZERO_VALUE);
+        } else {
+            // storage value = encode_fixed_size(U256) (trie.rs:824); the leaf
+            // stores that RLP as a string item -> wrap once more.
+            int vrlp_len = (vlen == 1 && val[31] < 0x80) ? 1 : 1 + vlen;
+            int vitem_len = vrlp_len == 1 ? 1 : 1 + vrlp_len;
+            int payload = hp_item_len(from, 64) + vitem_len;
+            int h = rlp_list_hdr_write(slot, payload);
+            int p = h + hp_item_write(slot + h, key, from, 64, 1);
+            if (vrlp_len > 1)
+                slot[p++] = (uint8_t)(0x80 + vrlp_len);
+            if (vlen > 1 || val[31] >= 0x80)
+                slot[p++] = (uint8_t)(0x80 + vlen);
+            for (int k = 0; k < vlen; ++k)
+                slot[p++] = val[32 - vlen + k];
+            int len = h + payload;
+            int nblocks = keccak_pad(slot, len);
+            uint64_t hash[4];
+            keccak_lds(slot64, nblocks, hash);
+
+            node_rec *r = &recs[i];
+            r->s = (uint32_t)i;
+            r->e = (uint32_t)(i + 1);
+            uint32_t seg = seg_id[i];
+            r->seg = seg;
+            r->depth = (int8_t)D;
+            uint8_t rl;
+            make_ref(slot, len, hash, r->ref, &rl);
+            r->ref_len = rl;
+            r->pad_ = 0;
+            depths[i] = (uint8_t)(D + 1);
+            atomicAdd(&hist_l[D + 1], 1u);
+            if (D == -1) // single-slot storage trie: root = keccak(leaf RLP)
+                memcpy(seg_roots + 32ull * seg, hash, 32);
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x < 66 && hist_l[threadIdx.x])
+        atomicAdd(&hist[threadIdx.x], hist_l[threadIdx.x]);
+}
+
+// Account leaf. value = RLP([nonce,balance,storage_root,code_hash])
+// (trie.rs:472-476); leaf rlp <= 2 + 34 + 2 + 113 = 151 B -> <= 2 blocks.
+#define SLOT_ACC 280
+__global__ void __launch_bounds__(BLOCK) k_leaf_account(
+    const sre_account_entry *__restrict__ acct, uint64_t na,
+    const uint8_t *__restrict__ storage_roots, const int8_t *__restrict__ lcp,
+    int subtree, node_rec *__restrict__ recs, uint8_t *__restrict__ depths,
+    uint32_t *__restrict__ hist, uint8_t *__restrict__ roots,
+    uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens)
+{
+    __shared__ uint8_t lds[BLOCK * SLOT_ACC + 66 * 4];
+    uint32_t *hist_l = (uint32_t *)(lds + BLOCK * SLOT_ACC);
+    if (threadIdx.x < 66)
+        hist_l[threadIdx.x] = 0;
+    __syncthreads();
+
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < na) {
+        uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_ACC;
+        uint64_t *slot64 = (uint64_t *)slot;
+
+        int8_t l0 = lcp[i], l1 = lcp[i + 1];
+        int D = l0 > l1 ? l0 : l1;
+        const uint8_t *key = acct[i].key;
+
+        // account-value field lengths
+        uint64_t nonce = acct[i].nonce;
+        int nlen = nonce ? (8 - (__clzll(nonce) >> 3)) : 0;
+        int nonce_rlp = (nlen == 0 || (nlen == 1 && nonce < 0x80)) ? 1 : 1 + nlen;
+        const uint8_t *bal = acct[i].balance;
+        int blen = min_be_len(bal);
+        int bal_rlp = (blen == 0 || (blen == 1 && bal[31] < 0x80)) ? 1 : 1 + blen;
+        int vw = nonce_rlp + bal_rlp + 33 + 33;      // account list payload
+        int vh = rlp_list_hdr_len(vw);               // its list header (1: vw<56? vw>=68 -> 2... see below)
+        int vtotal = vh + vw;                        // bytes of account RLP
+        int vs = vtotal < 56 ? 1 : 2;                // string wrapper header
+
+        // assemble + hash leaf with short key from `fr`
+        auto emit = [&](int fr, uint64_t hash[4]) -> int {
+#pragma unroll
+            for (int k = 0; k < SLOT_ACC / 8; ++k)
+                slot64[k] = 0;
+            int payload = hp_item_len(fr, 64) + vs + vtotal;
+            int h = rlp_list_hdr_write(slot, payload);
+            int p = h + hp_item_write(slot + h, key, fr, 64, 1);
+            if (vs == 1) {
+                slot[p++] = (uint8_t)(0x80 + vtotal);
+            } else {
+                slot[p++] = 0xb8;
+                slot[p++] = (uint8_t)vtotal;
+            }
+            p += rlp_list_hdr_write(slot + p, vw);
+            if (nlen == 0) {
+                slot[p++] = 0x80;
+            } else {
+                if (nonce_rlp > 1)
+                    slot[p++] = (uint8_t)(0x80 + nlen);
+                for (int k = nlen - 1; k >= 0; --k)
+                    slot[p++] = (uint8_t)(nonce >> (8 * k));
+            }
+            if (blen == 0) {
+                slot[p++] = 0x80;
+            } else {
+                if (bal_rlp > 1)
+                    slot[p++] = (uint8_t)(0x80 + blen);
+                for (int k = 0; k < blen; ++k)
+                    slot[p++] = bal[32 - blen + k];
+            }
+            slot[p++] = 0xa0;
+            for (int k = 0; k < 32; ++k)
+                slot[p++] = storage_roots[32 * i + k];
+            slot[p++] = 0xa0;
+            for (int k = 0; k < 32; ++k)
+                slot[p++] = acct[i].code_hash[k];
+            int len = h + payload;
+            int nblocks = keccak_pad(slot, len);
+            keccak_lds(slot64, nblocks, hash);
+            return len;
+        };
+
+        uint64_t hash[4];
+        int len = emit(D + 1, hash);
+        node_rec r;
+        r.s = (uint32_t)i;
+        r.e = (uint32_t)(i + 1);
+        r.seg = subtree ? (uint32_t)(key[0] >> 4) : 0u;
+        r.depth = (int8_t)D;
+        make_ref(slot, len, hash, r.ref, &r.ref_len);
+        r.pad_ = 0;
+        copy_rec(&recs[i], &r);
+        depths[i] = (uint8_t)(D + 1);
+        atomicAdd(&hist_l[D + 1], 1u);
+        if (D == -1) {
+            if (subtree) {
+                uint64_t chash[4];
+                int clen = emit(1, chash); // child-of-root-branch form
+                uint8_t clens;
+                make_ref(slot, clen, chash, child_refs + 33ull * r.seg, &clens);
+                child_lens[r.seg] = clens;
+                len = emit(0, hash); // standalone-root form
+            }
+            memcpy(roots + 32ull * r.seg, hash, 32);
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x < 66 && hist_l[threadIdx.x])
+        atomicAdd(&hist[threadIdx.x], hist_l[threadIdx.x]);
+}
+
+// ---------------------------------------------------------------------------
+// level machinery: selection / partition / merge / grouping
+// ---------------------------------------------------------------------------
+
+#define SEL_ITEMS 16
+
+// parallel exclusive scan of 256 per-thread counts in LDS; returns (excl,
+// total) — Hillis-Steele, 8 steps.
+__device__ __forceinline__ void block_scan(uint32_t *lds, uint32_t v,
+                                           uint32_t *excl, uint32_t *total)
+{
+    lds[threadIdx.x] = v;
+    __syncthreads();
+    for (uint32_t off = 1; off < BLOCK; off <<= 1) {
+        uint32_t x = threadIdx.x >= off ? lds[threadIdx.x - off] : 0;
+        __syncthreads();
+        lds[threadIdx.x] += x;
+        __syncthreads();
+    }
+    *excl = threadIdx.x ? lds[threadIdx.x - 1] : 0;
+    *total = lds[BLOCK - 1];
+    __syncthreads();
+}
+
+__global__ void k_sel_count(const uint8_t *__restrict__ depths, uint64_t n,
+                            uint32_t want, uint32_t *__restrict__ block_cnt)
+{
+    __shared__ uint32_t lds[BLOCK];
+    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SEL_ITEMS +
+                    (uint64_t)threadIdx.x * SEL_ITEMS;
+    uint32_t c = 0;
+    for (int k = 0; k < SEL_ITEMS; ++k) {
+        uint64_t i = base + k;
+        if (i < n && depths[i] == want)
+            c++;
+    }
+    uint32_t excl, total;
+    block_scan(lds, c, &excl, &total);
+    if (threadIdx.x == 0)
+        block_cnt[blockIdx.x] = total;
+}
+
+__global__ void k_sel_gather(const uint8_t *__restrict__ depths,
+                             const node_rec *__restrict__ recs, uint64_t n,
+                             uint32_t want, const uint32_t *__restrict__ block_off,
+                             node_rec *__restrict__ out)
+{
+    __shared__ uint32_t lds[BLOCK];
+    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SEL_ITEMS +
+                    (uint64_t)threadIdx.x * SEL_ITEMS;
+    uint32_t c = 0;
+    for (int k = 0; k < SEL_ITEMS; ++k) {
+        uint64_t i = base + k;
+        if (i < n && depths[i] == want)
+            c++;
+    }
+    uint32_t excl, total;
+    block_scan(lds, c, &excl, &total);
+    uint32_t pos = block_off[blockIdx.x] + excl;
+    for (int k = 0; k < SEL_ITEMS; ++k) {
+        uint64_t i = base + k;
+        if (i < n && depths[i] == want)
+            copy_rec(&out[pos++], &recs[i]);
+    }
+}
+
+// carry partition by record depth (sel = depth==want, rest = others)
+__global__ void k_part_count(const node_rec *__restrict__ recs, uint64_t n,
+                             int want_depth, uint32_t *__restrict__ bc_sel,
+                             uint32_t *__restrict__ bc_rest)
+{
+    __shared__ uint32_t lds[BLOCK];
+    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SEL_ITEMS +
+                    (uint64_t)threadIdx.x * SEL_ITEMS;
+    uint32_t cs = 0, cr = 0;
+    for (int k = 0; k < SEL_ITEMS; ++k) {
+        uint64_t i = base + k;
+        if (i < n) {
+            if (recs[i].depth == want_depth)
+                cs++;
+            else
+                cr++;
+        }
+    }
+    uint32_t excl, tot;
+    block_scan(lds, cs, &excl, &tot);
+    if (threadIdx.x == 0)
+        bc_sel[blockIdx.x] = tot;
+    block_scan(lds, cr, &excl, &tot);
+    if (threadIdx.x == 0)
+        bc_rest[blockIdx.x] = tot;
+}
+
+__global__ void k_part_gather(const node_rec *__restrict__ recs, uint64_t n,
+                              int want_depth, const uint32_t *__restrict__ off_sel,
+                              const uint32_t *__restrict__ off_rest,
+                              node_rec *__restrict__ out_sel,
+                              node_rec *__restrict__ out_rest)
+{
+    __shared__ uint32_t lds[BLOCK];
+    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SEL_ITEMS +
+                    (uint64_t)threadIdx.x * SEL_ITEMS;
+    uint32_t cs = 0, cr = 0;
+    for (int k = 0; k < SEL_ITEMS; ++k) {
+        uint64_t i = base + k;
+        if (i < n) {
+            if (recs[i].depth == want_depth)
+                cs++;
+            else
+                cr++;
+        }
+    }
+    uint32_t es, er, tot;
+    block_scan(lds, cs, &es, &tot);
+    block_scan(lds, cr, &er, &tot);
+    uint32_t ps = off_sel[blockIdx.x] + es;
+    uint32_t pr = off_rest[blockIdx.x] + er;
+    for (int k = 0; k < SEL_ITEMS; ++k) {
+        uint64_t i = base + k;
+        if (i < n) {
+            if (recs[i].depth == want_depth)
+                copy_rec(&out_sel[ps++], &recs[i]);
+            else
+                copy_rec(&out_rest[pr++], &recs[i]);
+        }
+    }
+}
+
+// merge two record arrays sorted by .s (distinct keys)
+__global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
+                          const node_rec *__restrict__ B, uint64_t nB,
+                          node_rec *__restrict__ out)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= nA)
+        return;
+    uint32_t key = A[i].s;
+    uint64_t lo = 0, hi = nB;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2;
+        if (B[mid].s < key)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    copy_rec(&out[i + lo], &A[i]);
+}
+
+__global__ void k_merge_b(const node_rec *__restrict__ A, uint64_t nA,
+                          const node_rec *__restrict__ B, uint64_t nB,
+                          node_rec *__restrict__ out)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= nB)
+        return;
+    uint32_t key = B[j].s;
+    uint64_t lo = 0, hi = nA;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2;
+        if (A[mid].s < key)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    copy_rec(&out[j + lo], &B[j]);
+}
+
+__global__ void k_group_flags(const node_rec *__restrict__ L, uint64_t n,
+                              const int8_t *__restrict__ lcp, int d,
+                              uint32_t *__restrict__ flags)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= n)
+        return;
+    uint32_t f;
+    if (j == 0 || L[j].s != L[j - 1].e)
+        f = 1;
+    else
+        f = (lcp[L[j].s] < d) ? 1u : 0u;
+    flags[j] = f;
+}
+
+// ---------------------------------------------------------------------------
+// branch kernel
+// ---------------------------------------------------------------------------
+
+#define SLOT_BR 552 // >= 4*136, 69-u64 stride (conflict-free b64 LDS reads)
+
+__global__ void __launch_bounds__(BLOCK) k_branch(
+    const node_rec *__restrict__ L, uint64_t n, const uint32_t *__restrict__ flags,
+    const uint32_t *__restrict__ gidx, const int8_t *__restrict__ lcp,
+    const uint8_t *__restrict__ keys, uint64_t key_stride, int d, int subtree,
+    node_rec *__restrict__ out, uint8_t *__restrict__ seg_roots,
+    uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
+    uint32_t *__restrict__ pending, uint32_t *__restrict__ err)
+{
+    __shared__ uint8_t lds[BLOCK * SLOT_BR];
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= n || !flags[j])
+        return;
+    uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_BR;
+    uint64_t *slot64 = (uint64_t *)slot;
+
+    uint64_t jend = j + 1;
+    while (jend < n && !flags[jend])
+        jend++;
+    int nmem = (int)(jend - j);
+    if (nmem < 2 || nmem > 16) {
+        atomicOr(err, 1u << E_INTERNAL);
+        return;
+    }
+    uint32_t s_first = L[j].s;
+    uint32_t e_last = L[jend - 1].e;
+    uint32_t seg = L[j].seg;
+    const uint8_t *key0 = keys + (uint64_t)s_first * key_stride;
+
+#pragma unroll 8
+    for (int k = 0; k < SLOT_BR / 8; ++k)
+        slot64[k] = 0;
+
+    // pass 1: payload length (members are in ascending-nibble order)
+    int payload = 1; // trailing empty value item
+    {
+        uint64_t m = j;
+        for (int b = 0; b < 16; ++b) {
+            if (m < jend && nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b) {
+                payload += L[m].ref_len;
+                m++;
+            } else {
+                payload += 1;
+            }
+        }
+        if (m != jend) { // members not in strict nibble order -> duplicate nibble
+            atomicOr(err, 1u << E_INTERNAL);
+            return;
+        }
+    }
+    int h = rlp_list_hdr_write(slot, payload);
+    // pass 2: write children
+    {
+        int p = h;
+        uint64_t m = j;
+        for (int b = 0; b < 16; ++b) {
+            if (m < jend && nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b) {
+                int rl = L[m].ref_len;
+                for (int k = 0; k < rl; ++k)
+                    slot[p + k] = L[m].ref[k];
+                p += rl;
+                m++;
+            } else {
+                slot[p++] = 0x80;
+            }
+        }
+        slot[p++] = 0x80;
+    }
+    int br_len = h + payload;
+    int nblocks = keccak_pad(slot, br_len);
+    uint64_t br_hash[4];
+    keccak_lds(slot64, nblocks, br_hash);
+    uint8_t br_ref[33], br_ref_len;
+    make_ref(slot, br_len, br_hash, br_ref, &br_ref_len);
+
+    int8_t pl = lcp[s_first], pr = lcp[e_last];
+    int P = pl > pr ? pl : pr;
+
+    // extension wrap over key0[from..d); produces hash (keccak of outermost
+    // node rlp) + embedded ref. Reuses the slot.
+    auto wrap = [&](int from, uint64_t hash[4], uint8_t *ref, uint8_t *ref_len) {
+        if (d == from) {
+#pragma unroll
+            for (int k = 0; k < 4; ++k)
+                hash[k] = br_hash[k];
+            for (int k = 0; k < br_ref_len; ++k)
+                ref[k] = br_ref[k];
+            *ref_len = br_ref_len;
+            return;
+        }
+#pragma unroll 8
+        for (int k = 0; k < SLOT_BR / 8; ++k)
+            slot64[k] = 0;
+        int pay = hp_item_len(from, d) + br_ref_len;
+        int hh = rlp_list_hdr_write(slot, pay);
+        int p = hh + hp_item_write(slot + hh, key0, from, d, 0);
+        for (int k = 0; k < br_ref_len; ++k)
+            slot[p++] = br_ref[k];
+        int len = hh + pay;
+        int nb2 = keccak_pad(slot, len);
+        keccak_lds(slot64, nb2, hash);
+        make_ref(slot, len, hash, ref, ref_len);
+    };
+
+    node_rec r;
+    r.s = s_first;
+    r.e = e_last;
+    r.seg = seg;
+    r.pad_ = 0;
+    if (P >= 0) {
+        r.depth = (int8_t)P;
+        uint64_t hash[4];
+        wrap(P + 1, hash, r.ref, &r.ref_len);
+        atomicAdd(&pending[P + 1], 1u);
+    } else {
+        r.depth = -1;
+        r.ref_len = 0; // dead record: root emitted below
+        uint64_t hash[4];
+        uint8_t ref[33], ref_len;
+        if (subtree) {
+            wrap(1, hash, ref, &ref_len); // child-of-root-branch form
+            for (int k = 0; k < ref_len; ++k)
+                child_refs[33ull * seg + k] = ref[k];
+            child_lens[seg] = ref_len;
+        }
+        wrap(0, hash, ref, &ref_len); // standalone form; root = keccak(rlp)
+        memcpy(seg_roots + 32ull * seg, hash, 32);
+    }
+    copy_rec(&out[gidx[j]], &r);
+}
+
+// ---------------------------------------------------------------------------
+// misc kernels
+// ---------------------------------------------------------------------------
+
+__global__ void k_scatter_roots(const uint8_t *__restrict__ seg_roots,
+                                const uint32_t *__restrict__ seg_acct, uint32_t n_seg,
+                                uint8_t *__restrict__ acct_roots)
+{
+    uint32_t s = blockIdx.x * blockDim.x + threadIdx.x;
+    if (s >= n_seg)
+        return;
+    const uint64_t *src = (const uint64_t *)(seg_roots + 32ull * s);
+    uint64_t *dst = (uint64_t *)(acct_roots + 32ull * seg_acct[s]);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+        dst[k] = src[k];
+}
+
+__global__ void k_fill_empty_roots(uint8_t *__restrict__ acct_roots, uint64_t na)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= na)
+        return;
+    for (int k = 0; k < 32; ++k)
+        acct_roots[32 * i + k] = D_EMPTY_ROOT[k];
+}
+
+__global__ void k_nibble_counts(const sre_account_entry *__restrict__ acct,
+                                uint64_t na, uint64_t *__restrict__ counts)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= na)
+        return;
+    atomicAdd((unsigned long long *)&counts[acct[i].key[0] >> 4], 1ull);
+}
+
+// root-branch assembly from gathered child refs (single thread; O(1) work)
+__global__ void k_finish_top(const uint8_t *__restrict__ child_refs,
+                             const uint8_t *__restrict__ child_lens,
+                             uint8_t *__restrict__ out_root)
+{
+    __shared__ uint8_t slot[SLOT_BR];
+    if (threadIdx.x != 0 || blockIdx.x != 0)
+        return;
+    uint64_t *slot64 = (uint64_t *)slot;
+    for (int k = 0; k < SLOT_BR / 8; ++k)
+        slot64[k] = 0;
+    int payload = 1;
+    for (int b = 0; b < 16; ++b)
+        payload += child_lens[b] ? child_lens[b] : 1;
+    int h = rlp_list_hdr_write(slot, payload);
+    int p = h;
+    for (int b = 0; b < 16; ++b) {
+        if (child_lens[b]) {
+            for (int k = 0; k < child_lens[b]; ++k)
+                slot[p++] = child_refs[33 * b + k];
+        } else {
+            slot[p++] = 0x80;
+        }
+    }
+    slot[p++] = 0x80;
+    int len = h + payload;
+    int nblocks = keccak_pad(slot, len);
+    uint64_t hash[4];
+    keccak_lds(slot64, nblocks, hash);
+    memcpy(out_root, hash, 32);
+}
+
+// ---------------------------------------------------------------------------
+// scan utilities (exclusive u32, recursive)
+// ---------------------------------------------------------------------------
+
+#define SCAN_ITEMS 8
+
+__global__ void k_scan_block(const uint32_t *__restrict__ in, uint64_t n,
+                             uint32_t *__restrict__ out, uint32_t *__restrict__ sums)
+{
+    __shared__ uint32_t lds[BLOCK];
+    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SCAN_ITEMS +
+                    (uint64_t)threadIdx.x * SCAN_ITEMS;
+    uint32_t v[SCAN_ITEMS];
+    uint32_t tsum = 0;
+#pragma unroll
+    for (int k = 0; k < SCAN_ITEMS; ++k) {
+        uint64_t i = base + k;
+        v[k] = i < n ? in[i] : 0;
+        tsum += v[k];
+    }
+    uint32_t excl, total;
+    block_scan(lds, tsum, &excl, &total);
+    if (threadIdx.x == 0 && sums)
+        sums[blockIdx.x] = total;
+    uint32_t run = excl;
+#pragma unroll
+    for (int k = 0; k < SCAN_ITEMS; ++k) {
+        uint64_t i = base + k;
+        if (i < n)
+            out[i] = run;
+        run += v[k];
+    }
+}
+
+__global__ void k_scan_add(uint32_t *__restrict__ out, uint64_t n,
+                           const uint32_t *__restrict__ sums_scanned)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= n)
+        return;
+    out[i] += sums_scanned[i / (BLOCK * SCAN_ITEMS)];
+}
+
+// ---------------------------------------------------------------------------
+// host side
+// ---------------------------------------------------------------------------
+
+#define HIP_CHECK(ctx, call)                                                     \
+    do {                                                                         \
+        hipError_t _e = (call);                                                  \
+        if (_e != hipSuccess) {                                                  \
+            set_err(ctx, std::string(#call) + ": " + hipGetErrorString(_e));     \
+            return -1;                                                           \
+        }                                                                        \
+    } while (0)
+
+struct sre_ctx {
+    int device = 0;
+    hipStream_t stream = nullptr;
+    std::string err;
+    const sre_account_entry *d_acct = nullptr;
+    uint64_t na = 0;
+    bool own_acct = false;
+    const sre_storage_entry *d_st = nullptr;
+    uint64_t ns = 0;
+    bool own_st = false;
+    sre_stats stats{};
+};
+
+static std::string g_err;
+
+static void set_err(sre_ctx *ctx, const std::string &msg)
+{
+    if (ctx)
+        ctx->err = msg;
+    else
+        g_err = msg;
+}
+
+extern "C" const char *sre_last_error(const sre_ctx *ctx)
+{
+    return ctx ? ctx->err.c_str() : g_err.c_str();
+}
+
+extern "C" sre_ctx *sre_create(int device)
+{
+    int count = 0;
+    hipError_t e = hipGetDeviceCount(&count);
+    if (e != hipSuccess || count <= device) {
+        set_err(nullptr, "sre_create: no HIP device " + std::to_string(device) +
+                             " (this engine is GPU-only by design)");
+        return nullptr;
+    }
+    if (hipSetDevice(device) != hipSuccess) {
+        set_err(nullptr, "sre_create: hipSetDevice failed");
+        return nullptr;
+    }
+    sre_ctx *ctx = new sre_ctx();
+    ctx->device = device;
+    if (hipStreamCreate(&ctx->stream) != hipSuccess) {
+        set_err(nullptr, "sre_create: hipStreamCreate failed");
+        delete ctx;
+        return nullptr;
+    }
+    return ctx;
+}
+
+extern "C" void sre_destroy(sre_ctx *ctx)
+{
+    if (!ctx)
+        return;
+    if (ctx->own_acct && ctx->d_acct)
+        hipFree((void *)ctx->d_acct);
+    if (ctx->own_st && ctx->d_st)
+        hipFree((void *)ctx->d_st);
+    if (ctx->stream)
+        hipStreamDestroy(ctx->stream);
+    delete ctx;
+}
+
+extern "C" int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entries,
+                                   uint64_t n)
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    if (ctx->own_acct && ctx->d_acct)
+        hipFree((void *)ctx->d_acct);
+    ctx->d_acct = nullptr;
+    void *p = nullptr;
+    if (n) {
+        HIP_CHECK(ctx, hipMalloc(&p, n * sizeof(sre_account_entry)));
+        HIP_CHECK(ctx, hipMemcpy(p, entries, n * sizeof(sre_account_entry),
+                                 hipMemcpyHostToDevice));
+    }
+    ctx->d_acct = (const sre_account_entry *)p;
+    ctx->na = n;
+    ctx->own_acct = true;
+    return 0;
+}
+
+extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries,
+                                  uint64_t n)
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    if (ctx->own_st && ctx->d_st)
+        hipFree((void *)ctx->d_st);
+    ctx->d_st = nullptr;
+    void *p = nullptr;
+    if (n) {
+        HIP_CHECK(ctx, hipMalloc(&p, n * sizeof(sre_storage_entry)));
+        HIP_CHECK(ctx, hipMemcpy(p, entries, n * sizeof(sre_storage_entry),
+                                 hipMemcpyHostToDevice));
+    }
+    ctx->d_st = (const sre_storage_entry *)p;
+    ctx->ns = n;
+    ctx->own_st = true;
+    return 0;
+}
+
+// Borrow device-resident inputs (zero-copy; caller keeps them alive).
+extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint64_t n)
+{
+    if (ctx->own_acct && ctx->d_acct)
+        hipFree((void *)ctx->d_acct);
+    ctx->d_acct = (const sre_account_entry *)d_entries;
+    ctx->na = n;
+    ctx->own_acct = false;
+    return 0;
+}
+
+extern "C" int sre_set_storage_device(sre_ctx *ctx, const void *d_entries, uint64_t n)
+{
+    if (ctx->own_st && ctx->d_st)
+        hipFree((void *)ctx->d_st);
+    ctx->d_st = (const sre_storage_entry *)d_entries;
+    ctx->ns = n;
+    ctx->own_st = false;
+    return 0;
+}
+
+extern "C" int sre_keccak_batch_device(sre_ctx *ctx, const void *d_in, uint64_t stride,
+                                       uint32_t len, uint64_t n, void *d_out)
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    if (len > 135) {
+        set_err(ctx, "sre_keccak_batch_device: len > 135 unsupported");
+        return -1;
+    }
+    if (n == 0)
+        return 0;
+    uint64_t blocks = (n + BLOCK - 1) / BLOCK;
+    hipLaunchKernelGGL(k_keccak_batch, dim3((uint32_t)blocks), dim3(BLOCK), 0,
+                       ctx->stream, (const uint8_t *)d_in, stride, len, n,
+                       (uint8_t *)d_out);
+    HIP_CHECK(ctx, hipGetLastError());
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    return 0;
+}
+
+struct DBuf {
+    void *p = nullptr;
+    ~DBuf()
+    {
+        if (p)
+            hipFree(p);
+    }
+    hipError_t alloc(size_t bytes)
+    {
+        if (p) {
+            hipFree(p);
+            p = nullptr;
+        }
+        return hipMalloc(&p, bytes ? bytes : 16);
+    }
+    template <typename T> T *as() { return (T *)p; }
+};
+
+static inline uint32_t grid_for(uint64_t n)
+{
+    return (uint32_t)((n + BLOCK - 1) / BLOCK);
+}
+
+static inline uint32_t sel_grid_for(uint64_t n)
+{
+    uint64_t per = (uint64_t)BLOCK * SEL_ITEMS;
+    return (uint32_t)((n + per - 1) / per);
+}
+
+// exclusive scan of u32 in[n] -> out[n]; total returned synchronously.
+static int scan_u32(sre_ctx *ctx, const uint32_t *d_in, uint32_t *d_out, uint64_t n,
+                    uint32_t *total)
+{
+    if (n == 0) {
+        if (total)
+            *total = 0;
+        return 0;
+    }
+    uint64_t per_block = (uint64_t)BLOCK * SCAN_ITEMS;
+    uint64_t nblocks = (n + per_block - 1) / per_block;
+    DBuf sums, sums_scanned;
+    HIP_CHECK(ctx, sums.alloc(nblocks * 4));
+    hipLaunchKernelGGL(k_scan_block, dim3((uint32_t)nblocks), dim3(BLOCK), 0,
+                       ctx->stream, d_in, n, d_out, sums.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    if (nblocks > 1) {
+        HIP_CHECK(ctx, sums_scanned.alloc(nblocks * 4));
+        if (scan_u32(ctx, sums.as<uint32_t>(), sums_scanned.as<uint32_t>(), nblocks,
+                     total))
+            return -1;
+        hipLaunchKernelGGL(k_scan_add, dim3(grid_for(n)), dim3(BLOCK), 0, ctx->stream,
+                           d_out, n, sums_scanned.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+    } else if (total) {
+        HIP_CHECK(ctx, hipMemcpyAsync(total, sums.as<uint32_t>(), 4,
+                                      hipMemcpyDeviceToHost, ctx->stream));
+        HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    }
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// level machinery driver
+// ---------------------------------------------------------------------------
+
+struct pass_out {
+    double leaf_ms = 0, branch_ms = 0;
+    uint64_t leaf_count = 0, leaf_blocks = 0, branch_count = 0, branch_blocks = 0;
+    uint64_t levels = 0;
+};
+
+static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_depths,
+                      const int8_t *d_lcp, const uint8_t *d_keys, uint64_t key_stride,
+                      const uint32_t *hist_host, int subtree, uint8_t *d_seg_roots,
+                      uint8_t *d_child_refs, uint8_t *d_child_lens, uint32_t *d_err,
+                      pass_out *po)
+{
+    int maxd = -1;
+    for (int d = 63; d >= 0; --d)
+        if (hist_host[d + 1]) {
+            maxd = d;
+            break;
+        }
+    if (maxd < 0)
+        return 0; // every leaf was already a segment root
+
+    DBuf carry, carry2, Lsel, Csel, Crest, Lbuf, newn, dead, flags, gidx, pend;
+    DBuf blk_a, blk_b, off_a, off_b;
+    uint64_t carry_n = 0;
+    uint64_t max_blocks = sel_grid_for(n) + 2;
+    HIP_CHECK(ctx, blk_a.alloc(max_blocks * 4));
+    HIP_CHECK(ctx, blk_b.alloc(max_blocks * 4));
+    HIP_CHECK(ctx, off_a.alloc(max_blocks * 4));
+    HIP_CHECK(ctx, off_b.alloc(max_blocks * 4));
+    HIP_CHECK(ctx, pend.alloc(66 * 4));
+    HIP_CHECK(ctx, hipMemsetAsync(pend.p, 0, 66 * 4, ctx->stream));
+
+    uint32_t pending_host[66] = {0};
+    hipEvent_t ev0, ev1;
+    hipEventCreate(&ev0);
+    hipEventCreate(&ev1);
+
+    for (int d = maxd; d >= 0; --d) {
+        uint64_t nA = hist_host[d + 1];
+        uint64_t nB = pending_host[d + 1];
+        if (nA == 0 && nB == 0)
+            continue;
+        po->levels++;
+        uint64_t n_level = nA + nB;
+        uint64_t n_rest = carry_n - nB;
+
+        // 1. select leaves with parent depth == d
+        if (nA) {
+            HIP_CHECK(ctx, Lsel.alloc(nA * sizeof(node_rec)));
+            uint32_t g = sel_grid_for(n);
+            hipLaunchKernelGGL(k_sel_count, dim3(g), dim3(BLOCK), 0, ctx->stream,
+                               d_depths, n, (uint32_t)(d + 1), blk_a.as<uint32_t>());
+            uint32_t tot = 0;
+            if (scan_u32(ctx, blk_a.as<uint32_t>(), off_a.as<uint32_t>(), g, &tot))
+                return -1;
+            if (tot != nA) {
+                set_err(ctx, "internal: leaf histogram mismatch");
+                return -1;
+            }
+            hipLaunchKernelGGL(k_sel_gather, dim3(g), dim3(BLOCK), 0, ctx->stream,
+                               d_depths, d_recs, n, (uint32_t)(d + 1),
+                               off_a.as<uint32_t>(), Lsel.as<node_rec>());
+            HIP_CHECK(ctx, hipGetLastError());
+        }
+        // 2. partition carry
+        if (nB) {
+            HIP_CHECK(ctx, Csel.alloc(nB * sizeof(node_rec)));
+            HIP_CHECK(ctx, Crest.alloc((n_rest ? n_rest : 1) * sizeof(node_rec)));
+            uint32_t g = sel_grid_for(carry_n);
+            hipLaunchKernelGGL(k_part_count, dim3(g), dim3(BLOCK), 0, ctx->stream,
+                               carry.as<node_rec>(), carry_n, d,
+                               blk_a.as<uint32_t>(), blk_b.as<uint32_t>());
+            uint32_t ts = 0, tr = 0;
+            if (scan_u32(ctx, blk_a.as<uint32_t>(), off_a.as<uint32_t>(), g, &ts))
+                return -1;
+            if (scan_u32(ctx, blk_b.as<uint32_t>(), off_b.as<uint32_t>(), g, &tr))
+                return -1;
+            if (ts != nB || tr != n_rest) {
+                set_err(ctx, "internal: pending count mismatch");
+                return -1;
+            }
+            hipLaunchKernelGGL(k_part_gather, dim3(g), dim3(BLOCK), 0, ctx->stream,
+                               carry.as<node_rec>(), carry_n, d, off_a.as<uint32_t>(),
+                               off_b.as<uint32_t>(), Csel.as<node_rec>(),
+                               Crest.as<node_rec>());
+            HIP_CHECK(ctx, hipGetLastError());
+        }
+        // 3. merge into level input L
+        node_rec *L;
+        if (nB == 0) {
+            L = Lsel.as<node_rec>();
+        } else if (nA == 0) {
+            L = Csel.as<node_rec>();
+        } else {
+            HIP_CHECK(ctx, Lbuf.alloc(n_level * sizeof(node_rec)));
+            hipLaunchKernelGGL(k_merge_a, dim3(grid_for(nA)), dim3(BLOCK), 0,
+                               ctx->stream, Lsel.as<node_rec>(), nA,
+                               Csel.as<node_rec>(), nB, Lbuf.as<node_rec>());
+            hipLaunchKernelGGL(k_merge_b, dim3(grid_for(nB)), dim3(BLOCK), 0,
+                               ctx->stream, Lsel.as<node_rec>(), nA,
+                               Csel.as<node_rec>(), nB, Lbuf.as<node_rec>());
+            HIP_CHECK(ctx, hipGetLastError());
+            L = Lbuf.as<node_rec>();
+        }
+        // 4. group flags + scan
+        HIP_CHECK(ctx, flags.alloc(n_level * 4));
+        HIP_CHECK(ctx, gidx.alloc(n_level * 4));
+        hipLaunchKernelGGL(k_group_flags, dim3(grid_for(n_level)), dim3(BLOCK), 0,
+                           ctx->stream, L, n_level, d_lcp, d, flags.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        uint32_t n_groups = 0;
+        if (scan_u32(ctx, flags.as<uint32_t>(), gidx.as<uint32_t>(), n_level,
+                     &n_groups))
+            return -1;
+        // 5. branch kernel
+        HIP_CHECK(ctx, newn.alloc((uint64_t)n_groups * sizeof(node_rec)));
+        hipEventRecord(ev0, ctx->stream);
+        hipLaunchKernelGGL(k_branch, dim3(grid_for(n_level)), dim3(BLOCK), 0,
+                           ctx->stream, L, n_level, flags.as<uint32_t>(),
+                           gidx.as<uint32_t>(), d_lcp, d_keys, key_stride, d, subtree,
+                           newn.as<node_rec>(), d_seg_roots, d_child_refs,
+                           d_child_lens, pend.as<uint32_t>(), d_err);
+        HIP_CHECK(ctx, hipGetLastError());
+        hipEventRecord(ev1, ctx->stream);
+        HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+        float ms = 0;
+        hipEventElapsedTime(&ms, ev0, ev1);
+        po->branch_ms += ms;
+        po->branch_count += n_groups;
+
+        // 6. read updated pending counts; compact alive nodes into new carry.
+        // pend[d+1] held exactly nB (new nodes always have P < d): consume it
+        // on host AND device so later readbacks don't resurrect it.
+        HIP_CHECK(ctx, hipMemcpy(pending_host, pend.p, 66 * 4, hipMemcpyDeviceToHost));
+        pending_host[d + 1] = 0;
+        HIP_CHECK(ctx, hipMemsetAsync((uint8_t *)pend.p + 4 * (d + 1), 0, 4,
+                                      ctx->stream));
+        uint64_t n_alive = 0;
+        {
+            uint64_t alive_acc = 0;
+            for (int k = 0; k < 66; ++k)
+                alive_acc += pending_host[k];
+            n_alive = alive_acc - (carry_n - nB); // new nodes still pending
+        }
+        // new carry = merge(Crest, alive(newn))
+        uint64_t new_carry_n = n_rest + n_alive;
+        if (new_carry_n == 0) {
+            carry_n = 0;
+        } else {
+            HIP_CHECK(ctx, carry2.alloc(new_carry_n * sizeof(node_rec)));
+            // compact alive out of newn (dead records have depth == -1)
+            DBuf alive;
+            HIP_CHECK(ctx, alive.alloc((n_alive ? n_alive : 1) * sizeof(node_rec)));
+            if (n_groups) {
+                HIP_CHECK(ctx, dead.alloc(((uint64_t)n_groups) * sizeof(node_rec)));
+                uint32_t g = sel_grid_for(n_groups);
+                hipLaunchKernelGGL(k_part_count, dim3(g), dim3(BLOCK), 0, ctx->stream,
+                                   newn.as<node_rec>(), n_groups, -1,
+                                   blk_a.as<uint32_t>(), blk_b.as<uint32_t>());
+                uint32_t ts = 0, tr = 0;
+                if (scan_u32(ctx, blk_a.as<uint32_t>(), off_a.as<uint32_t>(), g, &ts))
+                    return -1;
+                if (scan_u32(ctx, blk_b.as<uint32_t>(), off_b.as<uint32_t>(), g, &tr))
+                    return -1;
+                if (tr != n_alive) {
+                    set_err(ctx, "internal: alive count mismatch");
+                    return -1;
+                }
+                hipLaunchKernelGGL(k_part_gather, dim3(g), dim3(BLOCK), 0, ctx->stream,
+                                   newn.as<node_rec>(), n_groups, -1,
+                                   off_a.as<uint32_t>(), off_b.as<uint32_t>(),
+                                   dead.as<node_rec>(), alive.as<node_rec>());
+                HIP_CHECK(ctx, hipGetLastError());
+            }
+            hipLaunchKernelGGL(k_merge_a, dim3(grid_for(n_rest ? n_rest : 1)),
+                               dim3(BLOCK), 0, ctx->stream,
+                               nB ? Crest.as<node_rec>() : carry.as<node_rec>(),
+                               n_rest, alive.as<node_rec>(), n_alive,
+                               carry2.as<node_rec>());
+            hipLaunchKernelGGL(k_merge_b, dim3(grid_for(n_alive ? n_alive : 1)),
+                               dim3(BLOCK), 0, ctx->stream,
+                               nB ? Crest.as<node_rec>() : carry.as<node_rec>(),
+                               n_rest, alive.as<node_rec>(), n_alive,
+                               carry2.as<node_rec>());
+            HIP_CHECK(ctx, hipGetLastError());
+            HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+            std::swap(carry.p, carry2.p);
+            carry_n = new_carry_n;
+        }
+    }
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+    if (carry_n != 0) {
+        set_err(ctx, "internal: carry not empty after level 0");
+        return -1;
+    }
+    return 0;
+}
+
+// ---------------------------------------------------------------------------
+// passes
+// ---------------------------------------------------------------------------
+
+static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
+                            uint32_t *d_err)
+{
+    uint64_t ns = ctx->ns, na = ctx->na;
+    hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(na)), dim3(BLOCK), 0,
+                       ctx->stream, d_acct_roots, na);
+    HIP_CHECK(ctx, hipGetLastError());
+    if (ns == 0)
+        return 0;
+
+    DBuf flags, seg_id, lcp, recs, depths, hist;
+    HIP_CHECK(ctx, flags.alloc(ns * 4));
+    HIP_CHECK(ctx, seg_id.alloc(ns * 4));
+    HIP_CHECK(ctx, lcp.alloc(ns + 1));
+    HIP_CHECK(ctx, recs.alloc(ns * sizeof(node_rec)));
+    HIP_CHECK(ctx, depths.alloc(ns));
+    HIP_CHECK(ctx, hist.alloc(66 * 4));
+    HIP_CHECK(ctx, hipMemsetAsync(hist.p, 0, 66 * 4, ctx->stream));
+
+    hipLaunchKernelGGL(k_seg_flags, dim3(grid_for(ns)), dim3(BLOCK), 0, ctx->stream,
+                       ctx->d_st, ns, flags.as<uint32_t>(), d_err);
+    HIP_CHECK(ctx, hipGetLastError());
+    // seg_id[i] = number of segment starts at or before i, minus 1 for the
+    // start itself: the exclusive scan gives exactly the segment index for
+    // starts (starts-before excludes self) and non-starts alike.
+    uint32_t n_seg = 0;
+    if (scan_u32(ctx, flags.as<uint32_t>(), seg_id.as<uint32_t>(), ns, &n_seg))
+        return -1;
+
+    DBuf seg_start, seg_acct, seg_roots;
+    HIP_CHECK(ctx, seg_start.alloc((uint64_t)n_seg * 4));
+    HIP_CHECK(ctx, seg_acct.alloc((uint64_t)n_seg * 4));
+    HIP_CHECK(ctx, seg_roots.alloc((uint64_t)n_seg * 32));
+    hipLaunchKernelGGL(k_seg_starts, dim3(grid_for(ns)), dim3(BLOCK), 0, ctx->stream,
+                       flags.as<uint32_t>(), seg_id.as<uint32_t>(), ns,
+                       seg_start.as<uint32_t>());
+    hipLaunchKernelGGL(k_seg_acct, dim3(grid_for(n_seg)), dim3(BLOCK), 0, ctx->stream,
+                       ctx->d_st, seg_start.as<uint32_t>(), n_seg, ctx->d_acct, na,
+                       seg_acct.as<uint32_t>(), d_err);
+    hipLaunchKernelGGL(k_lcp_storage, dim3(grid_for(ns + 1)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_st, ns, seg_id.as<uint32_t>(),
+                       lcp.as<int8_t>(), d_err);
+    HIP_CHECK(ctx, hipGetLastError());
+
+    hipEvent_t ev0, ev1;
+    hipEventCreate(&ev0);
+    hipEventCreate(&ev1);
+    hipEventRecord(ev0, ctx->stream);
+    hipLaunchKernelGGL(k_leaf_storage, dim3(grid_for(ns)), dim3(BLOCK), 0, ctx->stream,
+                       ctx->d_st, ns, lcp.as<int8_t>(), seg_id.as<uint32_t>(),
+                       recs.as<node_rec>(), depths.as<uint8_t>(), hist.as<uint32_t>(),
+                       seg_roots.as<uint8_t>(), d_err);
+    HIP_CHECK(ctx, hipGetLastError());
+    hipEventRecord(ev1, ctx->stream);
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    po->leaf_ms += ms;
+    po->leaf_count += ns;
+    po->leaf_blocks += ns;
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+
+    uint32_t hist_host[66];
+    HIP_CHECK(ctx, hipMemcpy(hist_host, hist.p, 66 * 4, hipMemcpyDeviceToHost));
+
+    const uint8_t *keys =
+        (const uint8_t *)ctx->d_st + offsetof(sre_storage_entry, slot_key);
+    if (run_levels(ctx, ns, recs.as<node_rec>(), depths.as<uint8_t>(),
+                   lcp.as<int8_t>(), keys, sizeof(sre_storage_entry), hist_host, 0,
+                   seg_roots.as<uint8_t>(), nullptr, nullptr, d_err, po))
+        return -1;
+
+    hipLaunchKernelGGL(k_scatter_roots, dim3(grid_for(n_seg)), dim3(BLOCK), 0,
+                       ctx->stream, seg_roots.as<uint8_t>(), seg_acct.as<uint32_t>(),
+                       n_seg, d_acct_roots);
+    HIP_CHECK(ctx, hipGetLastError());
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    return 0;
+}
+
+static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int subtree,
+                            uint8_t *d_roots, uint8_t *d_child_refs,
+                            uint8_t *d_child_lens, pass_out *po, uint32_t *d_err)
+{
+    uint64_t na = ctx->na;
+    DBuf lcp, recs, depths, hist;
+    HIP_CHECK(ctx, lcp.alloc(na + 1));
+    HIP_CHECK(ctx, recs.alloc(na * sizeof(node_rec)));
+    HIP_CHECK(ctx, depths.alloc(na));
+    HIP_CHECK(ctx, hist.alloc(66 * 4));
+    HIP_CHECK(ctx, hipMemsetAsync(hist.p, 0, 66 * 4, ctx->stream));
+
+    hipLaunchKernelGGL(k_lcp_account, dim3(grid_for(na + 1)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_acct, na, subtree, lcp.as<int8_t>(), d_err);
+    HIP_CHECK(ctx, hipGetLastError());
+
+    hipEvent_t ev0, ev1;
+    hipEventCreate(&ev0);
+    hipEventCreate(&ev1);
+    hipEventRecord(ev0, ctx->stream);
+    hipLaunchKernelGGL(k_leaf_account, dim3(grid_for(na)), dim3(BLOCK), 0, ctx->stream,
+                       ctx->d_acct, na, d_storage_roots, lcp.as<int8_t>(), subtree,
+                       recs.as<node_rec>(), depths.as<uint8_t>(), hist.as<uint32_t>(),
+                       d_roots, d_child_refs, d_child_lens);
+    HIP_CHECK(ctx, hipGetLastError());
+    hipEventRecord(ev1, ctx->stream);
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    po->leaf_ms += ms;
+    po->leaf_count += na;
+    po->leaf_blocks += 2 * na;
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+
+    uint32_t hist_host[66];
+    HIP_CHECK(ctx, hipMemcpy(hist_host, hist.p, 66 * 4, hipMemcpyDeviceToHost));
+
+    const uint8_t *keys = (const uint8_t *)ctx->d_acct + offsetof(sre_account_entry, key);
+    return run_levels(ctx, na, recs.as<node_rec>(), depths.as<uint8_t>(),
+                      lcp.as<int8_t>(), keys, sizeof(sre_account_entry), hist_host,
+                      subtree, d_roots, d_child_refs, d_child_lens, d_err, po);
+}
+
+static int check_err(sre_ctx *ctx, uint32_t *d_err)
+{
+    uint32_t e = 0;
+    HIP_CHECK(ctx, hipMemcpy(&e, d_err, 4, hipMemcpyDeviceToHost));
+    if (e) {
+        std::string msg = "input/engine error:";
+        if (e & (1u << E_UNSORTED_ACCT))
+            msg += " accounts-not-sorted";
+        if (e & (1u << E_UNSORTED_STORAGE))
+            msg += " storage-not-sorted-or-duplicate";
+        if (e & (1u << E_ORPHAN_STORAGE))
+            msg += " storage-for-unknown-account";
+        if (e & (1u << E_ZERO_VALUE))
+            msg += " zero-storage-value";
+        if (e & (1u << E_INTERNAL))
+            msg += " internal-group-invariant";
+        set_err(ctx, msg);
+        return -1;
+    }
+    return 0;
+}
+
+static const uint8_t EMPTY_ROOT_H[32] = {
+    0x56, 0xe8, 0x1f, 0x17, 0x1b, 0xcc, 0x55, 0xa6, 0xff, 0x83, 0x45, 0xe6,
+    0x92, 0xc0, 0xf8, 0x6e, 0x5b, 0x48, 0xe0, 0x1b, 0x99, 0x6c, 0xad, 0xc0,
+    0x01, 0x62, 0x2f, 0xb5, 0xe3, 0x63, 0xb4, 0x21,
+};
+
+extern "C" int sre_root(sre_ctx *ctx, uint8_t out_root[32])
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    memset(&ctx->stats, 0, sizeof(ctx->stats));
+    if (ctx->na == 0) {
+        if (ctx->ns != 0) {
+            set_err(ctx, "storage entries without accounts");
+            return -1;
+        }
+        memcpy(out_root, EMPTY_ROOT_H, 32);
+        return 0;
+    }
+    hipEvent_t t0, t1;
+    hipEventCreate(&t0);
+    hipEventCreate(&t1);
+    hipEventRecord(t0, ctx->stream);
+
+    DBuf err, acct_roots, root;
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, acct_roots.alloc(ctx->na * 32));
+    HIP_CHECK(ctx, root.alloc(32));
+
+    pass_out po;
+    if (run_storage_pass(ctx, acct_roots.as<uint8_t>(), &po, err.as<uint32_t>()))
+        return -1;
+    if (run_account_pass(ctx, acct_roots.as<uint8_t>(), 0, root.as<uint8_t>(),
+                         nullptr, nullptr, &po, err.as<uint32_t>()))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+
+    hipEventRecord(t1, ctx->stream);
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    float total = 0;
+    hipEventElapsedTime(&total, t0, t1);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    ctx->stats.total_ms = total;
+    ctx->stats.leaf_hash_ms = po.leaf_ms;
+    ctx->stats.leaf_count = po.leaf_count;
+    ctx->stats.leaf_blocks = po.leaf_blocks;
+    ctx->stats.branch_hash_ms = po.branch_ms;
+    ctx->stats.branch_count = po.branch_count;
+    ctx->stats.branch_blocks = po.branch_blocks;
+    ctx->stats.levels = po.levels;
+
+    HIP_CHECK(ctx, hipMemcpy(out_root, root.p, 32, hipMemcpyDeviceToHost));
+    return 0;
+}
+
+extern "C" int sre_storage_roots(sre_ctx *ctx, uint8_t *out, uint64_t n)
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    if (n != ctx->na) {
+        set_err(ctx, "sre_storage_roots: n != uploaded account count");
+        return -1;
+    }
+    if (n == 0)
+        return 0;
+    DBuf err, acct_roots;
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, acct_roots.alloc(ctx->na * 32));
+    pass_out po;
+    if (run_storage_pass(ctx, acct_roots.as<uint8_t>(), &po, err.as<uint32_t>()))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    HIP_CHECK(ctx, hipMemcpy(out, acct_roots.p, ctx->na * 32, hipMemcpyDeviceToHost));
+    return 0;
+}
+
+extern "C" int sre_subtree_roots(sre_ctx *ctx, uint8_t out_child_refs[16][33],
+                                 uint8_t out_child_lens[16],
+                                 uint8_t out_root_hash[16][32], uint64_t out_counts[16])
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    memset(&ctx->stats, 0, sizeof(ctx->stats));
+    memset(out_child_lens, 0, 16);
+    memset(out_counts, 0, 16 * 8);
+    if (ctx->na == 0)
+        return 0;
+    hipEvent_t t0, t1;
+    hipEventCreate(&t0);
+    hipEventCreate(&t1);
+    hipEventRecord(t0, ctx->stream);
+
+    DBuf err, acct_roots, roots, crefs, clens, counts;
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, acct_roots.alloc(ctx->na * 32));
+    HIP_CHECK(ctx, roots.alloc(16 * 32));
+    HIP_CHECK(ctx, crefs.alloc(16 * 33));
+    HIP_CHECK(ctx, clens.alloc(16));
+    HIP_CHECK(ctx, counts.alloc(16 * 8));
+    HIP_CHECK(ctx, hipMemsetAsync(clens.p, 0, 16, ctx->stream));
+    HIP_CHECK(ctx, hipMemsetAsync(counts.p, 0, 16 * 8, ctx->stream));
+
+    pass_out po;
+    if (run_storage_pass(ctx, acct_roots.as<uint8_t>(), &po, err.as<uint32_t>()))
+        return -1;
+    if (run_account_pass(ctx, acct_roots.as<uint8_t>(), 1, roots.as<uint8_t>(),
+                         crefs.as<uint8_t>(), clens.as<uint8_t>(), &po,
+                         err.as<uint32_t>()))
+        return -1;
+    hipLaunchKernelGGL(k_nibble_counts, dim3(grid_for(ctx->na)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_acct, ctx->na, counts.as<uint64_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    hipEventRecord(t1, ctx->stream);
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    float total = 0;
+    hipEventElapsedTime(&total, t0, t1);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    ctx->stats.total_ms = total;
+    ctx->stats.leaf_hash_ms = po.leaf_ms;
+    ctx->stats.leaf_count = po.leaf_count;
+    ctx->stats.leaf_blocks = po.leaf_blocks;
+    ctx->stats.branch_hash_ms = po.branch_ms;
+    ctx->stats.branch_count = po.branch_count;
+    ctx->stats.branch_blocks = po.branch_blocks;
+    ctx->stats.levels = po.levels;
+
+    HIP_CHECK(ctx, hipMemcpy(out_child_refs, crefs.p, 16 * 33, hipMemcpyDeviceToHost));
+    HIP_CHECK(ctx, hipMemcpy(out_child_lens, clens.p, 16, hipMemcpyDeviceToHost));
+    HIP_CHECK(ctx, hipMemcpy(out_root_hash, roots.p, 16 * 32, hipMemcpyDeviceToHost));
+    HIP_CHECK(ctx, hipMemcpy(out_counts, counts.p, 16 * 8, hipMemcpyDeviceToHost));
+    return 0;
+}
+
+extern "C" int sre_finish_top(sre_ctx *ctx, const uint8_t child_refs[16][33],
+                              const uint8_t child_lens[16],
+                              const uint8_t root_hash[16][32],
+                              const uint64_t counts[16], uint8_t out_root[32])
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    (void)counts;
+    int populated = 0, last = -1;
+    for (int b = 0; b < 16; ++b)
+        if (child_lens[b]) {
+            populated++;
+            last = b;
+        }
+    if (populated == 0) {
+        memcpy(out_root, EMPTY_ROOT_H, 32);
+        return 0;
+    }
+    if (populated == 1) {
+        memcpy(out_root, root_hash[last], 32);
+        return 0;
+    }
+    DBuf crefs, clens, root;
+    HIP_CHECK(ctx, crefs.alloc(16 * 33));
+    HIP_CHECK(ctx, clens.alloc(16));
+    HIP_CHECK(ctx, root.alloc(32));
+    HIP_CHECK(ctx, hipMemcpy(crefs.p, child_refs, 16 * 33, hipMemcpyHostToDevice));
+    HIP_CHECK(ctx, hipMemcpy(clens.p, child_lens, 16, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(k_finish_top, dim3(1), dim3(64), 0, ctx->stream,
+                       crefs.as<uint8_t>(), clens.as<uint8_t>(), root.as<uint8_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    HIP_CHECK(ctx, hipMemcpy(out_root, root.p, 32, hipMemcpyDeviceToHost));
+    return 0;
+}
+
+extern "C" int sre_get_stats(sre_ctx *ctx, sre_stats *out)
+{
+    *out = ctx->stats;
+    return 0;
+}

@@ -1,0 +1,180 @@
+"""ctypes binding over the C-ABI of the MI355X state-root engine (libsre.so).
+
+Product path: GPU-only. Raises RuntimeError at construction when no HIP
+device / extension is present — there is no CPU fallback (the CPU oracle
+under oracle/ is test infrastructure and must never be imported here).
+
+Mirrors the reference surface it replaces (see include/sre.h):
+  StateRootEngine.root()           ~ StateRoot::root() /
+                                     StateRootProvider::state_root
+                                     (/root/reference/crates/storage/storage-api/src/trie.rs:13-41)
+  StateRootEngine.storage_roots()  ~ StorageRootProvider::storage_root (:45-58)
+  subtree_roots()/finish_top()     ~ the multi-GPU decomposition of
+                                     StateRoot::calculate (crates/trie/trie/src/trie.rs:171)
+"""
+import ctypes
+import os
+import subprocess
+
+import numpy as np
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+LIB = os.path.join(HERE, "libsre.so")
+SRC = os.path.join(HERE, "csrc", "sre.hip")
+
+ACCOUNT_DTYPE = np.dtype([
+    ("key", np.uint8, 32),
+    ("nonce", np.uint64),
+    ("balance", np.uint8, 32),
+    ("code_hash", np.uint8, 32),
+])
+STORAGE_DTYPE = np.dtype([
+    ("acct_key", np.uint8, 32),
+    ("slot_key", np.uint8, 32),
+    ("value", np.uint8, 32),
+])
+
+
+class SreStats(ctypes.Structure):
+    _fields_ = [
+        ("total_ms", ctypes.c_double),
+        ("leaf_hash_ms", ctypes.c_double),
+        ("leaf_count", ctypes.c_uint64),
+        ("leaf_blocks", ctypes.c_uint64),
+        ("branch_hash_ms", ctypes.c_double),
+        ("branch_count", ctypes.c_uint64),
+        ("branch_blocks", ctypes.c_uint64),
+        ("sort_ms", ctypes.c_double),
+        ("levels", ctypes.c_uint64),
+    ]
+
+
+def build(verbose=False):
+    """Compile libsre.so for gfx950 (in-tree; the .so travels with the repo)."""
+    cmd = ["hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-shared",
+           "-fPIC", SRC, "-o", LIB]
+    r = subprocess.run(cmd, capture_output=True, text=True)
+    if r.returncode != 0:
+        raise RuntimeError(f"hipcc failed:\n{r.stdout}\n{r.stderr}")
+    if verbose:
+        print(f"built {LIB}")
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(LIB):
+            raise RuntimeError(
+                f"{LIB} missing — run __graft_entry__.build() / reth_amd.engine.build()")
+        _lib = ctypes.CDLL(LIB)
+        _lib.sre_create.restype = ctypes.c_void_p
+        _lib.sre_create.argtypes = [ctypes.c_int]
+        _lib.sre_last_error.restype = ctypes.c_char_p
+        _lib.sre_last_error.argtypes = [ctypes.c_void_p]
+        _lib.sre_destroy.argtypes = [ctypes.c_void_p]
+    return _lib
+
+
+def _np_ptr(arr):
+    if len(arr) == 0:
+        return None
+    return arr.ctypes.data_as(ctypes.c_void_p)
+
+
+class StateRootEngine:
+    def __init__(self, device=0):
+        self._lib = lib()
+        self._ctx = self._lib.sre_create(device)
+        if not self._ctx:
+            raise RuntimeError("sre_create failed (GPU required, no fallback): "
+                               + self._lib.sre_last_error(None).decode())
+        self._keep = []  # borrowed device tensors kept alive
+
+    def close(self):
+        if getattr(self, "_ctx", None):
+            self._lib.sre_destroy(self._ctx)
+            self._ctx = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    def _check(self, rc):
+        if rc != 0:
+            raise RuntimeError(self._lib.sre_last_error(
+                ctypes.c_void_p(self._ctx)).decode())
+
+    # ---- input upload ----
+    def upload(self, accounts: np.ndarray, storage: np.ndarray):
+        assert accounts.dtype == ACCOUNT_DTYPE and storage.dtype == STORAGE_DTYPE
+        self._check(self._lib.sre_upload_accounts(
+            ctypes.c_void_p(self._ctx), _np_ptr(accounts), len(accounts)))
+        self._check(self._lib.sre_upload_storage(
+            ctypes.c_void_p(self._ctx), _np_ptr(storage), len(storage)))
+
+    def set_device_tensors(self, acct_u8, st_u8):
+        """Borrow torch GPU tensors: acct (na,104) uint8, st (ns,96) uint8,
+        contiguous, laid out as sre_account_entry / sre_storage_entry."""
+        assert acct_u8.dtype.itemsize == 1 and acct_u8.is_contiguous()
+        assert st_u8.dtype.itemsize == 1 and st_u8.is_contiguous()
+        assert acct_u8.shape[1] == 104 and st_u8.shape[1] == 96
+        self._keep = [acct_u8, st_u8]
+        self._check(self._lib.sre_set_accounts_device(
+            ctypes.c_void_p(self._ctx), ctypes.c_void_p(acct_u8.data_ptr()),
+            acct_u8.shape[0]))
+        self._check(self._lib.sre_set_storage_device(
+            ctypes.c_void_p(self._ctx), ctypes.c_void_p(st_u8.data_ptr()),
+            st_u8.shape[0]))
+
+    # ---- compute ----
+    def root(self) -> bytes:
+        out = (ctypes.c_uint8 * 32)()
+        self._check(self._lib.sre_root(ctypes.c_void_p(self._ctx), out))
+        return bytes(out)
+
+    def storage_roots(self, n) -> np.ndarray:
+        out = np.empty((n, 32), dtype=np.uint8)
+        self._check(self._lib.sre_storage_roots(
+            ctypes.c_void_p(self._ctx), _np_ptr(out), n))
+        return out
+
+    def subtree_roots(self):
+        refs = np.zeros((16, 33), dtype=np.uint8)
+        lens = np.zeros(16, dtype=np.uint8)
+        roots = np.zeros((16, 32), dtype=np.uint8)
+        counts = np.zeros(16, dtype=np.uint64)
+        self._check(self._lib.sre_subtree_roots(
+            ctypes.c_void_p(self._ctx), _np_ptr(refs), _np_ptr(lens),
+            _np_ptr(roots), _np_ptr(counts)))
+        return refs, lens, roots, counts
+
+    def finish_top(self, refs, lens, roots, counts) -> bytes:
+        out = (ctypes.c_uint8 * 32)()
+        self._check(self._lib.sre_finish_top(
+            ctypes.c_void_p(self._ctx),
+            _np_ptr(np.ascontiguousarray(refs, dtype=np.uint8)),
+            _np_ptr(np.ascontiguousarray(lens, dtype=np.uint8)),
+            _np_ptr(np.ascontiguousarray(roots, dtype=np.uint8)),
+            _np_ptr(np.ascontiguousarray(counts, dtype=np.uint64)), out))
+        return bytes(out)
+
+    def keccak_batch_device(self, in_tensor, msg_len, out_tensor):
+        """Hash n messages of msg_len bytes at a fixed stride (device memory).
+        in_tensor: (n, stride) uint8 cuda tensor; out_tensor: (n, 32) uint8."""
+        n, stride = in_tensor.shape
+        self._check(self._lib.sre_keccak_batch_device(
+            ctypes.c_void_p(self._ctx), ctypes.c_void_p(in_tensor.data_ptr()),
+            ctypes.c_uint64(stride), ctypes.c_uint32(msg_len), ctypes.c_uint64(n),
+            ctypes.c_void_p(out_tensor.data_ptr())))
+        return out_tensor
+
+    def stats(self) -> dict:
+        s = SreStats()
+        self._check(self._lib.sre_get_stats(ctypes.c_void_p(self._ctx),
+                                            ctypes.byref(s)))
+        return {f[0]: getattr(s, f[0]) for f in SreStats._fields_}
