@@ -300,7 +300,7 @@ __global__ void __launch_bounds__(BLOCK) k_keccak_batch(
     const uint8_t *__restrict__ in, uint64_t stride, uint32_t len, uint64_t n,
     uint8_t *__restrict__ out)
 {
-    __shared__ uint8_t lds[BLOCK * SLOT_KB];
+    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_KB];
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= n)
         return;
@@ -430,7 +430,7 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_storage(
     uint32_t *__restrict__ hist, uint8_t *__restrict__ seg_roots,
     uint32_t *__restrict__ err)
 {
-    __shared__ uint8_t lds[BLOCK * SLOT_STO + 66 * 4];
+    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_STO + 66 * 4];
     uint32_t *hist_l = (uint32_t *)(lds + BLOCK * SLOT_STO);
     if (threadIdx.x < 66)
         hist_l[threadIdx.x] = 0;
@@ -503,7 +503,7 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
     uint32_t *__restrict__ hist, uint8_t *__restrict__ roots,
     uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens)
 {
-    __shared__ uint8_t lds[BLOCK * SLOT_ACC + 66 * 4];
+    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_ACC + 66 * 4];
     uint32_t *hist_l = (uint32_t *)(lds + BLOCK * SLOT_ACC);
     if (threadIdx.x < 66)
         hist_l[threadIdx.x] = 0;
@@ -798,7 +798,7 @@ __global__ void __launch_bounds__(BLOCK) k_branch(
     uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
     uint32_t *__restrict__ pending, uint32_t *__restrict__ err)
 {
-    __shared__ uint8_t lds[BLOCK * SLOT_BR];
+    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_BR];
     uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (j >= n || !flags[j])
         return;
@@ -961,7 +961,7 @@ __global__ void k_finish_top(const uint8_t *__restrict__ child_refs,
                              const uint8_t *__restrict__ child_lens,
                              uint8_t *__restrict__ out_root)
 {
-    __shared__ uint8_t slot[SLOT_BR];
+    __shared__ __align__(16) uint8_t slot[SLOT_BR];
     if (threadIdx.x != 0 || blockIdx.x != 0)
         return;
     uint64_t *slot64 = (uint64_t *)slot;
