@@ -342,6 +342,17 @@ __global__ void k_seg_flags(const sre_storage_entry *__restrict__ st, uint64_t n
     flags[i] = (l < 64) ? 1u : 0u;
 }
 
+// seg_id = inclusive_scan(flags) - 1: the host computes the exclusive scan,
+// this fixes it up per element (excl + flag - 1).
+__global__ void k_seg_fix(const uint32_t *__restrict__ flags,
+                          uint32_t *__restrict__ seg_id, uint64_t ns)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= ns)
+        return;
+    seg_id[i] += flags[i] - 1;
+}
+
 __global__ void k_seg_starts(const uint32_t *__restrict__ flags,
                              const uint32_t *__restrict__ seg_id, uint64_t ns,
                              uint32_t *__restrict__ seg_start)
@@ -1476,12 +1487,13 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
     hipLaunchKernelGGL(k_seg_flags, dim3(grid_for(ns)), dim3(BLOCK), 0, ctx->stream,
                        ctx->d_st, ns, flags.as<uint32_t>(), d_err);
     HIP_CHECK(ctx, hipGetLastError());
-    // seg_id[i] = number of segment starts at or before i, minus 1 for the
-    // start itself: the exclusive scan gives exactly the segment index for
-    // starts (starts-before excludes self) and non-starts alike.
+    // seg_id[i] = inclusive_scan(flags)[i] - 1 (segment index of entry i)
     uint32_t n_seg = 0;
     if (scan_u32(ctx, flags.as<uint32_t>(), seg_id.as<uint32_t>(), ns, &n_seg))
         return -1;
+    hipLaunchKernelGGL(k_seg_fix, dim3(grid_for(ns)), dim3(BLOCK), 0, ctx->stream,
+                       flags.as<uint32_t>(), seg_id.as<uint32_t>(), ns);
+    HIP_CHECK(ctx, hipGetLastError());
 
     DBuf seg_start, seg_acct, seg_roots;
     HIP_CHECK(ctx, seg_start.alloc((uint64_t)n_seg * 4));
