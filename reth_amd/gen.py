@@ -240,21 +240,28 @@ def gen_state_torch(n_accounts, slots_per_account, keccak_batch_device,
         msg = torch.cat([_bytes_le(cc), _bytes_le(jj)], dim=1).contiguous()
         skeys = torch.empty((ns, 32), dtype=torch.uint8, device=device)
         keccak_batch_device(msg, 16, skeys)
+        del msg
         grp = torch.arange(na, dtype=torch.int64, device=device) \
                    .repeat_interleave(slots_per_account)
         sord = _argsort_keys_t(skeys, group=grp)
+        del grp
         x = ((cc << 32) | jj) ^ _i64(S_VAL)
         L = (1 + (_sm64_t(x) & 31)).to(torch.int64)
         top = (1 + _umod(_sm64_t(x ^ _i64(S_VB)), 255)).to(torch.uint8)
         body = torch.cat([_bytes_le(_sm64_t(x ^ _i64(S_VK[k]))) for k in range(4)],
                          dim=1)
+        del cc, jj, x
         colim = torch.arange(32, device=device)[None, :]
         vals = torch.where(colim >= (32 - L)[:, None], body,
                            torch.zeros_like(body))
+        del body
         vals[torch.arange(ns, device=device), 32 - L] = top
+        del top, L
         st[:, 0:32] = acct[:, 0:32].repeat_interleave(slots_per_account, dim=0)[sord]
         st[:, 32:64] = skeys[sord]
+        del skeys
         st[:, 64:96] = vals[sord]
+        del vals, sord
     return acct, st
 
 
