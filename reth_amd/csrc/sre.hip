@@ -816,12 +816,25 @@ __global__ void __launch_bounds__(BLOCK) k_branch(
     uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_BR;
     uint64_t *slot64 = (uint64_t *)slot;
 
+    // on any invariant violation: flag the error and emit a dead record so
+    // out[] stays fully initialized for the downstream partition/merge.
+    auto bail = [&]() {
+        atomicOr(err, 1u << E_INTERNAL);
+        node_rec r;
+        r.s = L[j].s;
+        r.e = L[j].e;
+        r.seg = L[j].seg;
+        r.depth = -1;
+        r.ref_len = 0;
+        r.pad_ = 0;
+        copy_rec(&out[gidx[j]], &r);
+    };
     uint64_t jend = j + 1;
-    while (jend < n && !flags[jend])
+    while (jend < n && !flags[jend] && jend - j <= 17)
         jend++;
     int nmem = (int)(jend - j);
     if (nmem < 2 || nmem > 16) {
-        atomicOr(err, 1u << E_INTERNAL);
+        bail();
         return;
     }
     uint32_t s_first = L[j].s;
@@ -845,8 +858,11 @@ __global__ void __launch_bounds__(BLOCK) k_branch(
                 payload += 1;
             }
         }
-        if (m != jend) { // members not in strict nibble order -> duplicate nibble
-            atomicOr(err, 1u << E_INTERNAL);
+        // m != jend: members not in strict nibble order (duplicate nibble);
+        // payload bound: 16*33+1 = 529 max — anything larger means a corrupt
+        // record and would overflow the LDS slot.
+        if (m != jend || payload > 529) {
+            bail();
             return;
         }
     }
@@ -1268,6 +1284,8 @@ struct pass_out {
     uint64_t levels = 0;
 };
 
+static int check_err(sre_ctx *ctx, uint32_t *d_err);
+
 static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_depths,
                       const int8_t *d_lcp, const uint8_t *d_keys, uint64_t key_stride,
                       const uint32_t *hist_host, int subtree, uint8_t *d_seg_roots,
@@ -1509,6 +1527,11 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
                        ctx->stream, ctx->d_st, ns, seg_id.as<uint32_t>(),
                        lcp.as<int8_t>(), d_err);
     HIP_CHECK(ctx, hipGetLastError());
+    // input-contract violations (unsorted/orphan entries) are flagged by the
+    // kernels above; bail BEFORE the trie machinery runs on garbage lcps.
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    if (check_err(ctx, d_err))
+        return -1;
 
     hipEvent_t ev0, ev1;
     hipEventCreate(&ev0);
@@ -1562,6 +1585,9 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
     hipLaunchKernelGGL(k_lcp_account, dim3(grid_for(na + 1)), dim3(BLOCK), 0,
                        ctx->stream, ctx->d_acct, na, subtree, lcp.as<int8_t>(), d_err);
     HIP_CHECK(ctx, hipGetLastError());
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    if (check_err(ctx, d_err))
+        return -1;
 
     hipEvent_t ev0, ev1;
     hipEventCreate(&ev0);
