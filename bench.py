@@ -75,6 +75,9 @@ def main():
     t_gen = time.perf_counter() - t_gen0
     na, ns = acct_t.shape[0], st_t.shape[0]
     eng.set_device_tensors(acct_t, st_t)
+    # release torch's cached generation blocks back to HIP so the engine's
+    # own allocations (records, level buffers) don't OOM at the 10M x 64 size
+    torch.cuda.empty_cache()
 
     def step():
         if world > 1:

@@ -799,83 +799,85 @@ __global__ void k_group_flags(const node_rec *__restrict__ L, uint64_t n,
 // branch kernel
 // ---------------------------------------------------------------------------
 
-#define SLOT_BR 552 // >= 4*136, 69-u64 stride (conflict-free b64 LDS reads)
+#define SLOT_BR 552 // >= 4*136 zero-padded branch RLP slot in global scratch
 
-__global__ void __launch_bounds__(BLOCK) k_branch(
-    const node_rec *__restrict__ L, uint64_t n, const uint32_t *__restrict__ flags,
-    const uint32_t *__restrict__ gidx, const int8_t *__restrict__ lcp,
-    const uint8_t *__restrict__ keys, uint64_t key_stride, int d, int subtree,
-    node_rec *__restrict__ out, uint8_t *__restrict__ seg_roots,
-    uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
-    uint32_t *__restrict__ pending, uint32_t *__restrict__ err)
+// per-group metadata produced by the assemble kernel
+struct br_meta {
+    uint32_t s, e, seg;
+    uint16_t br_len;   // 0 marks an invariant-violation group (error flagged)
+    int8_t P;          // parent depth (-1 = segment root)
+    uint8_t d;         // branch depth (level)
+};
+static_assert(sizeof(br_meta) == 16, "br_meta must be 16 bytes");
+
+// scatter group-start positions: gs[gidx[j]] = j (gs[n_groups] set by host)
+__global__ void k_group_starts(const uint32_t *__restrict__ flags,
+                               const uint32_t *__restrict__ gidx, uint64_t n,
+                               uint32_t *__restrict__ gs)
 {
-    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_BR];
     uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (j >= n || !flags[j])
+    if (j >= n)
         return;
-    uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_BR;
+    if (flags[j])
+        gs[gidx[j]] = (uint32_t)j;
+}
+
+// Assemble one branch node per lane into a zero-padded, keccak-padded
+// SLOT_BR-byte global scratch slot. No LDS -> full occupancy hides the
+// scattered child-ref gathers.
+__global__ void k_branch_assemble(
+    const node_rec *__restrict__ L, const uint32_t *__restrict__ gs,
+    uint32_t n_groups, const int8_t *__restrict__ lcp,
+    const uint8_t *__restrict__ keys, uint64_t key_stride, int d,
+    uint8_t *__restrict__ scratch, br_meta *__restrict__ meta,
+    uint32_t *__restrict__ err)
+{
+    uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups)
+        return;
+    uint64_t j = gs[g], jend = gs[g + 1];
+    br_meta mt;
+    mt.s = L[j].s;
+    mt.e = L[jend - 1].e;
+    mt.seg = L[j].seg;
+    mt.d = (uint8_t)d;
+    int8_t pl = lcp[mt.s], pr = lcp[mt.e];
+    mt.P = pl > pr ? pl : pr;
+
+    uint8_t *slot = scratch + (uint64_t)g * SLOT_BR;
     uint64_t *slot64 = (uint64_t *)slot;
 
-    // on any invariant violation: flag the error and emit a dead record so
-    // out[] stays fully initialized for the downstream partition/merge.
-    auto bail = [&]() {
-        atomicOr(err, 1u << E_INTERNAL);
-        node_rec r;
-        r.s = L[j].s;
-        r.e = L[j].e;
-        r.seg = L[j].seg;
-        r.depth = -1;
-        r.ref_len = 0;
-        r.pad_ = 0;
-        copy_rec(&out[gidx[j]], &r);
-    };
-    uint64_t jend = j + 1;
-    while (jend < n && !flags[jend] && jend - j <= 17)
-        jend++;
     int nmem = (int)(jend - j);
-    if (nmem < 2 || nmem > 16) {
-        bail();
+    // validate ascending-nibble member order and compute the payload:
+    // absent nibbles contribute one 0x80 byte, present ones their ref,
+    // plus the trailing empty value item.
+    uint64_t m = j;
+    for (int b = 0; b < 16 && m < jend; ++b)
+        if (nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b)
+            m++;
+    int payload = 1 + (16 - nmem);
+    for (uint64_t k = j; k < jend; ++k)
+        payload += L[k].ref_len;
+    if (nmem < 2 || nmem > 16 || m != jend || payload > 529) {
+        atomicOr(err, 1u << E_INTERNAL);
+        mt.br_len = 0;
+        meta[g] = mt;
         return;
     }
-    uint32_t s_first = L[j].s;
-    uint32_t e_last = L[jend - 1].e;
-    uint32_t seg = L[j].seg;
-    const uint8_t *key0 = keys + (uint64_t)s_first * key_stride;
-
-#pragma unroll 8
+#pragma unroll 4
     for (int k = 0; k < SLOT_BR / 8; ++k)
         slot64[k] = 0;
-
-    // pass 1: payload length (members are in ascending-nibble order)
-    int payload = 1; // trailing empty value item
-    {
-        uint64_t m = j;
-        for (int b = 0; b < 16; ++b) {
-            if (m < jend && nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b) {
-                payload += L[m].ref_len;
-                m++;
-            } else {
-                payload += 1;
-            }
-        }
-        // m != jend: members not in strict nibble order (duplicate nibble);
-        // payload bound: 16*33+1 = 529 max — anything larger means a corrupt
-        // record and would overflow the LDS slot.
-        if (m != jend || payload > 529) {
-            bail();
-            return;
-        }
-    }
     int h = rlp_list_hdr_write(slot, payload);
-    // pass 2: write children
+    // pass 2: write children in nibble order
     {
         int p = h;
-        uint64_t m = j;
+        m = j;
         for (int b = 0; b < 16; ++b) {
             if (m < jend && nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b) {
                 int rl = L[m].ref_len;
+                const uint8_t *ref = L[m].ref;
                 for (int k = 0; k < rl; ++k)
-                    slot[p + k] = L[m].ref[k];
+                    slot[p + k] = ref[k];
                 p += rl;
                 m++;
             } else {
@@ -885,17 +887,73 @@ __global__ void __launch_bounds__(BLOCK) k_branch(
         slot[p++] = 0x80;
     }
     int br_len = h + payload;
-    int nblocks = keccak_pad(slot, br_len);
+    keccak_pad(slot, br_len);
+    mt.br_len = (uint16_t)br_len;
+    meta[g] = mt;
+}
+
+// Hash one branch per lane: absorb the padded slot straight from global,
+// then do extension/root wraps in a small LDS slot (88 B -> high occupancy).
+#define SLOT_EXT 88
+__global__ void __launch_bounds__(BLOCK) k_branch_hash(
+    const uint8_t *__restrict__ scratch, const br_meta *__restrict__ meta,
+    uint32_t n_groups, const uint8_t *__restrict__ keys, uint64_t key_stride,
+    int subtree, node_rec *__restrict__ out, uint8_t *__restrict__ seg_roots,
+    uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
+    uint32_t *__restrict__ pending, uint32_t *__restrict__ err)
+{
+    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_EXT];
+    uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups)
+        return;
+    br_meta mt = meta[g];
+    node_rec *r = &out[g]; // write the record in place (no scratch spill)
+    r->s = mt.s;
+    r->e = mt.e;
+    r->seg = mt.seg;
+    r->pad_ = 0;
+    if (mt.br_len == 0) { // error group: dead record (err already flagged)
+        r->depth = -1;
+        r->ref_len = 0;
+        return;
+    }
+    int d = mt.d;
+    const uint64_t *slot64 = (const uint64_t *)(scratch + (uint64_t)g * SLOT_BR);
+    int nblocks = mt.br_len / 136 + 1;
     uint64_t br_hash[4];
-    keccak_lds(slot64, nblocks, br_hash);
+    {
+        uint64_t s[25];
+#pragma unroll
+        for (int i = 0; i < 25; ++i)
+            s[i] = 0;
+        for (int blk = 0; blk < nblocks; ++blk) {
+#pragma unroll
+            for (int i = 0; i < 17; ++i)
+                s[i] ^= slot64[blk * 17 + i];
+            keccak_f(s);
+        }
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+            br_hash[i] = s[i];
+    }
     uint8_t br_ref[33], br_ref_len;
-    make_ref(slot, br_len, br_hash, br_ref, &br_ref_len);
+    // make_ref needs the raw rlp only for the inline (<32 B) case
+    if (mt.br_len < 32) {
+        br_ref_len = (uint8_t)mt.br_len;
+        const uint8_t *sl = scratch + (uint64_t)g * SLOT_BR;
+        for (int k = 0; k < br_ref_len; ++k)
+            br_ref[k] = sl[k];
+    } else {
+        br_ref_len = 33;
+        br_ref[0] = 0xa0;
+        memcpy(br_ref + 1, br_hash, 32);
+    }
+    int kblocks = nblocks;
+    const uint8_t *key0 = keys + (uint64_t)mt.s * key_stride;
+    uint8_t *ext = lds + (uint64_t)threadIdx.x * SLOT_EXT;
+    uint64_t *ext64 = (uint64_t *)ext;
 
-    int8_t pl = lcp[s_first], pr = lcp[e_last];
-    int P = pl > pr ? pl : pr;
-
-    // extension wrap over key0[from..d); produces hash (keccak of outermost
-    // node rlp) + embedded ref. Reuses the slot.
+    // extension wrap over key0[from..d) in the LDS slot (<= 72 B)
     auto wrap = [&](int from, uint64_t hash[4], uint8_t *ref, uint8_t *ref_len) {
         if (d == from) {
 #pragma unroll
@@ -906,45 +964,64 @@ __global__ void __launch_bounds__(BLOCK) k_branch(
             *ref_len = br_ref_len;
             return;
         }
-#pragma unroll 8
-        for (int k = 0; k < SLOT_BR / 8; ++k)
-            slot64[k] = 0;
+#pragma unroll
+        for (int k = 0; k < SLOT_EXT / 8; ++k)
+            ext64[k] = 0;
         int pay = hp_item_len(from, d) + br_ref_len;
-        int hh = rlp_list_hdr_write(slot, pay);
-        int p = hh + hp_item_write(slot + hh, key0, from, d, 0);
+        int hh = rlp_list_hdr_write(ext, pay);
+        int p = hh + hp_item_write(ext + hh, key0, from, d, 0);
         for (int k = 0; k < br_ref_len; ++k)
-            slot[p++] = br_ref[k];
-        int len = hh + pay;
-        int nb2 = keccak_pad(slot, len);
-        keccak_lds(slot64, nb2, hash);
-        make_ref(slot, len, hash, ref, ref_len);
+            ext[p++] = br_ref[k];
+        int len = hh + pay; // <= 69 < SLOT_EXT
+        ext[len] = 0x01;    // pad start; end bit lands in lane 16 below
+        // single 136-B keccak block: absorb SLOT_EXT bytes + implicit zeros
+        uint64_t s[25];
+#pragma unroll
+        for (int i = 0; i < 25; ++i)
+            s[i] = 0;
+#pragma unroll
+        for (int i = 0; i < SLOT_EXT / 8; ++i)
+            s[i] ^= ext64[i];
+        s[16] ^= 0x8000000000000000ULL; // pad end bit of the 136-B block
+        keccak_f(s);
+#pragma unroll
+        for (int k = 0; k < 4; ++k)
+            hash[k] = s[k];
+        if (len < 32) {
+            *ref_len = (uint8_t)len;
+            for (int k = 0; k < len; ++k)
+                ref[k] = ext[k];
+        } else {
+            *ref_len = 33;
+            ref[0] = 0xa0;
+            memcpy(ref + 1, hash, 32);
+        }
+        kblocks += 1;
     };
 
-    node_rec r;
-    r.s = s_first;
-    r.e = e_last;
-    r.seg = seg;
-    r.pad_ = 0;
-    if (P >= 0) {
-        r.depth = (int8_t)P;
+    if (mt.P >= 0) {
+        r->depth = mt.P;
         uint64_t hash[4];
-        wrap(P + 1, hash, r.ref, &r.ref_len);
-        atomicAdd(&pending[P + 1], 1u);
+        uint8_t rl;
+        wrap(mt.P + 1, hash, r->ref, &rl);
+        r->ref_len = rl;
+        atomicAdd(&pending[mt.P + 1], 1u);
     } else {
-        r.depth = -1;
-        r.ref_len = 0; // dead record: root emitted below
+        r->depth = -1;
+        r->ref_len = 0;
         uint64_t hash[4];
-        uint8_t ref[33], ref_len;
+        uint8_t ref_len;
         if (subtree) {
-            wrap(1, hash, ref, &ref_len); // child-of-root-branch form
-            for (int k = 0; k < ref_len; ++k)
-                child_refs[33ull * seg + k] = ref[k];
-            child_lens[seg] = ref_len;
+            wrap(1, hash, child_refs + 33ull * mt.seg, &ref_len);
+            child_lens[mt.seg] = ref_len;
         }
-        wrap(0, hash, ref, &ref_len); // standalone form; root = keccak(rlp)
-        memcpy(seg_roots + 32ull * seg, hash, 32);
+        // standalone form: only the hash matters; write the ref into the
+        // (dead) record's ref bytes to keep everything in global memory
+        wrap(0, hash, r->ref, &ref_len);
+        memcpy(seg_roots + 32ull * mt.seg, hash, 32);
+        r->ref_len = 0;
     }
-    copy_rec(&out[gidx[j]], &r);
+    atomicAdd(&pending[65], (uint32_t)kblocks);
 }
 
 // ---------------------------------------------------------------------------
@@ -1082,7 +1159,43 @@ struct sre_ctx {
     uint64_t ns = 0;
     bool own_st = false;
     sre_stats stats{};
+    // size-class buffer pool: the level machinery allocates/frees dozens of
+    // transient arrays per level; hipMalloc latency would dominate small
+    // jobs. Freed buffers are cached by power-of-2 class and reused (also
+    // across bench steps). Freed for real in sre_destroy.
+    std::vector<std::pair<size_t, void *>> pool;
 };
+
+static size_t pool_class(size_t bytes)
+{
+    size_t c = 256;
+    while (c < bytes)
+        c <<= 1;
+    return c;
+}
+
+static void *pool_get(sre_ctx *ctx, size_t bytes)
+{
+    size_t cls = pool_class(bytes);
+    for (size_t i = 0; i < ctx->pool.size(); ++i) {
+        if (ctx->pool[i].first == cls) {
+            void *p = ctx->pool[i].second;
+            ctx->pool[i] = ctx->pool.back();
+            ctx->pool.pop_back();
+            return p;
+        }
+    }
+    void *p = nullptr;
+    if (hipMalloc(&p, cls) != hipSuccess)
+        return nullptr;
+    return p;
+}
+
+static void pool_put(sre_ctx *ctx, size_t bytes, void *p)
+{
+    if (p)
+        ctx->pool.emplace_back(pool_class(bytes), p);
+}
 
 static std::string g_err;
 
@@ -1126,6 +1239,9 @@ extern "C" void sre_destroy(sre_ctx *ctx)
 {
     if (!ctx)
         return;
+    for (auto &e : ctx->pool)
+        (void)hipFree(e.second);
+    ctx->pool.clear();
     if (ctx->own_acct && ctx->d_acct)
         hipFree((void *)ctx->d_acct);
     if (ctx->own_st && ctx->d_st)
@@ -1213,23 +1329,43 @@ extern "C" int sre_keccak_batch_device(sre_ctx *ctx, const void *d_in, uint64_t 
     return 0;
 }
 
+// Pool-backed transient device buffer (stream-ordered reuse is safe: all
+// work runs on ctx->stream).
 struct DBuf {
+    sre_ctx *ctx = nullptr;
     void *p = nullptr;
-    ~DBuf()
+    size_t sz = 0;
+    DBuf() = default;
+    explicit DBuf(sre_ctx *c) : ctx(c) {}
+    ~DBuf() { release(); }
+    void release()
     {
-        if (p)
-            hipFree(p);
+        if (p) {
+            if (ctx)
+                pool_put(ctx, sz, p);
+            else
+                (void)hipFree(p);
+            p = nullptr;
+        }
     }
     hipError_t alloc(size_t bytes)
     {
-        if (p) {
-            hipFree(p);
-            p = nullptr;
+        release();
+        sz = bytes ? bytes : 16;
+        if (ctx) {
+            p = pool_get(ctx, sz);
+            return p ? hipSuccess : hipErrorOutOfMemory;
         }
-        return hipMalloc(&p, bytes ? bytes : 16);
+        return hipMalloc(&p, sz);
     }
     template <typename T> T *as() { return (T *)p; }
 };
+
+static void swap_bufs(DBuf &a, DBuf &b)
+{
+    std::swap(a.p, b.p);
+    std::swap(a.sz, b.sz);
+}
 
 static inline uint32_t grid_for(uint64_t n)
 {
@@ -1253,7 +1389,7 @@ static int scan_u32(sre_ctx *ctx, const uint32_t *d_in, uint32_t *d_out, uint64_
     }
     uint64_t per_block = (uint64_t)BLOCK * SCAN_ITEMS;
     uint64_t nblocks = (n + per_block - 1) / per_block;
-    DBuf sums, sums_scanned;
+    DBuf sums(ctx), sums_scanned(ctx);
     HIP_CHECK(ctx, sums.alloc(nblocks * 4));
     hipLaunchKernelGGL(k_scan_block, dim3((uint32_t)nblocks), dim3(BLOCK), 0,
                        ctx->stream, d_in, n, d_out, sums.as<uint32_t>());
@@ -1301,8 +1437,10 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
     if (maxd < 0)
         return 0; // every leaf was already a segment root
 
-    DBuf carry, carry2, Lsel, Csel, Crest, Lbuf, newn, dead, flags, gidx, pend;
-    DBuf blk_a, blk_b, off_a, off_b;
+    DBuf carry(ctx), carry2(ctx), Lsel(ctx), Csel(ctx), Crest(ctx), Lbuf(ctx),
+        newn(ctx), dead(ctx), flags(ctx), gidx(ctx), pend(ctx);
+    DBuf gs(ctx), scratch(ctx), meta(ctx);
+    DBuf blk_a(ctx), blk_b(ctx), off_a(ctx), off_b(ctx);
     uint64_t carry_n = 0;
     uint64_t max_blocks = sel_grid_for(n) + 2;
     HIP_CHECK(ctx, blk_a.alloc(max_blocks * 4));
@@ -1394,12 +1532,27 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         if (scan_u32(ctx, flags.as<uint32_t>(), gidx.as<uint32_t>(), n_level,
                      &n_groups))
             return -1;
-        // 5. branch kernel
+        // 5. branch pipeline: group starts -> assemble (global scratch,
+        // full occupancy) -> hash (small LDS for ext wraps)
         HIP_CHECK(ctx, newn.alloc((uint64_t)n_groups * sizeof(node_rec)));
+        HIP_CHECK(ctx, gs.alloc(((uint64_t)n_groups + 1) * 4));
+        HIP_CHECK(ctx, scratch.alloc((uint64_t)n_groups * SLOT_BR));
+        HIP_CHECK(ctx, meta.alloc((uint64_t)n_groups * sizeof(br_meta)));
+        hipLaunchKernelGGL(k_group_starts, dim3(grid_for(n_level)), dim3(BLOCK), 0,
+                           ctx->stream, flags.as<uint32_t>(), gidx.as<uint32_t>(),
+                           n_level, gs.as<uint32_t>());
+        uint32_t n_level32 = (uint32_t)n_level;
+        HIP_CHECK(ctx, hipMemcpyAsync(gs.as<uint32_t>() + n_groups, &n_level32, 4,
+                                      hipMemcpyHostToDevice, ctx->stream));
         hipEventRecord(ev0, ctx->stream);
-        hipLaunchKernelGGL(k_branch, dim3(grid_for(n_level)), dim3(BLOCK), 0,
-                           ctx->stream, L, n_level, flags.as<uint32_t>(),
-                           gidx.as<uint32_t>(), d_lcp, d_keys, key_stride, d, subtree,
+        hipLaunchKernelGGL(k_branch_assemble, dim3(grid_for(n_groups)), dim3(BLOCK),
+                           0, ctx->stream, L, gs.as<uint32_t>(), n_groups,
+                           d_lcp, d_keys, key_stride, d, scratch.as<uint8_t>(),
+                           meta.as<br_meta>(), d_err);
+        HIP_CHECK(ctx, hipGetLastError());
+        hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(n_groups)), dim3(BLOCK), 0,
+                           ctx->stream, scratch.as<uint8_t>(), meta.as<br_meta>(),
+                           n_groups, d_keys, key_stride, subtree,
                            newn.as<node_rec>(), d_seg_roots, d_child_refs,
                            d_child_lens, pend.as<uint32_t>(), d_err);
         HIP_CHECK(ctx, hipGetLastError());
@@ -1420,10 +1573,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         uint64_t n_alive = 0;
         {
             uint64_t alive_acc = 0;
-            for (int k = 0; k < 66; ++k)
+            for (int k = 0; k < 65; ++k) // slot 65 = keccak-block stats
                 alive_acc += pending_host[k];
             n_alive = alive_acc - (carry_n - nB); // new nodes still pending
         }
+        po->branch_blocks = pending_host[65];
         // new carry = merge(Crest, alive(newn))
         uint64_t new_carry_n = n_rest + n_alive;
         if (new_carry_n == 0) {
@@ -1431,7 +1585,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         } else {
             HIP_CHECK(ctx, carry2.alloc(new_carry_n * sizeof(node_rec)));
             // compact alive out of newn (dead records have depth == -1)
-            DBuf alive;
+            DBuf alive(ctx);
             HIP_CHECK(ctx, alive.alloc((n_alive ? n_alive : 1) * sizeof(node_rec)));
             if (n_groups) {
                 HIP_CHECK(ctx, dead.alloc(((uint64_t)n_groups) * sizeof(node_rec)));
@@ -1466,7 +1620,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                carry2.as<node_rec>());
             HIP_CHECK(ctx, hipGetLastError());
             HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
-            std::swap(carry.p, carry2.p);
+            swap_bufs(carry, carry2);
             carry_n = new_carry_n;
         }
     }
@@ -1493,7 +1647,7 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
     if (ns == 0)
         return 0;
 
-    DBuf flags, seg_id, lcp, recs, depths, hist;
+    DBuf flags(ctx), seg_id(ctx), lcp(ctx), recs(ctx), depths(ctx), hist(ctx);
     HIP_CHECK(ctx, flags.alloc(ns * 4));
     HIP_CHECK(ctx, seg_id.alloc(ns * 4));
     HIP_CHECK(ctx, lcp.alloc(ns + 1));
@@ -1513,7 +1667,7 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
                        flags.as<uint32_t>(), seg_id.as<uint32_t>(), ns);
     HIP_CHECK(ctx, hipGetLastError());
 
-    DBuf seg_start, seg_acct, seg_roots;
+    DBuf seg_start(ctx), seg_acct(ctx), seg_roots(ctx);
     HIP_CHECK(ctx, seg_start.alloc((uint64_t)n_seg * 4));
     HIP_CHECK(ctx, seg_acct.alloc((uint64_t)n_seg * 4));
     HIP_CHECK(ctx, seg_roots.alloc((uint64_t)n_seg * 32));
@@ -1575,7 +1729,7 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
                             uint8_t *d_child_lens, pass_out *po, uint32_t *d_err)
 {
     uint64_t na = ctx->na;
-    DBuf lcp, recs, depths, hist;
+    DBuf lcp(ctx), recs(ctx), depths(ctx), hist(ctx);
     HIP_CHECK(ctx, lcp.alloc(na + 1));
     HIP_CHECK(ctx, recs.alloc(na * sizeof(node_rec)));
     HIP_CHECK(ctx, depths.alloc(na));
@@ -1662,7 +1816,7 @@ extern "C" int sre_root(sre_ctx *ctx, uint8_t out_root[32])
     hipEventCreate(&t1);
     hipEventRecord(t0, ctx->stream);
 
-    DBuf err, acct_roots, root;
+    DBuf err(ctx), acct_roots(ctx), root(ctx);
     HIP_CHECK(ctx, err.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
     HIP_CHECK(ctx, acct_roots.alloc(ctx->na * 32));
@@ -1705,7 +1859,7 @@ extern "C" int sre_storage_roots(sre_ctx *ctx, uint8_t *out, uint64_t n)
     }
     if (n == 0)
         return 0;
-    DBuf err, acct_roots;
+    DBuf err(ctx), acct_roots(ctx);
     HIP_CHECK(ctx, err.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
     HIP_CHECK(ctx, acct_roots.alloc(ctx->na * 32));
@@ -1733,7 +1887,7 @@ extern "C" int sre_subtree_roots(sre_ctx *ctx, uint8_t out_child_refs[16][33],
     hipEventCreate(&t1);
     hipEventRecord(t0, ctx->stream);
 
-    DBuf err, acct_roots, roots, crefs, clens, counts;
+    DBuf err(ctx), acct_roots(ctx), roots(ctx), crefs(ctx), clens(ctx), counts(ctx);
     HIP_CHECK(ctx, err.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
     HIP_CHECK(ctx, acct_roots.alloc(ctx->na * 32));
@@ -1799,7 +1953,7 @@ extern "C" int sre_finish_top(sre_ctx *ctx, const uint8_t child_refs[16][33],
         memcpy(out_root, root_hash[last], 32);
         return 0;
     }
-    DBuf crefs, clens, root;
+    DBuf crefs(ctx), clens(ctx), root(ctx);
     HIP_CHECK(ctx, crefs.alloc(16 * 33));
     HIP_CHECK(ctx, clens.alloc(16));
     HIP_CHECK(ctx, root.alloc(32));
