@@ -82,6 +82,27 @@ __device__ __forceinline__ uint64_t rotl64(uint64_t x, int n)
     return (x << n) | (x >> (64 - n));
 }
 
+// constant-amount 64-bit rotate as exactly two v_alignbit_b32 (the generic
+// shift form compiles to a 64-bit shift + 32-bit shift + or; measured in
+// profiles/ this is the keccak round's hottest primitive)
+__device__ __forceinline__ uint64_t rotl64c(uint64_t x, int n)
+{
+    uint32_t lo = (uint32_t)x, hi = (uint32_t)(x >> 32);
+    uint32_t nh, nl;
+    if (n == 32) {
+        nh = lo;
+        nl = hi;
+    } else if (n < 32) {
+        nh = __builtin_amdgcn_alignbit(hi, lo, 32 - n);
+        nl = __builtin_amdgcn_alignbit(lo, hi, 32 - n);
+    } else {
+        int m = n - 32;
+        nh = __builtin_amdgcn_alignbit(lo, hi, 32 - m);
+        nl = __builtin_amdgcn_alignbit(hi, lo, 32 - m);
+    }
+    return ((uint64_t)nh << 32) | nl;
+}
+
 // Keccak-f[1600]. Inner loops unrolled so every s[]/b[] index is a
 // compile-time constant (register-resident; dynamic indexing would spill to
 // scratch — cdna_hip_programming.md §5.4 rule 20).
@@ -98,7 +119,7 @@ __device__ void keccak_f(uint64_t s[25])
             c[x] = s[x] ^ s[x + 5] ^ s[x + 10] ^ s[x + 15] ^ s[x + 20];
 #pragma unroll
         for (int x = 0; x < 5; ++x)
-            d[x] = c[(x + 4) % 5] ^ rotl64(c[(x + 1) % 5], 1);
+            d[x] = c[(x + 4) % 5] ^ rotl64c(c[(x + 1) % 5], 1);
 #pragma unroll
         for (int i = 0; i < 25; ++i)
             s[i] ^= d[i % 5];
@@ -108,7 +129,7 @@ __device__ void keccak_f(uint64_t s[25])
             for (int y = 0; y < 5; ++y) {
                 const int src = x + 5 * y;
                 const int dst = y + 5 * ((2 * x + 3 * y) % 5);
-                b[dst] = ROT[src] ? rotl64(s[src], ROT[src]) : s[src];
+                b[dst] = ROT[src] ? rotl64c(s[src], ROT[src]) : s[src];
             }
         }
 #pragma unroll
@@ -1451,6 +1472,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
     HIP_CHECK(ctx, hipMemsetAsync(pend.p, 0, 66 * 4, ctx->stream));
 
     uint32_t pending_host[66] = {0};
+    uint64_t branch_blocks_base = po->branch_blocks;
     hipEvent_t ev0, ev1;
     hipEventCreate(&ev0);
     hipEventCreate(&ev1);
@@ -1533,11 +1555,15 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                      &n_groups))
             return -1;
         // 5. branch pipeline: group starts -> assemble (global scratch,
-        // full occupancy) -> hash (small LDS for ext wraps)
+        // full occupancy) -> hash (small LDS for ext wraps). Processed in
+        // chunks so the 552-B scratch slots stay bounded (a single level of
+        // the 10M x 64 build can hold >100M groups).
+        const uint64_t BR_CHUNK = 16ull << 20;
         HIP_CHECK(ctx, newn.alloc((uint64_t)n_groups * sizeof(node_rec)));
         HIP_CHECK(ctx, gs.alloc(((uint64_t)n_groups + 1) * 4));
-        HIP_CHECK(ctx, scratch.alloc((uint64_t)n_groups * SLOT_BR));
-        HIP_CHECK(ctx, meta.alloc((uint64_t)n_groups * sizeof(br_meta)));
+        uint64_t chunk = n_groups < BR_CHUNK ? n_groups : BR_CHUNK;
+        HIP_CHECK(ctx, scratch.alloc(chunk * SLOT_BR));
+        HIP_CHECK(ctx, meta.alloc(chunk * sizeof(br_meta)));
         hipLaunchKernelGGL(k_group_starts, dim3(grid_for(n_level)), dim3(BLOCK), 0,
                            ctx->stream, flags.as<uint32_t>(), gidx.as<uint32_t>(),
                            n_level, gs.as<uint32_t>());
@@ -1545,17 +1571,20 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         HIP_CHECK(ctx, hipMemcpyAsync(gs.as<uint32_t>() + n_groups, &n_level32, 4,
                                       hipMemcpyHostToDevice, ctx->stream));
         hipEventRecord(ev0, ctx->stream);
-        hipLaunchKernelGGL(k_branch_assemble, dim3(grid_for(n_groups)), dim3(BLOCK),
-                           0, ctx->stream, L, gs.as<uint32_t>(), n_groups,
-                           d_lcp, d_keys, key_stride, d, scratch.as<uint8_t>(),
-                           meta.as<br_meta>(), d_err);
-        HIP_CHECK(ctx, hipGetLastError());
-        hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(n_groups)), dim3(BLOCK), 0,
-                           ctx->stream, scratch.as<uint8_t>(), meta.as<br_meta>(),
-                           n_groups, d_keys, key_stride, subtree,
-                           newn.as<node_rec>(), d_seg_roots, d_child_refs,
-                           d_child_lens, pend.as<uint32_t>(), d_err);
-        HIP_CHECK(ctx, hipGetLastError());
+        for (uint64_t g0 = 0; g0 < n_groups; g0 += chunk) {
+            uint32_t gc = (uint32_t)(n_groups - g0 < chunk ? n_groups - g0 : chunk);
+            hipLaunchKernelGGL(k_branch_assemble, dim3(grid_for(gc)), dim3(BLOCK),
+                               0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
+                               d_lcp, d_keys, key_stride, d, scratch.as<uint8_t>(),
+                               meta.as<br_meta>(), d_err);
+            HIP_CHECK(ctx, hipGetLastError());
+            hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(gc)), dim3(BLOCK), 0,
+                               ctx->stream, scratch.as<uint8_t>(), meta.as<br_meta>(),
+                               gc, d_keys, key_stride, subtree,
+                               newn.as<node_rec>() + g0, d_seg_roots, d_child_refs,
+                               d_child_lens, pend.as<uint32_t>(), d_err);
+            HIP_CHECK(ctx, hipGetLastError());
+        }
         hipEventRecord(ev1, ctx->stream);
         HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
         float ms = 0;
@@ -1577,7 +1606,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                 alive_acc += pending_host[k];
             n_alive = alive_acc - (carry_n - nB); // new nodes still pending
         }
-        po->branch_blocks = pending_host[65];
+        po->branch_blocks = branch_blocks_base + pending_host[65];
         // new carry = merge(Crest, alive(newn))
         uint64_t new_carry_n = n_rest + n_alive;
         if (new_carry_n == 0) {
