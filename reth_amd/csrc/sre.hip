@@ -923,21 +923,32 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
     uint32_t *__restrict__ pending, uint32_t *__restrict__ err)
 {
-    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_EXT];
+    // pending[] updates are LDS-aggregated: one global atomic per counter
+    // per block instead of per group (a single hot counter word saturates at
+    // ~88 atomics/us chip-wide and was the dominant cost of this kernel).
+    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_EXT + 66 * 4];
+    uint32_t *hist_l = (uint32_t *)(lds + BLOCK * SLOT_EXT);
+    if (threadIdx.x < 66)
+        hist_l[threadIdx.x] = 0;
+    __syncthreads();
     uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
-    if (g >= n_groups)
-        return;
-    br_meta mt = meta[g];
-    node_rec *r = &out[g]; // write the record in place (no scratch spill)
-    r->s = mt.s;
-    r->e = mt.e;
-    r->seg = mt.seg;
-    r->pad_ = 0;
-    if (mt.br_len == 0) { // error group: dead record (err already flagged)
-        r->depth = -1;
-        r->ref_len = 0;
-        return;
+    bool active = g < n_groups;
+    br_meta mt{};
+    node_rec *r = nullptr;
+    if (active) {
+        mt = meta[g];
+        r = &out[g]; // write the record in place (no scratch spill)
+        r->s = mt.s;
+        r->e = mt.e;
+        r->seg = mt.seg;
+        r->pad_ = 0;
+        if (mt.br_len == 0) { // error group: dead record (err already flagged)
+            r->depth = -1;
+            r->ref_len = 0;
+            active = false;
+        }
     }
+    if (active) {
     int d = mt.d;
     const uint64_t *slot64 = (const uint64_t *)(scratch + (uint64_t)g * SLOT_BR);
     int nblocks = mt.br_len / 136 + 1;
@@ -1026,7 +1037,7 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         uint8_t rl;
         wrap(mt.P + 1, hash, r->ref, &rl);
         r->ref_len = rl;
-        atomicAdd(&pending[mt.P + 1], 1u);
+        atomicAdd(&hist_l[mt.P + 1], 1u);
     } else {
         r->depth = -1;
         r->ref_len = 0;
@@ -1042,7 +1053,11 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         memcpy(seg_roots + 32ull * mt.seg, hash, 32);
         r->ref_len = 0;
     }
-    atomicAdd(&pending[65], (uint32_t)kblocks);
+    atomicAdd(&hist_l[65], (uint32_t)kblocks);
+    } // active
+    __syncthreads();
+    if (threadIdx.x < 66 && hist_l[threadIdx.x])
+        atomicAdd(&pending[threadIdx.x], hist_l[threadIdx.x]);
 }
 
 // ---------------------------------------------------------------------------
