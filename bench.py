@@ -123,6 +123,8 @@ def main():
             "subtree composition mismatch"
 
     if rank != 0:
+        if dist:
+            dist.destroy_process_group()
         return
 
     # ---- roofline of the dominant kernel (leaf RLP+keccak), HIP-event timed
@@ -205,6 +207,8 @@ def main():
         "gen_seconds": round(t_gen, 1),
     }
     print(json.dumps(out))
+    if dist:
+        dist.destroy_process_group()
 
 
 if __name__ == "__main__":

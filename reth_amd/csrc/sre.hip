@@ -1090,10 +1090,17 @@ __global__ void k_fill_empty_roots(uint8_t *__restrict__ acct_roots, uint64_t na
 __global__ void k_nibble_counts(const sre_account_entry *__restrict__ acct,
                                 uint64_t na, uint64_t *__restrict__ counts)
 {
+    __shared__ uint32_t cnt_l[16];
+    if (threadIdx.x < 16)
+        cnt_l[threadIdx.x] = 0;
+    __syncthreads();
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= na)
-        return;
-    atomicAdd((unsigned long long *)&counts[acct[i].key[0] >> 4], 1ull);
+    if (i < na)
+        atomicAdd(&cnt_l[acct[i].key[0] >> 4], 1u);
+    __syncthreads();
+    if (threadIdx.x < 16 && cnt_l[threadIdx.x])
+        atomicAdd((unsigned long long *)&counts[threadIdx.x],
+                  (unsigned long long)cnt_l[threadIdx.x]);
 }
 
 // root-branch assembly from gathered child refs (single thread; O(1) work)
