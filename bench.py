@@ -10,9 +10,9 @@ Single GPU (default): the largest single-GPU configuration of the metric —
 Multi GPU (launched by the driver via torch.distributed.run, one rank per
 GPU over RCCL): accounts sharded by hashed-key top nibble (nibble % world ==
 rank); per-rank sre_subtree_roots, one all-gather of the 16 subtrie digests,
-root finished on every rank (reth_amd/sharding.py). scaling="weak" in the
-sense that per-GPU work shrinks as ranks grow while the TOTAL job (one root
-over the same 10M x 64 state) is fixed; the driver computes efficiency.
+root finished on every rank (reth_amd/sharding.py). scaling="strong": the
+TOTAL job (one root over the same 10M x 64 state) is fixed as ranks grow;
+the driver computes efficiency from the per-N values.
 
 CPU baseline: the C oracle (reth-algorithm restatement, single thread) timed
 on a bounded sample of the same workload on this box's host cores, scaled
