@@ -31,6 +31,23 @@ def main(path):
     print(f"\n# kernel-busy {span[0]:.1f} ms over wall span {span[1]:.1f} ms "
           f"(includes untimed generation + warmup + all steps)")
 
+    # PMC counters (from --pmc passes): aggregate per kernel per counter
+    pmc = next((t for t in tables if t.startswith("rocpd_pmc_event")), None)
+    info = next((t for t in tables if t.startswith("rocpd_info_pmc")), None)
+    if pmc and cur.execute(f"SELECT COUNT(*) FROM {pmc}").fetchone()[0]:
+        print("\n# PMC counters per kernel: counter sum over dispatches "
+              "(and per-dispatch average)")
+        rows = list(cur.execute(f"""
+            SELECT s.display_name, i.name, SUM(p.value), COUNT(*)
+            FROM {pmc} p
+            JOIN {disp} d ON p.event_id = d.event_id
+            JOIN {sym} s ON d.kernel_id = s.id
+            JOIN {info} i ON p.pmc_id = i.id
+            GROUP BY s.display_name, i.name ORDER BY 3 DESC"""))
+        for name, counter, total, cnt in rows[:60]:
+            print(f"{counter:>22} {total:18.0f} over {cnt:5d} dispatches "
+                  f"(avg {total/cnt:14.1f})  {name[:60]}")
+
 
 if __name__ == "__main__":
     main(sys.argv[1])
