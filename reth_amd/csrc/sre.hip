@@ -885,9 +885,14 @@ __global__ void k_branch_assemble(
         meta[g] = mt;
         return;
     }
-#pragma unroll 4
-    for (int k = 0; k < SLOT_BR / 8; ++k)
-        slot64[k] = 0;
+    // zero only the keccak blocks this node occupies (the hash kernel reads
+    // exactly nb*17 u64s) — most branches fit one 136-B block
+    {
+        int hdr = rlp_list_hdr_len(payload);
+        int nb = (hdr + payload) / 136 + 1;
+        for (int k = 0; k < nb * 17; ++k)
+            slot64[k] = 0;
+    }
     int h = rlp_list_hdr_write(slot, payload);
     // pass 2: write children in nibble order
     {
