@@ -843,16 +843,19 @@ __global__ void k_group_starts(const uint32_t *__restrict__ flags,
         gs[gidx[j]] = (uint32_t)j;
 }
 
-// Assemble one branch node per lane into a zero-padded, keccak-padded
-// SLOT_BR-byte global scratch slot. No LDS -> full occupancy hides the
-// scattered child-ref gathers.
-__global__ void k_branch_assemble(
+// Assemble one branch node per lane. RLP bytes are staged in a per-lane LDS
+// slot (byte stores to LDS are cheap; divergent byte stores to global were
+// address-throughput bound) and copied out as u64s; child refs are read as
+// aligned u32 words of the 48-byte records instead of per-byte.
+#define BLOCK_A 128u
+__global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     const node_rec *__restrict__ L, const uint32_t *__restrict__ gs,
     uint32_t n_groups, const int8_t *__restrict__ lcp,
     const uint8_t *__restrict__ keys, uint64_t key_stride, int d,
     uint8_t *__restrict__ scratch, br_meta *__restrict__ meta,
     uint32_t *__restrict__ err)
 {
+    __shared__ __align__(16) uint8_t lds[BLOCK_A * SLOT_BR];
     uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
     if (g >= n_groups)
         return;
@@ -865,13 +868,11 @@ __global__ void k_branch_assemble(
     int8_t pl = lcp[mt.s], pr = lcp[mt.e];
     mt.P = pl > pr ? pl : pr;
 
-    uint8_t *slot = scratch + (uint64_t)g * SLOT_BR;
+    uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_BR;
     uint64_t *slot64 = (uint64_t *)slot;
 
     int nmem = (int)(jend - j);
-    // validate ascending-nibble member order and compute the payload:
-    // absent nibbles contribute one 0x80 byte, present ones their ref,
-    // plus the trailing empty value item.
+    // validate ascending-nibble member order and compute the payload
     uint64_t m = j;
     for (int b = 0; b < 16 && m < jend; ++b)
         if (nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b)
@@ -885,25 +886,29 @@ __global__ void k_branch_assemble(
         meta[g] = mt;
         return;
     }
-    // zero only the keccak blocks this node occupies (the hash kernel reads
-    // exactly nb*17 u64s) — most branches fit one 136-B block
-    {
-        int hdr = rlp_list_hdr_len(payload);
-        int nb = (hdr + payload) / 136 + 1;
-        for (int k = 0; k < nb * 17; ++k)
-            slot64[k] = 0;
-    }
+    int hdr = rlp_list_hdr_len(payload);
+    int nb = (hdr + payload) / 136 + 1;
+    for (int k = 0; k < nb * 17; ++k)
+        slot64[k] = 0;
     int h = rlp_list_hdr_write(slot, payload);
-    // pass 2: write children in nibble order
+    // write children in nibble order; refs read as 9 aligned u32s per record
     {
         int p = h;
         m = j;
         for (int b = 0; b < 16; ++b) {
             if (m < jend && nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b) {
                 int rl = L[m].ref_len;
-                const uint8_t *ref = L[m].ref;
-                for (int k = 0; k < rl; ++k)
-                    slot[p + k] = ref[k];
+                const uint32_t *rec32 = (const uint32_t *)&L[m];
+                uint32_t w[9];
+#pragma unroll
+                for (int k = 0; k < 9; ++k)
+                    w[k] = rec32[3 + k]; // bytes 12..48 of the record
+                // ref byte i lives at record byte 14+i
+#pragma unroll
+                for (int i = 0; i < 33; ++i)
+                    if (i < rl)
+                        slot[p + i] = (uint8_t)(w[((14 + i) >> 2) - 3]
+                                                >> (8 * ((14 + i) & 3)));
                 p += rl;
                 m++;
             } else {
@@ -914,6 +919,10 @@ __global__ void k_branch_assemble(
     }
     int br_len = h + payload;
     keccak_pad(slot, br_len);
+    // copy the occupied blocks to global scratch as u64s
+    uint64_t *out64 = (uint64_t *)(scratch + (uint64_t)g * SLOT_BR);
+    for (int k = 0; k < nb * 17; ++k)
+        out64[k] = slot64[k];
     mt.br_len = (uint16_t)br_len;
     meta[g] = mt;
 }
@@ -1600,7 +1609,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         hipEventRecord(ev0, ctx->stream);
         for (uint64_t g0 = 0; g0 < n_groups; g0 += chunk) {
             uint32_t gc = (uint32_t)(n_groups - g0 < chunk ? n_groups - g0 : chunk);
-            hipLaunchKernelGGL(k_branch_assemble, dim3(grid_for(gc)), dim3(BLOCK),
+            hipLaunchKernelGGL(k_branch_assemble,
+                               dim3((gc + BLOCK_A - 1) / BLOCK_A), dim3(BLOCK_A),
                                0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
                                d_lcp, d_keys, key_stride, d, scratch.as<uint8_t>(),
                                meta.as<br_meta>(), d_err);
