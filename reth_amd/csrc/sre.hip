@@ -277,7 +277,7 @@ __device__ __forceinline__ int min_be_len(const uint8_t *v)
 }
 
 // active trie node record, 48 bytes
-struct node_rec {
+struct __attribute__((aligned(16))) node_rec {
     uint32_t s, e;    // covered leaf interval [s, e)
     uint32_t seg;     // segment id (storage: account segment; account: 0/top nibble)
     int8_t depth;     // parent branch depth this node waits for (-1 = root)
@@ -287,13 +287,15 @@ struct node_rec {
 };
 static_assert(sizeof(node_rec) == 48, "node_rec must be 48 bytes");
 
+// 3 x dwordx4: records are 16-B aligned (48-B stride in 256-B-aligned
+// allocations); scalar u32 copies were address-throughput bound.
 __device__ __forceinline__ void copy_rec(node_rec *dst, const node_rec *src)
 {
-    const uint32_t *s32 = (const uint32_t *)src;
-    uint32_t *d32 = (uint32_t *)dst;
+    const int4 *s4 = (const int4 *)src;
+    int4 *d4 = (int4 *)dst;
 #pragma unroll
-    for (int i = 0; i < 12; ++i)
-        d32[i] = s32[i];
+    for (int i = 0; i < 3; ++i)
+        d4[i] = s4[i];
 }
 
 // ref from an LDS-resident rlp + its precomputed hash
@@ -852,8 +854,8 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     const node_rec *__restrict__ L, const uint32_t *__restrict__ gs,
     uint32_t n_groups, const int8_t *__restrict__ lcp,
     const uint8_t *__restrict__ keys, uint64_t key_stride, int d,
-    uint8_t *__restrict__ scratch, br_meta *__restrict__ meta,
-    uint32_t *__restrict__ err)
+    uint8_t *__restrict__ scratch, uint64_t scratch_stride,
+    br_meta *__restrict__ meta, uint32_t *__restrict__ err)
 {
     __shared__ __align__(16) uint8_t lds[BLOCK_A * SLOT_BR];
     uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
@@ -919,10 +921,12 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     }
     int br_len = h + payload;
     keccak_pad(slot, br_len);
-    // copy the occupied blocks to global scratch as u64s
-    uint64_t *out64 = (uint64_t *)(scratch + (uint64_t)g * SLOT_BR);
+    // copy the occupied blocks to global scratch, COLUMN-MAJOR: u64 word k
+    // of group g lives at scratch64[k*stride + g], so consecutive lanes
+    // store (and the hash kernel loads) consecutive addresses.
+    uint64_t *out64 = (uint64_t *)scratch;
     for (int k = 0; k < nb * 17; ++k)
-        out64[k] = slot64[k];
+        out64[(uint64_t)k * scratch_stride + g] = slot64[k];
     mt.br_len = (uint16_t)br_len;
     meta[g] = mt;
 }
@@ -931,7 +935,8 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
 // then do extension/root wraps in a small LDS slot (88 B -> high occupancy).
 #define SLOT_EXT 88
 __global__ void __launch_bounds__(BLOCK) k_branch_hash(
-    const uint8_t *__restrict__ scratch, const br_meta *__restrict__ meta,
+    const uint8_t *__restrict__ scratch, uint64_t scratch_stride,
+    const br_meta *__restrict__ meta,
     uint32_t n_groups, const uint8_t *__restrict__ keys, uint64_t key_stride,
     int subtree, node_rec *__restrict__ out, uint8_t *__restrict__ seg_roots,
     uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
@@ -964,7 +969,7 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     }
     if (active) {
     int d = mt.d;
-    const uint64_t *slot64 = (const uint64_t *)(scratch + (uint64_t)g * SLOT_BR);
+    const uint64_t *scr64 = (const uint64_t *)scratch; // column-major
     int nblocks = mt.br_len / 136 + 1;
     uint64_t br_hash[4];
     {
@@ -975,7 +980,7 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         for (int blk = 0; blk < nblocks; ++blk) {
 #pragma unroll
             for (int i = 0; i < 17; ++i)
-                s[i] ^= slot64[blk * 17 + i];
+                s[i] ^= scr64[(uint64_t)(blk * 17 + i) * scratch_stride + g];
             keccak_f(s);
         }
 #pragma unroll
@@ -986,9 +991,11 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     // make_ref needs the raw rlp only for the inline (<32 B) case
     if (mt.br_len < 32) {
         br_ref_len = (uint8_t)mt.br_len;
-        const uint8_t *sl = scratch + (uint64_t)g * SLOT_BR;
-        for (int k = 0; k < br_ref_len; ++k)
-            br_ref[k] = sl[k];
+#pragma unroll
+        for (int k = 0; k < 31; ++k)
+            if (k < br_ref_len)
+                br_ref[k] = (uint8_t)(scr64[(uint64_t)(k >> 3) * scratch_stride + g]
+                                      >> (8 * (k & 7)));
     } else {
         br_ref_len = 33;
         br_ref[0] = 0xa0;
@@ -1613,10 +1620,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                dim3((gc + BLOCK_A - 1) / BLOCK_A), dim3(BLOCK_A),
                                0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
                                d_lcp, d_keys, key_stride, d, scratch.as<uint8_t>(),
-                               meta.as<br_meta>(), d_err);
+                               chunk, meta.as<br_meta>(), d_err);
             HIP_CHECK(ctx, hipGetLastError());
             hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(gc)), dim3(BLOCK), 0,
-                               ctx->stream, scratch.as<uint8_t>(), meta.as<br_meta>(),
+                               ctx->stream, scratch.as<uint8_t>(), chunk,
+                               meta.as<br_meta>(),
                                gc, d_keys, key_stride, subtree,
                                newn.as<node_rec>() + g0, d_seg_roots, d_child_refs,
                                d_child_lens, pend.as<uint32_t>(), d_err);
