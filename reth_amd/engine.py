@@ -165,7 +165,15 @@ class StateRootEngine:
 
     def keccak_batch_device(self, in_tensor, msg_len, out_tensor):
         """Hash n messages of msg_len bytes at a fixed stride (device memory).
-        in_tensor: (n, stride) uint8 cuda tensor; out_tensor: (n, 32) uint8."""
+        in_tensor: (n, stride) uint8 cuda tensor; out_tensor: (n, 32) uint8.
+
+        The kernel runs on the engine's own HIP stream: drain torch's stream
+        first so pending writes to in_tensor are visible (the engine
+        synchronizes its stream before returning, so later torch ops are
+        safe the other way around)."""
+        import torch
+        if in_tensor.is_cuda:
+            torch.cuda.synchronize(in_tensor.device)
         n, stride = in_tensor.shape
         self._check(self._lib.sre_keccak_batch_device(
             ctypes.c_void_p(self._ctx), ctypes.c_void_p(in_tensor.data_ptr()),
