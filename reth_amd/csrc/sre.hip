@@ -875,15 +875,23 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     uint64_t *slot64 = (uint64_t *)slot;
 
     int nmem = (int)(jend - j);
-    // validate ascending-nibble member order and compute the payload
-    uint64_t m = j;
-    for (int b = 0; b < 16 && m < jend; ++b)
-        if (nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b)
-            m++;
+    // ONE pass over the members: cache each member's child nibble (4 bits
+    // into a u64), validate strictly-ascending nibbles (implies distinct,
+    // in-order, and <= 16 of them), and accumulate the payload.
+    uint64_t nibs = 0;
     int payload = 1 + (16 - nmem);
-    for (uint64_t k = j; k < jend; ++k)
-        payload += L[k].ref_len;
-    if (nmem < 2 || nmem > 16 || m != jend || payload > 529) {
+    bool order_ok = nmem >= 2 && nmem <= 16;
+    {
+        int prev = -1;
+        for (uint64_t mm = j; mm < jend && order_ok; ++mm) {
+            int nbm = nib_of(keys + (uint64_t)L[mm].s * key_stride, d);
+            order_ok &= nbm > prev;
+            prev = nbm;
+            nibs |= (uint64_t)nbm << (4 * (mm - j));
+            payload += L[mm].ref_len;
+        }
+    }
+    if (!order_ok || payload > 529) {
         atomicOr(err, 1u << E_INTERNAL);
         mt.br_len = 0;
         meta[g] = mt;
@@ -897,9 +905,9 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     // write children in nibble order; refs read as 9 aligned u32s per record
     {
         int p = h;
-        m = j;
+        uint64_t m = j;
         for (int b = 0; b < 16; ++b) {
-            if (m < jend && nib_of(keys + (uint64_t)L[m].s * key_stride, d) == b) {
+            if (m < jend && (int)((nibs >> (4 * (m - j))) & 0xf) == b) {
                 int rl = L[m].ref_len;
                 const uint32_t *rec32 = (const uint32_t *)&L[m];
                 uint32_t w[9];
