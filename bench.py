@@ -26,6 +26,19 @@ import sys
 import time
 
 
+def _pmc_traffic(accounts, slots):
+    """PMC-measured HBM bytes (GB) per leaf-kernel launch for this shape,
+    from the committed rocprofv3 counter runs (profiles/traffic.json);
+    None when the shape was not profiled."""
+    try:
+        with open(os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                               "profiles", "traffic.json")) as f:
+            t = json.load(f)
+        return t.get(f"{accounts}x{slots}", {}).get("leaf_traffic_gb")
+    except Exception:
+        return None
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -191,7 +204,7 @@ def main():
             "peak": 8000.0,
             "unit": "GB/s",
             "frac": round(achieved_gbps / 8000.0, 4),
-            "traffic": None,
+            "traffic": _pmc_traffic(args.accounts, args.slots),
             "note": "leaf RLP+keccak kernel; algorithmic bytes "
                     "(96 B/storage leaf, 142 B/account leaf) / HIP-event "
                     "kernel time. The kernel is integer-VALU bound, not "
