@@ -846,11 +846,56 @@ __global__ void k_group_starts(const uint32_t *__restrict__ flags,
         gs[gidx[j]] = (uint32_t)j;
 }
 
-// Assemble one branch node per lane. RLP bytes are staged in a per-lane LDS
-// slot (byte stores to LDS are cheap; divergent byte stores to global were
-// address-throughput bound) and copied out as u64s; child refs are read as
-// aligned u32 words of the 48-byte records instead of per-byte.
-#define BLOCK_A 128u
+// Assemble one branch node per lane, streaming bytes through a register
+// appender straight into the COLUMN-MAJOR u64 scratch (no LDS at all): a
+// per-lane LDS slot capped occupancy at 4 waves/CU and left the kernel 99%
+// latency-stalled (profiles/r01_2Mx64_sq_pmc.txt); the appender keeps the
+// kernel at full occupancy and every scratch store coalesced.
+#define BLOCK_A 256u
+
+struct byte_appender {
+    uint64_t cur;
+    int pos;      // bytes filled in cur (0..7)
+    int widx;     // next column-major word index
+    uint64_t *base;
+    uint64_t stride;
+    uint32_t g;
+
+    __device__ __forceinline__ void init(uint64_t *b, uint64_t s, uint32_t gg)
+    {
+        cur = 0;
+        pos = 0;
+        widx = 0;
+        base = b;
+        stride = s;
+        g = gg;
+    }
+    __device__ __forceinline__ void put(uint8_t v)
+    {
+        cur |= (uint64_t)v << (8 * pos);
+        if (++pos == 8) {
+            base[(uint64_t)widx * stride + g] = cur;
+            widx++;
+            pos = 0;
+            cur = 0;
+        }
+    }
+    // finish the message: flush, zero-fill to the last word of block nb,
+    // and set the keccak pad end bit (0x80 in the final byte).
+    __device__ __forceinline__ void finish(int nb)
+    {
+        int last = nb * 17 - 1;
+        while (widx < last) {
+            base[(uint64_t)widx * stride + g] = cur;
+            widx++;
+            cur = 0;
+            pos = 0;
+        }
+        cur |= 0x8000000000000000ULL;
+        base[(uint64_t)last * stride + g] = cur;
+    }
+};
+
 __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     const node_rec *__restrict__ L, const uint32_t *__restrict__ gs,
     uint32_t n_groups, const int8_t *__restrict__ lcp,
@@ -858,7 +903,6 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     uint8_t *__restrict__ scratch, uint64_t scratch_stride,
     br_meta *__restrict__ meta, uint32_t *__restrict__ err)
 {
-    __shared__ __align__(16) uint8_t lds[BLOCK_A * SLOT_BR];
     uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
     if (g >= n_groups)
         return;
@@ -871,13 +915,10 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     int8_t pl = lcp[mt.s], pr = lcp[mt.e];
     mt.P = pl > pr ? pl : pr;
 
-    uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_BR;
-    uint64_t *slot64 = (uint64_t *)slot;
-
     int nmem = (int)(jend - j);
     // ONE pass over the members: cache each member's child nibble (4 bits
     // into a u64), validate strictly-ascending nibbles (implies distinct,
-    // in-order, and <= 16 of them), and accumulate the payload.
+    // in-order, and <= 16 of them), accumulate the payload.
     uint64_t nibs = 0;
     int payload = 1 + (16 - nmem);
     bool order_ok = nmem >= 2 && nmem <= 16;
@@ -897,14 +938,23 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
         meta[g] = mt;
         return;
     }
-    int hdr = rlp_list_hdr_len(payload);
-    int nb = (hdr + payload) / 136 + 1;
-    for (int k = 0; k < nb * 17; ++k)
-        slot64[k] = 0;
-    int h = rlp_list_hdr_write(slot, payload);
-    // write children in nibble order; refs read as 9 aligned u32s per record
+    int h = rlp_list_hdr_len(payload);
+    int br_len = h + payload;
+    int nb = br_len / 136 + 1;
+
+    byte_appender ap;
+    ap.init((uint64_t *)scratch, scratch_stride, g);
+    if (payload < 56) {
+        ap.put((uint8_t)(0xc0 + payload));
+    } else if (payload <= 255) {
+        ap.put(0xf8);
+        ap.put((uint8_t)payload);
+    } else {
+        ap.put(0xf9);
+        ap.put((uint8_t)(payload >> 8));
+        ap.put((uint8_t)payload);
+    }
     {
-        int p = h;
         uint64_t m = j;
         for (int b = 0; b < 16; ++b) {
             if (m < jend && (int)((nibs >> (4 * (m - j))) & 0xf) == b) {
@@ -918,24 +968,17 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
 #pragma unroll
                 for (int i = 0; i < 33; ++i)
                     if (i < rl)
-                        slot[p + i] = (uint8_t)(w[((14 + i) >> 2) - 3]
-                                                >> (8 * ((14 + i) & 3)));
-                p += rl;
+                        ap.put((uint8_t)(w[((14 + i) >> 2) - 3]
+                                         >> (8 * ((14 + i) & 3))));
                 m++;
             } else {
-                slot[p++] = 0x80;
+                ap.put(0x80);
             }
         }
-        slot[p++] = 0x80;
+        ap.put(0x80); // empty value item
     }
-    int br_len = h + payload;
-    keccak_pad(slot, br_len);
-    // copy the occupied blocks to global scratch, COLUMN-MAJOR: u64 word k
-    // of group g lives at scratch64[k*stride + g], so consecutive lanes
-    // store (and the hash kernel loads) consecutive addresses.
-    uint64_t *out64 = (uint64_t *)scratch;
-    for (int k = 0; k < nb * 17; ++k)
-        out64[(uint64_t)k * scratch_stride + g] = slot64[k];
+    ap.put(0x01); // keccak pad start
+    ap.finish(nb);
     mt.br_len = (uint16_t)br_len;
     meta[g] = mt;
 }
