@@ -349,21 +349,37 @@ __global__ void __launch_bounds__(BLOCK) k_keccak_batch(
 // segmentation + lcp kernels
 // ---------------------------------------------------------------------------
 
-__global__ void k_seg_flags(const sre_storage_entry *__restrict__ st, uint64_t ns,
-                            uint32_t *__restrict__ flags, uint32_t *__restrict__ err)
+// One pass over adjacent entry pairs produces BOTH the segment-start flags
+// (acct_key change) and the slot-key lcp array (with -1 sentinels exactly at
+// the segment boundaries) — each 96-byte entry is pulled once.
+__global__ void k_seg_flags_lcp(const sre_storage_entry *__restrict__ st,
+                                uint64_t ns, uint32_t *__restrict__ flags,
+                                int8_t *__restrict__ lcp,
+                                uint32_t *__restrict__ err)
 {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= ns)
+    if (i > ns)
         return;
-    if (i == 0) {
-        flags[0] = 1;
+    if (i == 0 || i == ns) {
+        lcp[i] = -1;
+        if (i == 0 && ns)
+            flags[0] = 1;
         return;
     }
     bool gt;
-    int l = key_lcp(st[i - 1].acct_key, st[i].acct_key, &gt);
+    int la = key_lcp(st[i - 1].acct_key, st[i].acct_key, &gt);
     if (gt)
         atomicOr(err, 1u << E_UNSORTED_STORAGE);
-    flags[i] = (l < 64) ? 1u : 0u;
+    if (la < 64) { // segment boundary
+        flags[i] = 1;
+        lcp[i] = -1;
+        return;
+    }
+    flags[i] = 0;
+    int l = key_lcp(st[i - 1].slot_key, st[i].slot_key, &gt);
+    if (gt || l == 64)
+        atomicOr(err, 1u << E_UNSORTED_STORAGE);
+    lcp[i] = (int8_t)l;
 }
 
 // seg_id = inclusive_scan(flags) - 1: the host computes the exclusive scan,
@@ -412,24 +428,6 @@ __global__ void k_seg_acct(const sre_storage_entry *__restrict__ st,
             lo = mid + 1;
     }
     atomicOr(err, 1u << E_ORPHAN_STORAGE);
-}
-
-__global__ void k_lcp_storage(const sre_storage_entry *__restrict__ st, uint64_t ns,
-                              const uint32_t *__restrict__ seg_id,
-                              int8_t *__restrict__ lcp, uint32_t *__restrict__ err)
-{
-    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i > ns)
-        return;
-    if (i == 0 || i == ns || seg_id[i] != seg_id[i - 1]) {
-        lcp[i] = -1;
-        return;
-    }
-    bool gt;
-    int l = key_lcp(st[i - 1].slot_key, st[i].slot_key, &gt);
-    if (gt || l == 64)
-        atomicOr(err, 1u << E_UNSORTED_STORAGE);
-    lcp[i] = (int8_t)l;
 }
 
 __global__ void k_lcp_account(const sre_account_entry *__restrict__ acct, uint64_t na,
@@ -1782,8 +1780,9 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
     HIP_CHECK(ctx, hist.alloc(66 * 4));
     HIP_CHECK(ctx, hipMemsetAsync(hist.p, 0, 66 * 4, ctx->stream));
 
-    hipLaunchKernelGGL(k_seg_flags, dim3(grid_for(ns)), dim3(BLOCK), 0, ctx->stream,
-                       ctx->d_st, ns, flags.as<uint32_t>(), d_err);
+    hipLaunchKernelGGL(k_seg_flags_lcp, dim3(grid_for(ns + 1)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_st, ns, flags.as<uint32_t>(),
+                       lcp.as<int8_t>(), d_err);
     HIP_CHECK(ctx, hipGetLastError());
     // seg_id[i] = inclusive_scan(flags)[i] - 1 (segment index of entry i)
     uint32_t n_seg = 0;
@@ -1803,10 +1802,6 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
     hipLaunchKernelGGL(k_seg_acct, dim3(grid_for(n_seg)), dim3(BLOCK), 0, ctx->stream,
                        ctx->d_st, seg_start.as<uint32_t>(), n_seg, ctx->d_acct, na,
                        seg_acct.as<uint32_t>(), d_err);
-    hipLaunchKernelGGL(k_lcp_storage, dim3(grid_for(ns + 1)), dim3(BLOCK), 0,
-                       ctx->stream, ctx->d_st, ns, seg_id.as<uint32_t>(),
-                       lcp.as<int8_t>(), d_err);
-    HIP_CHECK(ctx, hipGetLastError());
     // input-contract violations (unsorted/orphan entries) are flagged by the
     // kernels above; bail BEFORE the trie machinery runs on garbage lcps.
     HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
