@@ -1551,11 +1551,20 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
     if (maxd < 0)
         return 0; // every leaf was already a segment root
 
-    DBuf carry(ctx), carry2(ctx), Lsel(ctx), Csel(ctx), Crest(ctx), Lbuf(ctx),
-        newn(ctx), dead(ctx), flags(ctx), gidx(ctx), pend(ctx);
+    // Per-depth carry arrays: a branch node produced with parent depth p
+    // waits in carries[p] (sorted by interval start) until level p runs.
+    // This replaces partitioning/rewriting one big carry at every level —
+    // each node now enters exactly one carry once.
+    std::vector<DBuf> carries;
+    carries.reserve(64);
+    for (int i = 0; i < 64; ++i)
+        carries.emplace_back(ctx);
+    uint64_t carry_cnt[64] = {0};
+
+    DBuf Lsel(ctx), Lbuf(ctx), newn(ctx), trash(ctx), newp(ctx), cmerge(ctx);
+    DBuf flags(ctx), gidx(ctx), pend(ctx);
     DBuf gs(ctx), scratch(ctx), meta(ctx);
     DBuf blk_a(ctx), blk_b(ctx), off_a(ctx), off_b(ctx);
-    uint64_t carry_n = 0;
     uint64_t max_blocks = sel_grid_for(n) + 2;
     HIP_CHECK(ctx, blk_a.alloc(max_blocks * 4));
     HIP_CHECK(ctx, blk_b.alloc(max_blocks * 4));
@@ -1572,12 +1581,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 
     for (int d = maxd; d >= 0; --d) {
         uint64_t nA = hist_host[d + 1];
-        uint64_t nB = pending_host[d + 1];
+        uint64_t nB = carry_cnt[d];
         if (nA == 0 && nB == 0)
             continue;
         po->levels++;
         uint64_t n_level = nA + nB;
-        uint64_t n_rest = carry_n - nB;
 
         // 1. select leaves with parent depth == d
         if (nA) {
@@ -1597,47 +1605,24 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                off_a.as<uint32_t>(), Lsel.as<node_rec>());
             HIP_CHECK(ctx, hipGetLastError());
         }
-        // 2. partition carry
-        if (nB) {
-            HIP_CHECK(ctx, Csel.alloc(nB * sizeof(node_rec)));
-            HIP_CHECK(ctx, Crest.alloc((n_rest ? n_rest : 1) * sizeof(node_rec)));
-            uint32_t g = sel_grid_for(carry_n);
-            hipLaunchKernelGGL(k_part_count, dim3(g), dim3(BLOCK), 0, ctx->stream,
-                               carry.as<node_rec>(), carry_n, d,
-                               blk_a.as<uint32_t>(), blk_b.as<uint32_t>());
-            uint32_t ts = 0, tr = 0;
-            if (scan_u32(ctx, blk_a.as<uint32_t>(), off_a.as<uint32_t>(), g, &ts))
-                return -1;
-            if (scan_u32(ctx, blk_b.as<uint32_t>(), off_b.as<uint32_t>(), g, &tr))
-                return -1;
-            if (ts != nB || tr != n_rest) {
-                set_err(ctx, "internal: pending count mismatch");
-                return -1;
-            }
-            hipLaunchKernelGGL(k_part_gather, dim3(g), dim3(BLOCK), 0, ctx->stream,
-                               carry.as<node_rec>(), carry_n, d, off_a.as<uint32_t>(),
-                               off_b.as<uint32_t>(), Csel.as<node_rec>(),
-                               Crest.as<node_rec>());
-            HIP_CHECK(ctx, hipGetLastError());
-        }
-        // 3. merge into level input L
+        // 2. merge leaves with this depth's carry into the level input L
         node_rec *L;
         if (nB == 0) {
             L = Lsel.as<node_rec>();
         } else if (nA == 0) {
-            L = Csel.as<node_rec>();
+            L = carries[d].as<node_rec>();
         } else {
             HIP_CHECK(ctx, Lbuf.alloc(n_level * sizeof(node_rec)));
             hipLaunchKernelGGL(k_merge_a, dim3(grid_for(nA)), dim3(BLOCK), 0,
                                ctx->stream, Lsel.as<node_rec>(), nA,
-                               Csel.as<node_rec>(), nB, Lbuf.as<node_rec>());
+                               carries[d].as<node_rec>(), nB, Lbuf.as<node_rec>());
             hipLaunchKernelGGL(k_merge_b, dim3(grid_for(nB)), dim3(BLOCK), 0,
                                ctx->stream, Lsel.as<node_rec>(), nA,
-                               Csel.as<node_rec>(), nB, Lbuf.as<node_rec>());
+                               carries[d].as<node_rec>(), nB, Lbuf.as<node_rec>());
             HIP_CHECK(ctx, hipGetLastError());
             L = Lbuf.as<node_rec>();
         }
-        // 4. group flags + scan
+        // 3. group flags + scan
         HIP_CHECK(ctx, flags.alloc(n_level * 4));
         HIP_CHECK(ctx, gidx.alloc(n_level * 4));
         hipLaunchKernelGGL(k_group_flags, dim3(grid_for(n_level)), dim3(BLOCK), 0,
@@ -1647,10 +1632,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         if (scan_u32(ctx, flags.as<uint32_t>(), gidx.as<uint32_t>(), n_level,
                      &n_groups))
             return -1;
-        // 5. branch pipeline: group starts -> assemble (global scratch,
-        // full occupancy) -> hash (small LDS for ext wraps). Processed in
-        // chunks so the 552-B scratch slots stay bounded (a single level of
-        // the 10M x 64 build can hold >100M groups).
+        // 4. branch pipeline: group starts -> assemble -> hash, chunked so
+        // the 552-B scratch slots stay bounded.
         const uint64_t BR_CHUNK = 16ull << 20;
         HIP_CHECK(ctx, newn.alloc((uint64_t)n_groups * sizeof(node_rec)));
         HIP_CHECK(ctx, gs.alloc(((uint64_t)n_groups + 1) * 4));
@@ -1687,73 +1670,74 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         po->branch_ms += ms;
         po->branch_count += n_groups;
 
-        // 6. read updated pending counts; compact alive nodes into new carry.
-        // pend[d+1] held exactly nB (new nodes always have P < d): consume it
-        // on host AND device so later readbacks don't resurrect it.
+        // this level's inputs are consumed
+        carries[d].release();
+        carry_cnt[d] = 0;
+
+        // 5. distribute the new nodes into their target per-depth carries.
+        // pend[p+1] - previous snapshot = nodes newly pending at depth p.
+        uint32_t prev_pending[66];
+        memcpy(prev_pending, pending_host, sizeof(prev_pending));
         HIP_CHECK(ctx, hipMemcpy(pending_host, pend.p, 66 * 4, hipMemcpyDeviceToHost));
-        pending_host[d + 1] = 0;
+        pending_host[d + 1] = 0; // consumed (and d's carry was emptied above)
         HIP_CHECK(ctx, hipMemsetAsync((uint8_t *)pend.p + 4 * (d + 1), 0, 4,
                                       ctx->stream));
-        uint64_t n_alive = 0;
-        {
-            uint64_t alive_acc = 0;
-            for (int k = 0; k < 65; ++k) // slot 65 = keccak-block stats
-                alive_acc += pending_host[k];
-            n_alive = alive_acc - (carry_n - nB); // new nodes still pending
-        }
         po->branch_blocks = branch_blocks_base + pending_host[65];
-        // new carry = merge(Crest, alive(newn))
-        uint64_t new_carry_n = n_rest + n_alive;
-        if (new_carry_n == 0) {
-            carry_n = 0;
-        } else {
-            HIP_CHECK(ctx, carry2.alloc(new_carry_n * sizeof(node_rec)));
-            // compact alive out of newn (dead records have depth == -1)
-            DBuf alive(ctx);
-            HIP_CHECK(ctx, alive.alloc((n_alive ? n_alive : 1) * sizeof(node_rec)));
-            if (n_groups) {
-                HIP_CHECK(ctx, dead.alloc(((uint64_t)n_groups) * sizeof(node_rec)));
-                uint32_t g = sel_grid_for(n_groups);
-                hipLaunchKernelGGL(k_part_count, dim3(g), dim3(BLOCK), 0, ctx->stream,
-                                   newn.as<node_rec>(), n_groups, -1,
-                                   blk_a.as<uint32_t>(), blk_b.as<uint32_t>());
-                uint32_t ts = 0, tr = 0;
-                if (scan_u32(ctx, blk_a.as<uint32_t>(), off_a.as<uint32_t>(), g, &ts))
-                    return -1;
-                if (scan_u32(ctx, blk_b.as<uint32_t>(), off_b.as<uint32_t>(), g, &tr))
-                    return -1;
-                if (tr != n_alive) {
-                    set_err(ctx, "internal: alive count mismatch");
-                    return -1;
-                }
-                hipLaunchKernelGGL(k_part_gather, dim3(g), dim3(BLOCK), 0, ctx->stream,
-                                   newn.as<node_rec>(), n_groups, -1,
-                                   off_a.as<uint32_t>(), off_b.as<uint32_t>(),
-                                   dead.as<node_rec>(), alive.as<node_rec>());
-                HIP_CHECK(ctx, hipGetLastError());
+        for (int p = d - 1; p >= 0; --p) {
+            uint64_t fresh = pending_host[p + 1] - prev_pending[p + 1];
+            if (fresh == 0)
+                continue;
+            // extract depth-p nodes out of newn (stable -> sorted by s)
+            HIP_CHECK(ctx, newp.alloc(fresh * sizeof(node_rec)));
+            HIP_CHECK(ctx, trash.alloc((uint64_t)n_groups * sizeof(node_rec)));
+            uint32_t g = sel_grid_for(n_groups);
+            hipLaunchKernelGGL(k_part_count, dim3(g), dim3(BLOCK), 0, ctx->stream,
+                               newn.as<node_rec>(), n_groups, p,
+                               blk_a.as<uint32_t>(), blk_b.as<uint32_t>());
+            uint32_t ts = 0;
+            if (scan_u32(ctx, blk_a.as<uint32_t>(), off_a.as<uint32_t>(), g, &ts))
+                return -1;
+            uint32_t tr = 0;
+            if (scan_u32(ctx, blk_b.as<uint32_t>(), off_b.as<uint32_t>(), g, &tr))
+                return -1;
+            if (ts != fresh) {
+                set_err(ctx, "internal: fresh-node count mismatch");
+                return -1;
             }
-            hipLaunchKernelGGL(k_merge_a, dim3(grid_for(n_rest ? n_rest : 1)),
-                               dim3(BLOCK), 0, ctx->stream,
-                               nB ? Crest.as<node_rec>() : carry.as<node_rec>(),
-                               n_rest, alive.as<node_rec>(), n_alive,
-                               carry2.as<node_rec>());
-            hipLaunchKernelGGL(k_merge_b, dim3(grid_for(n_alive ? n_alive : 1)),
-                               dim3(BLOCK), 0, ctx->stream,
-                               nB ? Crest.as<node_rec>() : carry.as<node_rec>(),
-                               n_rest, alive.as<node_rec>(), n_alive,
-                               carry2.as<node_rec>());
+            hipLaunchKernelGGL(k_part_gather, dim3(g), dim3(BLOCK), 0, ctx->stream,
+                               newn.as<node_rec>(), n_groups, p,
+                               off_a.as<uint32_t>(), off_b.as<uint32_t>(),
+                               newp.as<node_rec>(), trash.as<node_rec>());
             HIP_CHECK(ctx, hipGetLastError());
-            HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
-            swap_bufs(carry, carry2);
-            carry_n = new_carry_n;
+            if (carry_cnt[p] == 0) {
+                swap_bufs(carries[p], newp);
+                carry_cnt[p] = fresh;
+            } else {
+                uint64_t total = carry_cnt[p] + fresh;
+                HIP_CHECK(ctx, cmerge.alloc(total * sizeof(node_rec)));
+                hipLaunchKernelGGL(k_merge_a, dim3(grid_for(carry_cnt[p])),
+                                   dim3(BLOCK), 0, ctx->stream,
+                                   carries[p].as<node_rec>(), carry_cnt[p],
+                                   newp.as<node_rec>(), fresh,
+                                   cmerge.as<node_rec>());
+                hipLaunchKernelGGL(k_merge_b, dim3(grid_for(fresh)), dim3(BLOCK), 0,
+                                   ctx->stream, carries[p].as<node_rec>(),
+                                   carry_cnt[p], newp.as<node_rec>(), fresh,
+                                   cmerge.as<node_rec>());
+                HIP_CHECK(ctx, hipGetLastError());
+                HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+                swap_bufs(carries[p], cmerge);
+                carry_cnt[p] = total;
+            }
         }
     }
     hipEventDestroy(ev0);
     hipEventDestroy(ev1);
-    if (carry_n != 0) {
-        set_err(ctx, "internal: carry not empty after level 0");
-        return -1;
-    }
+    for (int p = 0; p < 64; ++p)
+        if (carry_cnt[p] != 0) {
+            set_err(ctx, "internal: carry not empty after level 0");
+            return -1;
+        }
     return 0;
 }
 
