@@ -284,7 +284,7 @@ struct __attribute__((aligned(16))) node_rec {
     int8_t depth;     // parent branch depth this node waits for (-1 = root)
     uint8_t ref_len;  // 1..33 (0 marks a consumed root record)
     uint8_t ref[33];
-    uint8_t pad_;
+    uint8_t pad_;     // child nibble at `depth` (nib(key, depth); 0 for roots)
 };
 static_assert(sizeof(node_rec) == 48, "node_rec must be 48 bytes");
 
@@ -514,7 +514,9 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_storage(
             uint8_t rl;
             make_ref(slot, len, hash, r->ref, &rl);
             r->ref_len = rl;
-            r->pad_ = 0;
+            // child nibble at the parent branch (known now; saves the branch
+            // assembler a scattered key gather per member)
+            r->pad_ = D >= 0 ? nib_of(key, D) : 0;
             depths[i] = (uint8_t)(D + 1);
             atomicAdd(&hist_l[D + 1], 1u);
             if (D == -1) // single-slot storage trie: root = keccak(leaf RLP)
@@ -614,7 +616,7 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
         r.seg = subtree ? (uint32_t)(key[0] >> 4) : 0u;
         r.depth = (int8_t)D;
         make_ref(slot, len, hash, r.ref, &r.ref_len);
-        r.pad_ = 0;
+        r.pad_ = D >= 0 ? nib_of(key, D) : 0;
         copy_rec(&recs[i], &r);
         depths[i] = (uint8_t)(D + 1);
         atomicAdd(&hist_l[D + 1], 1u);
@@ -923,7 +925,7 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     {
         int prev = -1;
         for (uint64_t mm = j; mm < jend && order_ok; ++mm) {
-            int nbm = nib_of(keys + (uint64_t)L[mm].s * key_stride, d);
+            int nbm = L[mm].pad_; // child nibble, stored at node creation
             order_ok &= nbm > prev;
             prev = nbm;
             nibs |= (uint64_t)nbm << (4 * (mm - j));
@@ -1108,6 +1110,7 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         uint8_t rl;
         wrap(mt.P + 1, hash, r->ref, &rl);
         r->ref_len = rl;
+        r->pad_ = nib_of(key0, mt.P);
         atomicAdd(&hist_l[mt.P + 1], 1u);
     } else {
         r->depth = -1;
