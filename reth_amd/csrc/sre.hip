@@ -2005,6 +2005,10 @@ extern "C" int sre_subtree_roots(sre_ctx *ctx, uint8_t out_child_refs[16][33],
     HIP_CHECK(ctx, counts.alloc(16 * 8));
     HIP_CHECK(ctx, hipMemsetAsync(clens.p, 0, 16, ctx->stream));
     HIP_CHECK(ctx, hipMemsetAsync(counts.p, 0, 16 * 8, ctx->stream));
+    // zero refs/roots too: pool-recycled buffers hold stale bytes, and rows
+    // for absent nibbles must come back deterministically zeroed
+    HIP_CHECK(ctx, hipMemsetAsync(crefs.p, 0, 16 * 33, ctx->stream));
+    HIP_CHECK(ctx, hipMemsetAsync(roots.p, 0, 16 * 32, ctx->stream));
 
     pass_out po;
     if (run_storage_pass(ctx, acct_roots.as<uint8_t>(), &po, err.as<uint32_t>()))
