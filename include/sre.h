@@ -68,6 +68,32 @@ typedef struct {
 
 typedef struct sre_ctx sre_ctx;
 
+/* One stored trie node — the engine's representation of reth's
+ * `BranchNodeCompact` + its path (crates/trie/common/src/updates.rs:17;
+ * node shape per crates/trie/common/src/hash_builder/state.rs:15-48 and the
+ * pinned expectations of crates/trie/db/tests/trie.rs:519-589 /
+ * crates/trie/trie/src/node_iter.rs:383-465):
+ *   - a branch node is stored iff hash_mask != 0;
+ *   - hash_mask bit c is set iff child c's subtree top (after any extension)
+ *     is a branch whose RLP is >= 32 bytes; `hashes` holds those branch-RLP
+ *     keccaks in ascending nibble order;
+ *   - tree_mask bit c is set iff child c's subtree top branch is itself
+ *     stored;
+ *   - the path-[] row (root is a branch) carries root_hash.
+ * kind 0 = account trie; kind 1 = storage trie of acct_key. 624 bytes. */
+typedef struct {
+    uint8_t  acct_key[32];   /* kind 1 only */
+    uint8_t  kind;
+    uint8_t  path_len;       /* nibbles, 0..63 */
+    uint8_t  path[32];       /* packed nibbles, high nibble first */
+    uint8_t  num_hashes;     /* popcount(hash_mask) */
+    uint8_t  root_hash_set;  /* 1 => root_hash valid (path_len == 0) */
+    uint16_t state_mask, tree_mask, hash_mask;
+    uint8_t  root_hash[32];
+    uint8_t  hashes[16][32];
+    uint8_t  pad_[6];
+} sre_update_row;
+
 /* Per-call timing/throughput stats (HIP-event measured, for bench reporting;
  * mirrors the spirit of crates/trie/trie/src/metrics.rs:7-60). Times in ms. */
 typedef struct {
@@ -119,6 +145,17 @@ int sre_finish_top(sre_ctx *ctx,
                    const uint8_t  root_hash[16][32],
                    const uint64_t counts[16],
                    uint8_t out_root[32]);
+
+/* Compute the state root AND retain the stored trie nodes (TrieUpdates) —
+ * the surface of StateRootProvider::state_root_with_updates
+ * (crates/storage/storage-api/src/trie.rs:30) / StateRoot::root_with_updates
+ * feeding TrieWriter::write_trie_updates (storage-api/src/trie.rs:176-188).
+ * Rows are retrieved afterwards with sre_updates_count / sre_updates_get
+ * (account-trie rows first, then storage rows grouped by account; each list
+ * sorted by path). A full rebuild has no removed_nodes. */
+int sre_root_with_updates(sre_ctx *ctx, uint8_t out_root[32]);
+int64_t sre_updates_count(sre_ctx *ctx);
+int sre_updates_get(sre_ctx *ctx, sre_update_row *out, uint64_t max_rows);
 
 int sre_get_stats(sre_ctx *ctx, sre_stats *out);
 

@@ -31,6 +31,22 @@ STORAGE_DTYPE = np.dtype([
 ])  # 96 bytes, matches sre_storage_entry
 assert STORAGE_DTYPE.itemsize == 96
 
+UPDATE_DTYPE = np.dtype([
+    ("acct_key", np.uint8, 32),
+    ("kind", np.uint8),
+    ("path_len", np.uint8),
+    ("path", np.uint8, 32),
+    ("num_hashes", np.uint8),
+    ("root_hash_set", np.uint8),
+    ("state_mask", "<u2"),
+    ("tree_mask", "<u2"),
+    ("hash_mask", "<u2"),
+    ("root_hash", np.uint8, 32),
+    ("hashes", np.uint8, (16, 32)),
+    ("pad", np.uint8, 6),
+])  # 624 bytes, matches sre_update_row
+assert UPDATE_DTYPE.itemsize == 624
+
 _lib = None
 
 
@@ -99,6 +115,26 @@ def subtree_roots(accounts: np.ndarray, storage: np.ndarray):
                                    _ptr(refs), _ptr(lens), _ptr(roots),
                                    _ptr(counts)))
     return refs, lens, roots, counts
+
+
+def state_root_with_updates(accounts: np.ndarray, storage: np.ndarray):
+    """Returns (root_bytes, rows ndarray of UPDATE_DTYPE) — the stored
+    trie nodes (TrieUpdates) of a full rebuild."""
+    assert accounts.dtype == ACCOUNT_DTYPE and storage.dtype == STORAGE_DTYPE
+    out = (ctypes.c_uint8 * 32)()
+    rows_p = ctypes.c_void_p()
+    n_rows = ctypes.c_uint64()
+    _check(lib().okc_state_root_with_updates(
+        _ptr(accounts), len(accounts), _ptr(storage), len(storage), out,
+        ctypes.byref(rows_p), ctypes.byref(n_rows)))
+    n = n_rows.value
+    if n:
+        buf = ctypes.string_at(rows_p.value, n * UPDATE_DTYPE.itemsize)
+        rows = np.frombuffer(buf, dtype=UPDATE_DTYPE).copy()
+        lib().okc_free_updates(rows_p)
+    else:
+        rows = np.zeros(0, dtype=UPDATE_DTYPE)
+    return bytes(out), rows
 
 
 def finish_top(refs, lens, roots, counts) -> bytes:

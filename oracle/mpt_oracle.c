@@ -153,13 +153,44 @@ static void make_ref(const uint8_t *rlp, size_t len, ref_t *ref)
 
 /* ---------------- generic sorted-stream trie build ---------------- */
 
+/* TrieUpdates collector (see sre_update_row in sre.h for the semantics,
+ * pinned by /root/reference/crates/trie/db/tests/trie.rs:519-589 and
+ * crates/trie/trie/src/node_iter.rs:383-465). */
+typedef struct {
+    sre_update_row *rows;
+    uint64_t n, cap;
+    uint8_t kind;                /* 0 account trie, 1 storage trie */
+    const uint8_t *acct_key;     /* kind 1 */
+} upd_ctx;
+
+/* what a parent branch needs to know about a child subtree */
+typedef struct {
+    uint8_t is_branch;           /* subtree top (after ext) is a branch */
+    uint8_t branch_hashed;       /* that branch's RLP >= 32 bytes */
+    uint8_t stored;              /* that branch was emitted (hash_mask != 0) */
+    uint8_t branch_hash[32];
+} child_info;
+
 typedef struct {
     const uint8_t *keys;   /* 32-byte keys at key_stride apart, sorted asc */
     size_t key_stride;
     /* writes the leaf's RLP-encoded value for leaf idx, returns length */
     size_t (*value)(const void *vctx, uint64_t idx, uint8_t *out);
     const void *vctx;
+    upd_ctx *upd;          /* optional TrieUpdates collector */
 } trie_src;
+
+static int upd_push(upd_ctx *u, const sre_update_row *row)
+{
+    if (u->n == u->cap) {
+        u->cap = u->cap ? u->cap * 2 : 64;
+        u->rows = (sre_update_row *)realloc(u->rows, u->cap * sizeof(*u->rows));
+        if (!u->rows)
+            return -1;
+    }
+    u->rows[u->n++] = *row;
+    return 0;
+}
 
 #define NODE_MAX 532 /* 17-item branch: 16*33+1 payload + 3-byte header */
 
@@ -167,8 +198,10 @@ typedef struct {
  * [0,pos). Writes the node's RLP to out (caller buffer >= NODE_MAX), returns
  * its length, and fills *ref with the embedded reference. */
 static size_t build(const trie_src *src, uint64_t lo, uint64_t hi, int pos,
-                    uint8_t *out, ref_t *ref)
+                    uint8_t *out, ref_t *ref, child_info *ci)
 {
+    if (ci)
+        memset(ci, 0, sizeof(*ci));
     const uint8_t *key_lo = src->keys + lo * src->key_stride;
     if (hi - lo == 1) {
         uint8_t val[128];
@@ -192,7 +225,9 @@ static size_t build(const trie_src *src, uint64_t lo, uint64_t hi, int pos,
         /* extension node over key[pos..p) */
         ref_t child;
         uint8_t child_rlp[NODE_MAX];
-        build(src, lo, hi, p, child_rlp, &child);
+        /* masks/storage see through the extension: the stored-trie model
+         * keys nodes by the BRANCH path and reconstructs extensions */
+        build(src, lo, hi, p, child_rlp, &child, ci);
         uint8_t hp[33];
         size_t hp_len = hp_encode(key_lo, pos, p, 0, hp);
         uint8_t payload[72];
@@ -206,6 +241,9 @@ static size_t build(const trie_src *src, uint64_t lo, uint64_t hi, int pos,
     }
     /* branch at pos */
     uint8_t payload[NODE_MAX];
+    sre_update_row row;
+    memset(&row, 0, sizeof(row));
+    int nhash = 0;
     size_t w = 0;
     uint64_t i = lo;
     for (int b = 0; b < 16; b++) {
@@ -217,9 +255,17 @@ static size_t build(const trie_src *src, uint64_t lo, uint64_t hi, int pos,
         } else {
             ref_t child;
             uint8_t child_rlp[NODE_MAX];
-            build(src, i, j, pos + 1, child_rlp, &child);
+            child_info cci;
+            build(src, i, j, pos + 1, child_rlp, &child, &cci);
             memcpy(payload + w, child.b, child.len);
             w += child.len;
+            row.state_mask |= (uint16_t)(1u << b);
+            if (cci.is_branch && cci.branch_hashed) {
+                row.hash_mask |= (uint16_t)(1u << b);
+                memcpy(row.hashes[nhash++], cci.branch_hash, 32);
+            }
+            if (cci.stored)
+                row.tree_mask |= (uint16_t)(1u << b);
             i = j;
         }
     }
@@ -227,6 +273,35 @@ static size_t build(const trie_src *src, uint64_t lo, uint64_t hi, int pos,
     size_t h = rlp_list_hdr(w, out);
     memcpy(out + h, payload, w);
     make_ref(out, h + w, ref);
+    if (ci || (src->upd && row.hash_mask)) {
+        uint8_t bh[32];
+        okc_keccak256(out, h + w, bh);
+        if (ci) {
+            ci->is_branch = 1;
+            ci->branch_hashed = (h + w) >= 32;
+            ci->stored = row.hash_mask != 0;
+            memcpy(ci->branch_hash, bh, 32);
+        }
+        if (src->upd && row.hash_mask) { /* stored iff hash_mask != 0 */
+            row.kind = src->upd->kind;
+            if (src->upd->acct_key)
+                memcpy(row.acct_key, src->upd->acct_key, 32);
+            row.path_len = (uint8_t)pos;
+            for (int k = 0; k < pos; k++) {
+                uint8_t nb_ = nib(key_lo, k);
+                if (k & 1)
+                    row.path[k / 2] |= nb_;
+                else
+                    row.path[k / 2] = (uint8_t)(nb_ << 4);
+            }
+            row.num_hashes = (uint8_t)nhash;
+            if (pos == 0) { /* root branch row carries the root hash */
+                row.root_hash_set = 1;
+                memcpy(row.root_hash, bh, 32);
+            }
+            upd_push(src->upd, &row);
+        }
+    }
     return h + w;
 }
 
@@ -284,9 +359,9 @@ static int is_zero32(const uint8_t *p)
 
 /* Per-account storage roots; also validates all input contracts.
  * roots_out: na*32 bytes. */
-int okc_storage_roots(const sre_account_entry *acct, uint64_t na,
-                      const sre_storage_entry *st, uint64_t ns,
-                      uint8_t *roots_out)
+static int storage_roots_impl(const sre_account_entry *acct, uint64_t na,
+                              const sre_storage_entry *st, uint64_t ns,
+                              uint8_t *roots_out, upd_ctx *upd)
 {
     for (uint64_t i = 1; i < na; i++)
         if (memcmp(acct[i - 1].key, acct[i].key, 32) >= 0)
@@ -301,6 +376,7 @@ int okc_storage_roots(const sre_account_entry *acct, uint64_t na,
             return 4;
 
     trie_src src;
+    src.upd = NULL;
     src.key_stride = sizeof(sre_storage_entry);
     src.value = storage_value;
     src.vctx = st;
@@ -319,15 +395,104 @@ int okc_storage_roots(const sre_account_entry *acct, uint64_t na,
             /* re-base keys AND value ctx so build's idx 0 == entry lo */
             src.keys = st[lo].slot_key;
             src.vctx = st + lo;
+            if (upd) {
+                upd->kind = 1;
+                upd->acct_key = acct[j].key;
+                src.upd = upd;
+            }
             uint8_t rlp[NODE_MAX];
             ref_t ref;
-            size_t len = build(&src, 0, i - lo, 0, rlp, &ref);
+            size_t len = build(&src, 0, i - lo, 0, rlp, &ref, NULL);
             okc_keccak256(rlp, len, roots_out + 32 * j);
         }
     }
     if (i != ns)
         return 3;
     return 0;
+}
+
+int okc_storage_roots(const sre_account_entry *acct, uint64_t na,
+                      const sre_storage_entry *st, uint64_t ns,
+                      uint8_t *roots_out)
+{
+    return storage_roots_impl(acct, na, st, ns, roots_out, NULL);
+}
+
+/* sort rows: account-trie rows first (by path), then storage rows grouped
+ * by account key (by path within) — the order updates.rs `into_sorted`
+ * produces. */
+static int cmp_rows(const void *a_, const void *b_)
+{
+    const sre_update_row *a = (const sre_update_row *)a_;
+    const sre_update_row *b = (const sre_update_row *)b_;
+    if (a->kind != b->kind)
+        return a->kind < b->kind ? -1 : 1;
+    if (a->kind == 1) {
+        int c = memcmp(a->acct_key, b->acct_key, 32);
+        if (c)
+            return c;
+    }
+    int minl = a->path_len < b->path_len ? a->path_len : b->path_len;
+    for (int k = 0; k < minl; k++) {
+        uint8_t na_ = (a->path[k / 2] >> ((k & 1) ? 0 : 4)) & 0xf;
+        uint8_t nb_ = (b->path[k / 2] >> ((k & 1) ? 0 : 4)) & 0xf;
+        if (na_ != nb_)
+            return na_ < nb_ ? -1 : 1;
+    }
+    if (a->path_len != b->path_len)
+        return a->path_len < b->path_len ? -1 : 1;
+    return 0;
+}
+
+/* state root + TrieUpdates (StateRoot::root_with_updates semantics,
+ * crates/trie/trie/src/trie.rs:141; full rebuild => no removed nodes). */
+int okc_state_root_with_updates(const sre_account_entry *acct, uint64_t na,
+                                const sre_storage_entry *st, uint64_t ns,
+                                uint8_t out_root[32],
+                                sre_update_row **rows_out, uint64_t *n_rows)
+{
+    *rows_out = NULL;
+    *n_rows = 0;
+    if (na == 0) {
+        if (ns != 0)
+            return 3;
+        memcpy(out_root, EMPTY_ROOT, 32);
+        return 0;
+    }
+    uint8_t *roots = (uint8_t *)malloc(na * 32);
+    if (!roots)
+        return 5;
+    upd_ctx u;
+    memset(&u, 0, sizeof(u));
+    int rc = storage_roots_impl(acct, na, st, ns, roots, &u);
+    if (rc) {
+        free(roots);
+        free(u.rows);
+        return rc;
+    }
+    u.kind = 0;
+    u.acct_key = NULL;
+    acct_vctx vc = { acct, roots };
+    trie_src src;
+    src.upd = &u;
+    src.keys = acct[0].key;
+    src.key_stride = sizeof(sre_account_entry);
+    src.value = account_value;
+    src.vctx = &vc;
+    uint8_t rlp[NODE_MAX];
+    ref_t ref;
+    size_t len = build(&src, 0, na, 0, rlp, &ref, NULL);
+    okc_keccak256(rlp, len, out_root);
+    free(roots);
+    qsort(u.rows, u.n, sizeof(sre_update_row), cmp_rows);
+    *rows_out = u.rows;
+    *n_rows = u.n;
+    return 0;
+}
+
+void okc_free_updates(sre_update_row *rows)
+{
+    free(rows);
 }
 
 int okc_state_root(const sre_account_entry *acct, uint64_t na,
@@ -350,13 +515,14 @@ int okc_state_root(const sre_account_entry *acct, uint64_t na,
     }
     acct_vctx vc = { acct, roots };
     trie_src src;
+    src.upd = NULL;
     src.keys = acct[0].key;
     src.key_stride = sizeof(sre_account_entry);
     src.value = account_value;
     src.vctx = &vc;
     uint8_t rlp[NODE_MAX];
     ref_t ref;
-    size_t len = build(&src, 0, na, 0, rlp, &ref);
+    size_t len = build(&src, 0, na, 0, rlp, &ref, NULL);
     okc_keccak256(rlp, len, out);
     free(roots);
     return 0;
@@ -389,6 +555,7 @@ int okc_subtree_roots(const sre_account_entry *acct, uint64_t na,
             j++;
         acct_vctx vc = { acct + i, roots + 32 * i };
         trie_src src;
+        src.upd = NULL;
         src.keys = acct[i].key;
         src.key_stride = sizeof(sre_account_entry);
         src.value = account_value;
@@ -396,11 +563,11 @@ int okc_subtree_roots(const sre_account_entry *acct, uint64_t na,
         uint8_t rlp[NODE_MAX];
         ref_t ref;
         /* child-of-root-branch form: path consumed through nibble 0 */
-        build(&src, 0, j - i, 1, rlp, &ref);
+        build(&src, 0, j - i, 1, rlp, &ref, NULL);
         memcpy(out_child_refs[b], ref.b, ref.len);
         out_child_lens[b] = ref.len;
         /* standalone-trie form (only used when this nibble is the whole trie) */
-        size_t len = build(&src, 0, j - i, 0, rlp, &ref);
+        size_t len = build(&src, 0, j - i, 0, rlp, &ref, NULL);
         okc_keccak256(rlp, len, out_root_hash[b]);
         out_counts[b] = j - i;
         i = j;
