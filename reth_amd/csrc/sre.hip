@@ -39,6 +39,7 @@
 #include <cstdint>
 #include <cstdio>
 #include <cstring>
+#include <algorithm>
 #include <string>
 #include <vector>
 
@@ -831,8 +832,10 @@ struct br_meta {
     uint16_t br_len;   // 0 marks an invariant-violation group (error flagged)
     int8_t P;          // parent depth (-1 = segment root)
     uint8_t d;         // branch depth (level)
+    uint8_t flags;     // bit0: stored (hash_mask != 0 — has a hashed-branch child)
+    uint8_t pad_[3];
 };
-static_assert(sizeof(br_meta) == 16, "br_meta must be 16 bytes");
+static_assert(sizeof(br_meta) == 20, "br_meta must be 20 bytes");
 
 // scatter group-start positions: gs[gidx[j]] = j (gs[n_groups] set by host)
 __global__ void k_group_starts(const uint32_t *__restrict__ flags,
@@ -922,19 +925,24 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     uint64_t nibs = 0;
     int payload = 1 + (16 - nmem);
     bool order_ok = nmem >= 2 && nmem <= 16;
+    bool stored = false; // this branch gets a TrieUpdates row (hash_mask != 0)
     {
         int prev = -1;
         for (uint64_t mm = j; mm < jend && order_ok; ++mm) {
-            int nbm = L[mm].pad_; // child nibble, stored at node creation
+            uint8_t pb = L[mm].pad_;
+            int nbm = pb & 0xF; // child nibble, stored at node creation
             order_ok &= nbm > prev;
             prev = nbm;
             nibs |= (uint64_t)nbm << (4 * (mm - j));
             payload += L[mm].ref_len;
+            stored |= (pb & 0x20) != 0; // member is a hashed branch
         }
     }
+    mt.flags = stored ? 1 : 0;
     if (!order_ok || payload > 529) {
         atomicOr(err, 1u << E_INTERNAL);
         mt.br_len = 0;
+        mt.flags = 0;
         meta[g] = mt;
         return;
     }
@@ -992,7 +1000,9 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     uint32_t n_groups, const uint8_t *__restrict__ keys, uint64_t key_stride,
     int subtree, node_rec *__restrict__ out, uint8_t *__restrict__ seg_roots,
     uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
-    uint32_t *__restrict__ pending, uint32_t *__restrict__ err)
+    uint32_t *__restrict__ pending, uint32_t *__restrict__ err,
+    uint8_t *__restrict__ bhash_by_s /* updates mode: branch hash by interval
+                                        start; null otherwise */)
 {
     // pending[] updates are LDS-aggregated: one global atomic per counter
     // per block instead of per group (a single hot counter word saturates at
@@ -1104,13 +1114,17 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         kblocks += 1;
     };
 
+    if (bhash_by_s)
+        memcpy(bhash_by_s + 32ull * mt.s, br_hash, 32);
+    uint8_t upd_bits = (uint8_t)(((mt.flags & 1) << 4) |
+                                 ((mt.br_len >= 32 ? 1 : 0) << 5));
     if (mt.P >= 0) {
         r->depth = mt.P;
         uint64_t hash[4];
         uint8_t rl;
         wrap(mt.P + 1, hash, r->ref, &rl);
         r->ref_len = rl;
-        r->pad_ = nib_of(key0, mt.P);
+        r->pad_ = (uint8_t)(nib_of(key0, mt.P) | upd_bits);
         atomicAdd(&hist_l[mt.P + 1], 1u);
     } else {
         r->depth = -1;
@@ -1132,6 +1146,77 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     __syncthreads();
     if (threadIdx.x < 66 && hist_l[threadIdx.x])
         atomicAdd(&pending[threadIdx.x], hist_l[threadIdx.x]);
+}
+
+// TrieUpdates emission (updates mode): one row per STORED branch
+// (hash_mask != 0 — semantics in sre.h, pinned by the reference tests).
+// Runs per level chunk after k_branch_hash; row slots via LDS-aggregated
+// atomic bump. kind: 0 account trie, 1 storage trie (acct_key patched on
+// host from the stashed seg id).
+__global__ void __launch_bounds__(BLOCK) k_emit_updates(
+    const node_rec *__restrict__ L, const uint32_t *__restrict__ gs,
+    uint32_t n_groups, const br_meta *__restrict__ meta,
+    const uint8_t *__restrict__ keys, uint64_t key_stride,
+    const uint8_t *__restrict__ bhash_by_s, int kind,
+    sre_update_row *__restrict__ rows, uint32_t *__restrict__ row_counter)
+{
+    __shared__ uint32_t lds[BLOCK];
+    __shared__ uint32_t base;
+    uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    bool stored = g < n_groups && meta[g].br_len != 0 && (meta[g].flags & 1);
+    uint32_t excl, total;
+    block_scan(lds, stored ? 1u : 0u, &excl, &total);
+    if (threadIdx.x == 0)
+        base = total ? atomicAdd(row_counter, total) : 0;
+    __syncthreads();
+    if (!stored)
+        return;
+    br_meta mt = meta[g];
+    sre_update_row *row = &rows[base + excl];
+    uint64_t j = gs[g], jend = gs[g + 1];
+    uint16_t state = 0, tree = 0, hashm = 0;
+    int nh = 0;
+    for (uint64_t m = j; m < jend; ++m) {
+        uint8_t pb = L[m].pad_;
+        int b = pb & 0xF;
+        state |= (uint16_t)(1u << b);
+        if (pb & 0x20) { // child subtree top is a hashed branch
+            hashm |= (uint16_t)(1u << b);
+            memcpy(row->hashes[nh++], bhash_by_s + 32ull * L[m].s, 32);
+        }
+        if (pb & 0x10) // child branch is itself stored
+            tree |= (uint16_t)(1u << b);
+    }
+    for (int k = nh; k < 16; ++k)
+        for (int q = 0; q < 32; ++q)
+            row->hashes[k][q] = 0;
+    row->state_mask = state;
+    row->tree_mask = tree;
+    row->hash_mask = hashm;
+    row->num_hashes = (uint8_t)nh;
+    row->kind = (uint8_t)kind;
+    int d = mt.d;
+    row->path_len = (uint8_t)d;
+    const uint8_t *key0 = keys + (uint64_t)mt.s * key_stride;
+    for (int k = 0; k < 32; ++k) {
+        uint8_t hi = (2 * k < d) ? nib_of(key0, 2 * k) : 0;
+        uint8_t lo = (2 * k + 1 < d) ? nib_of(key0, 2 * k + 1) : 0;
+        row->path[k] = (uint8_t)((hi << 4) | lo);
+    }
+    if (d == 0) { // root branch row carries the root hash
+        row->root_hash_set = 1;
+        memcpy(row->root_hash, bhash_by_s + 32ull * mt.s, 32);
+    } else {
+        row->root_hash_set = 0;
+        for (int q = 0; q < 32; ++q)
+            row->root_hash[q] = 0;
+    }
+    // stash seg for the host-side acct_key patch; zero the rest
+    for (int q = 0; q < 32; ++q)
+        row->acct_key[q] = 0;
+    memcpy(row->pad_, &mt.seg, 4);
+    row->pad_[4] = 0;
+    row->pad_[5] = 0;
 }
 
 // ---------------------------------------------------------------------------
@@ -1276,6 +1361,9 @@ struct sre_ctx {
     uint64_t ns = 0;
     bool own_st = false;
     sre_stats stats{};
+    // TrieUpdates retention (sre_root_with_updates)
+    bool retain_updates = false;
+    std::vector<sre_update_row> updates;
     // size-class buffer pool: the level machinery allocates/frees dozens of
     // transient arrays per level; hipMalloc latency would dominate small
     // jobs. Freed buffers are cached by power-of-2 class and reused (also
@@ -1539,11 +1627,13 @@ struct pass_out {
 
 static int check_err(sre_ctx *ctx, uint32_t *d_err);
 
+// updates_kind: -1 = off; 0/1 = retain TrieUpdates rows of this trie kind
+// (d_bhash then holds 32 B per underlying entry for branch-hash lookups).
 static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_depths,
                       const int8_t *d_lcp, const uint8_t *d_keys, uint64_t key_stride,
                       const uint32_t *hist_host, int subtree, uint8_t *d_seg_roots,
                       uint8_t *d_child_refs, uint8_t *d_child_lens, uint32_t *d_err,
-                      pass_out *po)
+                      pass_out *po, int updates_kind, uint8_t *d_bhash)
 {
     int maxd = -1;
     for (int d = 63; d >= 0; --d)
@@ -1566,7 +1656,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 
     DBuf Lsel(ctx), Lbuf(ctx), newn(ctx), trash(ctx), newp(ctx), cmerge(ctx);
     DBuf flags(ctx), gidx(ctx), pend(ctx);
-    DBuf gs(ctx), scratch(ctx), meta(ctx);
+    DBuf gs(ctx), scratch(ctx), meta(ctx), urows(ctx), urow_cnt(ctx);
     DBuf blk_a(ctx), blk_b(ctx), off_a(ctx), off_b(ctx);
     uint64_t max_blocks = sel_grid_for(n) + 2;
     HIP_CHECK(ctx, blk_a.alloc(max_blocks * 4));
@@ -1663,8 +1753,32 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                meta.as<br_meta>(),
                                gc, d_keys, key_stride, subtree,
                                newn.as<node_rec>() + g0, d_seg_roots, d_child_refs,
-                               d_child_lens, pend.as<uint32_t>(), d_err);
+                               d_child_lens, pend.as<uint32_t>(), d_err, d_bhash);
             HIP_CHECK(ctx, hipGetLastError());
+            if (updates_kind >= 0) {
+                if (!urows.p) {
+                    HIP_CHECK(ctx, urows.alloc(chunk * sizeof(sre_update_row)));
+                    HIP_CHECK(ctx, urow_cnt.alloc(4));
+                }
+                HIP_CHECK(ctx, hipMemsetAsync(urow_cnt.p, 0, 4, ctx->stream));
+                hipLaunchKernelGGL(k_emit_updates, dim3(grid_for(gc)), dim3(BLOCK),
+                                   0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
+                                   meta.as<br_meta>(), d_keys, key_stride, d_bhash,
+                                   updates_kind, urows.as<sre_update_row>(),
+                                   urow_cnt.as<uint32_t>());
+                HIP_CHECK(ctx, hipGetLastError());
+                uint32_t nrows = 0;
+                HIP_CHECK(ctx, hipMemcpy(&nrows, urow_cnt.p, 4,
+                                         hipMemcpyDeviceToHost));
+                if (nrows) {
+                    size_t old_sz = ctx->updates.size();
+                    ctx->updates.resize(old_sz + nrows);
+                    HIP_CHECK(ctx, hipMemcpy(ctx->updates.data() + old_sz,
+                                             urows.p,
+                                             (uint64_t)nrows * sizeof(sre_update_row),
+                                             hipMemcpyDeviceToHost));
+                }
+            }
         }
         hipEventRecord(ev1, ctx->stream);
         HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
@@ -1817,12 +1931,35 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
     uint32_t hist_host[66];
     HIP_CHECK(ctx, hipMemcpy(hist_host, hist.p, 66 * 4, hipMemcpyDeviceToHost));
 
+    DBuf bhash(ctx);
+    if (ctx->retain_updates)
+        HIP_CHECK(ctx, bhash.alloc(ns * 32));
+    size_t upd_start = ctx->updates.size();
     const uint8_t *keys =
         (const uint8_t *)ctx->d_st + offsetof(sre_storage_entry, slot_key);
     if (run_levels(ctx, ns, recs.as<node_rec>(), depths.as<uint8_t>(),
                    lcp.as<int8_t>(), keys, sizeof(sre_storage_entry), hist_host, 0,
-                   seg_roots.as<uint8_t>(), nullptr, nullptr, d_err, po))
+                   seg_roots.as<uint8_t>(), nullptr, nullptr, d_err, po,
+                   ctx->retain_updates ? 1 : -1, bhash.as<uint8_t>()))
         return -1;
+    if (ctx->retain_updates && ctx->updates.size() > upd_start) {
+        // patch acct_key from the stashed seg ids: seg -> account index ->
+        // 32-byte hashed key (strided D2H of the key column)
+        std::vector<uint32_t> seg_acct_h(n_seg);
+        HIP_CHECK(ctx, hipMemcpy(seg_acct_h.data(), seg_acct.p,
+                                 (uint64_t)n_seg * 4, hipMemcpyDeviceToHost));
+        std::vector<uint8_t> keys_h((uint64_t)na * 32);
+        HIP_CHECK(ctx, hipMemcpy2D(keys_h.data(), 32, ctx->d_acct,
+                                   sizeof(sre_account_entry), 32, na,
+                                   hipMemcpyDeviceToHost));
+        for (size_t r = upd_start; r < ctx->updates.size(); ++r) {
+            sre_update_row &row = ctx->updates[r];
+            uint32_t seg;
+            memcpy(&seg, row.pad_, 4);
+            memset(row.pad_, 0, sizeof(row.pad_));
+            memcpy(row.acct_key, keys_h.data() + 32ull * seg_acct_h[seg], 32);
+        }
+    }
 
     hipLaunchKernelGGL(k_scatter_roots, dim3(grid_for(n_seg)), dim3(BLOCK), 0,
                        ctx->stream, seg_roots.as<uint8_t>(), seg_acct.as<uint32_t>(),
@@ -1873,10 +2010,21 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
     uint32_t hist_host[66];
     HIP_CHECK(ctx, hipMemcpy(hist_host, hist.p, 66 * 4, hipMemcpyDeviceToHost));
 
+    DBuf bhash(ctx);
+    if (ctx->retain_updates)
+        HIP_CHECK(ctx, bhash.alloc(na * 32));
+    size_t upd_start = ctx->updates.size();
     const uint8_t *keys = (const uint8_t *)ctx->d_acct + offsetof(sre_account_entry, key);
-    return run_levels(ctx, na, recs.as<node_rec>(), depths.as<uint8_t>(),
-                      lcp.as<int8_t>(), keys, sizeof(sre_account_entry), hist_host,
-                      subtree, d_roots, d_child_refs, d_child_lens, d_err, po);
+    if (run_levels(ctx, na, recs.as<node_rec>(), depths.as<uint8_t>(),
+                   lcp.as<int8_t>(), keys, sizeof(sre_account_entry), hist_host,
+                   subtree, d_roots, d_child_refs, d_child_lens, d_err, po,
+                   ctx->retain_updates ? 0 : -1, bhash.as<uint8_t>()))
+        return -1;
+    if (ctx->retain_updates) {
+        for (size_t r = upd_start; r < ctx->updates.size(); ++r)
+            memset(ctx->updates[r].pad_, 0, sizeof(ctx->updates[r].pad_));
+    }
+    return 0;
 }
 
 static int check_err(sre_ctx *ctx, uint32_t *d_err)
@@ -2076,6 +2224,53 @@ extern "C" int sre_finish_top(sre_ctx *ctx, const uint8_t child_refs[16][33],
     HIP_CHECK(ctx, hipGetLastError());
     HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
     HIP_CHECK(ctx, hipMemcpy(out_root, root.p, 32, hipMemcpyDeviceToHost));
+    return 0;
+}
+
+// same ordering as reth's TrieUpdates::into_sorted (updates.rs): account
+// rows first (path-sorted), then storage rows grouped by account key
+static bool row_less(const sre_update_row &a, const sre_update_row &b)
+{
+    if (a.kind != b.kind)
+        return a.kind < b.kind;
+    if (a.kind == 1) {
+        int c = memcmp(a.acct_key, b.acct_key, 32);
+        if (c)
+            return c < 0;
+    }
+    int minl = a.path_len < b.path_len ? a.path_len : b.path_len;
+    for (int k = 0; k < minl; ++k) {
+        uint8_t na_ = (a.path[k / 2] >> ((k & 1) ? 0 : 4)) & 0xf;
+        uint8_t nb_ = (b.path[k / 2] >> ((k & 1) ? 0 : 4)) & 0xf;
+        if (na_ != nb_)
+            return na_ < nb_;
+    }
+    return a.path_len < b.path_len;
+}
+
+extern "C" int sre_root_with_updates(sre_ctx *ctx, uint8_t out_root[32])
+{
+    ctx->retain_updates = true;
+    ctx->updates.clear();
+    int rc = sre_root(ctx, out_root);
+    ctx->retain_updates = false;
+    if (rc)
+        return rc;
+    std::sort(ctx->updates.begin(), ctx->updates.end(), row_less);
+    return 0;
+}
+
+extern "C" int64_t sre_updates_count(sre_ctx *ctx)
+{
+    return (int64_t)ctx->updates.size();
+}
+
+extern "C" int sre_updates_get(sre_ctx *ctx, sre_update_row *out, uint64_t max_rows)
+{
+    uint64_t n = ctx->updates.size();
+    if (n > max_rows)
+        n = max_rows;
+    memcpy(out, ctx->updates.data(), n * sizeof(sre_update_row));
     return 0;
 }
 

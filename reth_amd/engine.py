@@ -34,6 +34,23 @@ STORAGE_DTYPE = np.dtype([
     ("value", np.uint8, 32),
 ])
 
+# sre_update_row (include/sre.h): one stored BranchNodeCompact + its path
+UPDATE_DTYPE = np.dtype([
+    ("acct_key", np.uint8, 32),
+    ("kind", np.uint8),
+    ("path_len", np.uint8),
+    ("path", np.uint8, 32),
+    ("num_hashes", np.uint8),
+    ("root_hash_set", np.uint8),
+    ("state_mask", "<u2"),
+    ("tree_mask", "<u2"),
+    ("hash_mask", "<u2"),
+    ("root_hash", np.uint8, 32),
+    ("hashes", np.uint8, (16, 32)),
+    ("pad", np.uint8, 6),
+])
+assert UPDATE_DTYPE.itemsize == 624
+
 
 class SreStats(ctypes.Structure):
     _fields_ = [
@@ -75,6 +92,8 @@ def lib():
         _lib.sre_last_error.restype = ctypes.c_char_p
         _lib.sre_last_error.argtypes = [ctypes.c_void_p]
         _lib.sre_destroy.argtypes = [ctypes.c_void_p]
+        _lib.sre_updates_count.restype = ctypes.c_int64
+        _lib.sre_updates_count.argtypes = [ctypes.c_void_p]
     return _lib
 
 
@@ -136,6 +155,20 @@ class StateRootEngine:
         out = (ctypes.c_uint8 * 32)()
         self._check(self._lib.sre_root(ctypes.c_void_p(self._ctx), out))
         return bytes(out)
+
+    def root_with_updates(self):
+        """State root + TrieUpdates rows (UPDATE_DTYPE), sorted like reth's
+        TrieUpdates::into_sorted. Surface of
+        StateRootProvider::state_root_with_updates."""
+        out = (ctypes.c_uint8 * 32)()
+        self._check(self._lib.sre_root_with_updates(
+            ctypes.c_void_p(self._ctx), out))
+        n = self._lib.sre_updates_count(ctypes.c_void_p(self._ctx))
+        rows = np.zeros(n, dtype=UPDATE_DTYPE)
+        if n:
+            self._check(self._lib.sre_updates_get(
+                ctypes.c_void_p(self._ctx), _np_ptr(rows), n))
+        return bytes(out), rows
 
     def storage_roots(self, n) -> np.ndarray:
         out = np.empty((n, 32), dtype=np.uint8)
