@@ -1733,6 +1733,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         uint64_t chunk = n_groups < BR_CHUNK ? n_groups : BR_CHUNK;
         HIP_CHECK(ctx, scratch.alloc(chunk * SLOT_BR));
         HIP_CHECK(ctx, meta.alloc(chunk * sizeof(br_meta)));
+        if (updates_kind >= 0) {
+            // per-level: capacity must cover THIS level's chunk
+            HIP_CHECK(ctx, urows.alloc(chunk * sizeof(sre_update_row)));
+            HIP_CHECK(ctx, urow_cnt.alloc(4));
+        }
         hipLaunchKernelGGL(k_group_starts, dim3(grid_for(n_level)), dim3(BLOCK), 0,
                            ctx->stream, flags.as<uint32_t>(), gidx.as<uint32_t>(),
                            n_level, gs.as<uint32_t>());
@@ -1756,10 +1761,6 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                d_child_lens, pend.as<uint32_t>(), d_err, d_bhash);
             HIP_CHECK(ctx, hipGetLastError());
             if (updates_kind >= 0) {
-                if (!urows.p) {
-                    HIP_CHECK(ctx, urows.alloc(chunk * sizeof(sre_update_row)));
-                    HIP_CHECK(ctx, urow_cnt.alloc(4));
-                }
                 HIP_CHECK(ctx, hipMemsetAsync(urow_cnt.p, 0, 4, ctx->stream));
                 hipLaunchKernelGGL(k_emit_updates, dim3(grid_for(gc)), dim3(BLOCK),
                                    0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
