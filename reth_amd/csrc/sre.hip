@@ -1001,8 +1001,10 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     int subtree, node_rec *__restrict__ out, uint8_t *__restrict__ seg_roots,
     uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
     uint32_t *__restrict__ pending, uint32_t *__restrict__ err,
-    uint8_t *__restrict__ bhash_by_s /* updates mode: branch hash by interval
-                                        start; null otherwise */)
+    uint8_t *__restrict__ bhash_by_s, /* updates mode: branch hash by
+                                         interval start; null otherwise */
+    sre_update_row *__restrict__ urows, /* updates mode: this level's rows */
+    const uint32_t *__restrict__ urowidx)
 {
     // pending[] updates are LDS-aggregated: one global atomic per counter
     // per block instead of per group (a single hot counter word saturates at
@@ -1114,8 +1116,15 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         kblocks += 1;
     };
 
-    if (bhash_by_s)
+    if (bhash_by_s) {
         memcpy(bhash_by_s + 32ull * mt.s, br_hash, 32);
+        if (mt.d == 0 && urowidx && urowidx[g] != 0xFFFFFFFFu) {
+            // path-[] row: root branch carries its own hash
+            sre_update_row *ur = &urows[urowidx[g]];
+            ur->root_hash_set = 1;
+            memcpy(ur->root_hash, br_hash, 32);
+        }
+    }
     uint8_t upd_bits = (uint8_t)(((mt.flags & 1) << 4) |
                                  ((mt.br_len >= 32 ? 1 : 0) << 5));
     if (mt.P >= 0) {
@@ -1158,7 +1167,8 @@ __global__ void __launch_bounds__(BLOCK) k_emit_updates(
     uint32_t n_groups, const br_meta *__restrict__ meta,
     const uint8_t *__restrict__ keys, uint64_t key_stride,
     const uint8_t *__restrict__ bhash_by_s, int kind,
-    sre_update_row *__restrict__ rows, uint32_t *__restrict__ row_counter)
+    sre_update_row *__restrict__ rows, uint32_t *__restrict__ row_counter,
+    uint32_t *__restrict__ rowidx /* per group: row slot or ~0 */)
 {
     __shared__ uint32_t lds[BLOCK];
     __shared__ uint32_t base;
@@ -1169,6 +1179,8 @@ __global__ void __launch_bounds__(BLOCK) k_emit_updates(
     if (threadIdx.x == 0)
         base = total ? atomicAdd(row_counter, total) : 0;
     __syncthreads();
+    if (g < n_groups)
+        rowidx[g] = stored ? base + excl : 0xFFFFFFFFu;
     if (!stored)
         return;
     br_meta mt = meta[g];
@@ -1203,14 +1215,13 @@ __global__ void __launch_bounds__(BLOCK) k_emit_updates(
         uint8_t lo = (2 * k + 1 < d) ? nib_of(key0, 2 * k + 1) : 0;
         row->path[k] = (uint8_t)((hi << 4) | lo);
     }
-    if (d == 0) { // root branch row carries the root hash
-        row->root_hash_set = 1;
-        memcpy(row->root_hash, bhash_by_s + 32ull * mt.s, 32);
-    } else {
-        row->root_hash_set = 0;
-        for (int q = 0; q < 32; ++q)
-            row->root_hash[q] = 0;
-    }
+    // root_hash (d == 0 rows) is patched by k_branch_hash afterwards: this
+    // kernel runs BEFORE the level's hashing so the children's bhash_by_s
+    // entries are not yet overwritten by ancestors sharing the same
+    // leftmost interval start.
+    row->root_hash_set = 0;
+    for (int q = 0; q < 32; ++q)
+        row->root_hash[q] = 0;
     // stash seg for the host-side acct_key patch; zero the rest
     for (int q = 0; q < 32; ++q)
         row->acct_key[q] = 0;
@@ -1656,7 +1667,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 
     DBuf Lsel(ctx), Lbuf(ctx), newn(ctx), trash(ctx), newp(ctx), cmerge(ctx);
     DBuf flags(ctx), gidx(ctx), pend(ctx);
-    DBuf gs(ctx), scratch(ctx), meta(ctx), urows(ctx), urow_cnt(ctx);
+    DBuf gs(ctx), scratch(ctx), meta(ctx), urows(ctx), urow_cnt(ctx),
+        urowidx(ctx);
     DBuf blk_a(ctx), blk_b(ctx), off_a(ctx), off_b(ctx);
     uint64_t max_blocks = sel_grid_for(n) + 2;
     HIP_CHECK(ctx, blk_a.alloc(max_blocks * 4));
@@ -1737,6 +1749,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
             // per-level: capacity must cover THIS level's chunk
             HIP_CHECK(ctx, urows.alloc(chunk * sizeof(sre_update_row)));
             HIP_CHECK(ctx, urow_cnt.alloc(4));
+            HIP_CHECK(ctx, urowidx.alloc(chunk * 4));
         }
         hipLaunchKernelGGL(k_group_starts, dim3(grid_for(n_level)), dim3(BLOCK), 0,
                            ctx->stream, flags.as<uint32_t>(), gidx.as<uint32_t>(),
@@ -1753,21 +1766,29 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                d_lcp, d_keys, key_stride, d, scratch.as<uint8_t>(),
                                chunk, meta.as<br_meta>(), d_err);
             HIP_CHECK(ctx, hipGetLastError());
-            hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(gc)), dim3(BLOCK), 0,
-                               ctx->stream, scratch.as<uint8_t>(), chunk,
-                               meta.as<br_meta>(),
-                               gc, d_keys, key_stride, subtree,
-                               newn.as<node_rec>() + g0, d_seg_roots, d_child_refs,
-                               d_child_lens, pend.as<uint32_t>(), d_err, d_bhash);
-            HIP_CHECK(ctx, hipGetLastError());
             if (updates_kind >= 0) {
+                // emit BEFORE hashing: children's bhash entries must not yet
+                // be overwritten by this level's nodes (shared leftmost s)
                 HIP_CHECK(ctx, hipMemsetAsync(urow_cnt.p, 0, 4, ctx->stream));
                 hipLaunchKernelGGL(k_emit_updates, dim3(grid_for(gc)), dim3(BLOCK),
                                    0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
                                    meta.as<br_meta>(), d_keys, key_stride, d_bhash,
                                    updates_kind, urows.as<sre_update_row>(),
-                                   urow_cnt.as<uint32_t>());
+                                   urow_cnt.as<uint32_t>(), urowidx.as<uint32_t>());
                 HIP_CHECK(ctx, hipGetLastError());
+            }
+            hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(gc)), dim3(BLOCK), 0,
+                               ctx->stream, scratch.as<uint8_t>(), chunk,
+                               meta.as<br_meta>(),
+                               gc, d_keys, key_stride, subtree,
+                               newn.as<node_rec>() + g0, d_seg_roots, d_child_refs,
+                               d_child_lens, pend.as<uint32_t>(), d_err, d_bhash,
+                               updates_kind >= 0 ? urows.as<sre_update_row>()
+                                                 : nullptr,
+                               updates_kind >= 0 ? urowidx.as<uint32_t>()
+                                                 : nullptr);
+            HIP_CHECK(ctx, hipGetLastError());
+            if (updates_kind >= 0) {
                 uint32_t nrows = 0;
                 HIP_CHECK(ctx, hipMemcpy(&nrows, urow_cnt.p, 4,
                                          hipMemcpyDeviceToHost));
