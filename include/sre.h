@@ -146,6 +146,32 @@ int sre_finish_top(sre_ctx *ctx,
                    const uint64_t counts[16],
                    uint8_t out_root[32]);
 
+/* One account row of a HashedPostState overlay delta. deleted=1 destroys
+ * the account AND wipes its storage (HashedStorage::new(wiped=true),
+ * crates/trie/common/src/hashed_state.rs:425-440). 112 bytes. */
+typedef struct {
+    uint8_t  key[32];
+    uint64_t nonce;
+    uint8_t  balance[32];
+    uint8_t  code_hash[32];
+    uint8_t  deleted;
+    uint8_t  pad_[7];
+} sre_account_delta;
+
+/* Apply a HashedPostState delta to the resident state — the overlay-merge
+ * semantics of HashedPostStateCursor (crates/trie/trie/src/hashed_cursor/
+ * post_state.rs:89,313,355): post-state wins, a zero storage value deletes
+ * the slot, deleted accounts disappear along with all their storage. Both
+ * delta arrays sorted ascending (by key / (acct_key, slot_key)). The
+ * resident state is REPLACED by the merged result (device-resident), so a
+ * following sre_root computes the post-delta root — the incremental-root
+ * entry (DatabaseStateRoot::incremental_root semantics at the state level,
+ * crates/trie/db/src/state.rs:62; BASELINE configs[4]). A storage delta row
+ * for an account deleted in the same delta is invalid. */
+int sre_apply_delta(sre_ctx *ctx,
+                    const sre_account_delta *acct_delta, uint64_t n_acct,
+                    const sre_storage_entry *st_delta, uint64_t n_st);
+
 /* Compute the state root AND retain the stored trie nodes (TrieUpdates) —
  * the surface of StateRootProvider::state_root_with_updates
  * (crates/storage/storage-api/src/trie.rs:30) / StateRoot::root_with_updates

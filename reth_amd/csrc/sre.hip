@@ -1231,6 +1231,196 @@ __global__ void __launch_bounds__(BLOCK) k_emit_updates(
 }
 
 // ---------------------------------------------------------------------------
+// overlay-delta merge (sre_apply_delta): post-state wins, zero deletes,
+// deleted accounts wipe storage (post_state.rs:89,313,355 semantics)
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ int cmp_key32(const uint8_t *a, const uint8_t *b)
+{
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+        uint64_t wa = be64_at(a + 8 * k), wb = be64_at(b + 8 * k);
+        if (wa != wb)
+            return wa < wb ? -1 : 1;
+    }
+    return 0;
+}
+
+__device__ __forceinline__ int cmp_key64(const uint8_t *a, const uint8_t *b)
+{
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+        uint64_t wa = be64_at(a + 8 * k), wb = be64_at(b + 8 * k);
+        if (wa != wb)
+            return wa < wb ? -1 : 1;
+    }
+    return 0;
+}
+
+// lower_bound over keys at `stride`, comparing `klen` (32 or 64) bytes
+__device__ __forceinline__ uint64_t lb_keys(const uint8_t *base, uint64_t stride,
+                                            uint64_t n, const uint8_t *key,
+                                            int klen)
+{
+    uint64_t lo = 0, hi = n;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2;
+        int c = klen == 32 ? cmp_key32(base + mid * stride, key)
+                           : cmp_key64(base + mid * stride, key);
+        if (c < 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    return lo;
+}
+
+// base account survives iff its key is absent from the delta
+__global__ void k_ovl_base_acct_flags(const sre_account_entry *__restrict__ base,
+                                      uint64_t nb,
+                                      const sre_account_delta *__restrict__ dl,
+                                      uint64_t nd, uint32_t *__restrict__ flags)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i > nb)
+        return;
+    if (i == nb) { // scan sentinel
+        flags[i] = 0;
+        return;
+    }
+    uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_account_delta), nd,
+                         base[i].key, 32);
+    flags[i] = (p < nd && cmp_key32(dl[p].key, base[i].key) == 0) ? 0u : 1u;
+}
+
+// delta account row materializes iff not deleted (+ order validation)
+__global__ void k_ovl_delta_acct_flags(const sre_account_delta *__restrict__ dl,
+                                       uint64_t nd, uint32_t *__restrict__ flags,
+                                       uint32_t *__restrict__ del_flags,
+                                       uint32_t *__restrict__ err)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j > nd)
+        return;
+    if (j == nd) {
+        flags[j] = 0;
+        del_flags[j] = 0;
+        return;
+    }
+    if (j > 0 && cmp_key32(dl[j - 1].key, dl[j].key) >= 0)
+        atomicOr(err, 1u << E_UNSORTED_ACCT);
+    flags[j] = dl[j].deleted ? 0u : 1u;
+    del_flags[j] = dl[j].deleted ? 1u : 0u;
+}
+
+__global__ void k_ovl_scatter_acct(const sre_account_entry *__restrict__ base,
+                                   uint64_t nb, const uint32_t *__restrict__ bexcl,
+                                   const sre_account_delta *__restrict__ dl,
+                                   uint64_t nd, const uint32_t *__restrict__ dexcl,
+                                   sre_account_entry *__restrict__ out)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < nb && bexcl[i] != bexcl[i + 1]) { // base survivor
+        uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_account_delta), nd,
+                             base[i].key, 32);
+        out[bexcl[i] + dexcl[p]] = base[i];
+    }
+    uint64_t j = i; // reuse the same grid for the (smaller) delta side
+    if (j < nd && dexcl[j] != dexcl[j + 1]) {
+        uint64_t p = lb_keys((const uint8_t *)base, sizeof(sre_account_entry), nb,
+                             dl[j].key, 32);
+        sre_account_entry e;
+        memcpy(e.key, dl[j].key, 32);
+        e.nonce = dl[j].nonce;
+        memcpy(e.balance, dl[j].balance, 32);
+        memcpy(e.code_hash, dl[j].code_hash, 32);
+        out[dexcl[j] + bexcl[p]] = e;
+    }
+}
+
+// gather the keys of deleted accounts (sorted subset of the sorted delta)
+__global__ void k_ovl_gather_deleted(const sre_account_delta *__restrict__ dl,
+                                     uint64_t nd,
+                                     const uint32_t *__restrict__ delexcl,
+                                     uint8_t *__restrict__ out_keys)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= nd || delexcl[j] == delexcl[j + 1])
+        return;
+    memcpy(out_keys + 32ull * delexcl[j], dl[j].key, 32);
+}
+
+// base storage survives iff acct not deleted and (acct,slot) not in delta
+__global__ void k_ovl_base_st_flags(const sre_storage_entry *__restrict__ base,
+                                    uint64_t nb,
+                                    const sre_storage_entry *__restrict__ dl,
+                                    uint64_t nd,
+                                    const uint8_t *__restrict__ del_keys,
+                                    uint64_t ndel, uint32_t *__restrict__ flags)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i > nb)
+        return;
+    if (i == nb) {
+        flags[i] = 0;
+        return;
+    }
+    uint64_t q = lb_keys(del_keys, 32, ndel, base[i].acct_key, 32);
+    if (q < ndel && cmp_key32(del_keys + 32 * q, base[i].acct_key) == 0) {
+        flags[i] = 0; // wiped with its account
+        return;
+    }
+    uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_storage_entry), nd,
+                         base[i].acct_key, 64);
+    flags[i] = (p < nd && cmp_key64(dl[p].acct_key, base[i].acct_key) == 0)
+                   ? 0u
+                   : 1u;
+}
+
+// delta storage row materializes iff value != 0; row for a deleted account
+// is an input error (+ order validation)
+__global__ void k_ovl_delta_st_flags(const sre_storage_entry *__restrict__ dl,
+                                     uint64_t nd,
+                                     const uint8_t *__restrict__ del_keys,
+                                     uint64_t ndel, uint32_t *__restrict__ flags,
+                                     uint32_t *__restrict__ err)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j > nd)
+        return;
+    if (j == nd) {
+        flags[j] = 0;
+        return;
+    }
+    if (j > 0 && cmp_key64(dl[j - 1].acct_key, dl[j].acct_key) >= 0)
+        atomicOr(err, 1u << E_UNSORTED_STORAGE);
+    uint64_t q = lb_keys(del_keys, 32, ndel, dl[j].acct_key, 32);
+    if (q < ndel && cmp_key32(del_keys + 32 * q, dl[j].acct_key) == 0)
+        atomicOr(err, 1u << E_ORPHAN_STORAGE);
+    flags[j] = min_be_len(dl[j].value) ? 1u : 0u; // zero value = delete
+}
+
+__global__ void k_ovl_scatter_st(const sre_storage_entry *__restrict__ base,
+                                 uint64_t nb, const uint32_t *__restrict__ bexcl,
+                                 const sre_storage_entry *__restrict__ dl,
+                                 uint64_t nd, const uint32_t *__restrict__ dexcl,
+                                 sre_storage_entry *__restrict__ out)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < nb && bexcl[i] != bexcl[i + 1]) {
+        uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_storage_entry), nd,
+                             base[i].acct_key, 64);
+        out[bexcl[i] + dexcl[p]] = base[i];
+    }
+    uint64_t j = i;
+    if (j < nd && dexcl[j] != dexcl[j + 1]) {
+        uint64_t p = lb_keys((const uint8_t *)base, sizeof(sre_storage_entry), nb,
+                             dl[j].acct_key, 64);
+        out[dexcl[j] + bexcl[p]] = dl[j];
+    }
+}
+
+// ---------------------------------------------------------------------------
 // misc kernels
 // ---------------------------------------------------------------------------
 
@@ -2268,6 +2458,123 @@ static bool row_less(const sre_update_row &a, const sre_update_row &b)
             return na_ < nb_;
     }
     return a.path_len < b.path_len;
+}
+
+extern "C" int sre_apply_delta(sre_ctx *ctx,
+                               const sre_account_delta *acct_delta,
+                               uint64_t n_acct,
+                               const sre_storage_entry *st_delta, uint64_t n_st)
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    uint64_t nb = ctx->na, ns = ctx->ns;
+    DBuf dl_a(ctx), dl_s(ctx), err(ctx);
+    DBuf fa(ctx), fd(ctx), fdel(ctx), sa(ctx), sd(ctx), sdel(ctx);
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, dl_a.alloc((n_acct ? n_acct : 1) * sizeof(sre_account_delta)));
+    HIP_CHECK(ctx, dl_s.alloc((n_st ? n_st : 1) * sizeof(sre_storage_entry)));
+    if (n_acct)
+        HIP_CHECK(ctx, hipMemcpyAsync(dl_a.p, acct_delta,
+                                      n_acct * sizeof(sre_account_delta),
+                                      hipMemcpyHostToDevice, ctx->stream));
+    if (n_st)
+        HIP_CHECK(ctx, hipMemcpyAsync(dl_s.p, st_delta,
+                                      n_st * sizeof(sre_storage_entry),
+                                      hipMemcpyHostToDevice, ctx->stream));
+
+    // ---- accounts ----
+    HIP_CHECK(ctx, fa.alloc((nb + 1) * 4));
+    HIP_CHECK(ctx, fd.alloc((n_acct + 1) * 4));
+    HIP_CHECK(ctx, fdel.alloc((n_acct + 1) * 4));
+    hipLaunchKernelGGL(k_ovl_base_acct_flags, dim3(grid_for(nb + 1)), dim3(BLOCK),
+                       0, ctx->stream, ctx->d_acct, nb,
+                       dl_a.as<sre_account_delta>(), n_acct, fa.as<uint32_t>());
+    hipLaunchKernelGGL(k_ovl_delta_acct_flags, dim3(grid_for(n_acct + 1)),
+                       dim3(BLOCK), 0, ctx->stream, dl_a.as<sre_account_delta>(),
+                       n_acct, fd.as<uint32_t>(), fdel.as<uint32_t>(),
+                       err.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    uint32_t Bk = 0, Dk = 0, Ndel = 0;
+    DBuf ea(ctx), ed(ctx), edel(ctx);
+    HIP_CHECK(ctx, ea.alloc((nb + 1) * 4));
+    HIP_CHECK(ctx, ed.alloc((n_acct + 1) * 4));
+    HIP_CHECK(ctx, edel.alloc((n_acct + 1) * 4));
+    if (scan_u32(ctx, fa.as<uint32_t>(), ea.as<uint32_t>(), nb + 1, &Bk))
+        return -1;
+    if (scan_u32(ctx, fd.as<uint32_t>(), ed.as<uint32_t>(), n_acct + 1, &Dk))
+        return -1;
+    if (scan_u32(ctx, fdel.as<uint32_t>(), edel.as<uint32_t>(), n_acct + 1, &Ndel))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    uint64_t new_na = Bk + Dk;
+    void *new_acct = nullptr;
+    HIP_CHECK(ctx, hipMalloc(&new_acct,
+                             (new_na ? new_na : 1) * sizeof(sre_account_entry)));
+    uint64_t gmax = nb > n_acct ? nb : n_acct;
+    if (gmax)
+        hipLaunchKernelGGL(k_ovl_scatter_acct, dim3(grid_for(gmax)), dim3(BLOCK), 0,
+                           ctx->stream, ctx->d_acct, nb, ea.as<uint32_t>(),
+                           dl_a.as<sre_account_delta>(), n_acct, ed.as<uint32_t>(),
+                           (sre_account_entry *)new_acct);
+    HIP_CHECK(ctx, hipGetLastError());
+    HIP_CHECK(ctx, sdel.alloc((Ndel ? Ndel : 1) * 32));
+    if (n_acct)
+        hipLaunchKernelGGL(k_ovl_gather_deleted, dim3(grid_for(n_acct)),
+                           dim3(BLOCK), 0, ctx->stream,
+                           dl_a.as<sre_account_delta>(), n_acct,
+                           edel.as<uint32_t>(), sdel.as<uint8_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+
+    // ---- storage ----
+    HIP_CHECK(ctx, sa.alloc((ns + 1) * 4));
+    HIP_CHECK(ctx, sd.alloc((n_st + 1) * 4));
+    hipLaunchKernelGGL(k_ovl_base_st_flags, dim3(grid_for(ns + 1)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_st, ns, dl_s.as<sre_storage_entry>(),
+                       n_st, sdel.as<uint8_t>(), (uint64_t)Ndel,
+                       sa.as<uint32_t>());
+    hipLaunchKernelGGL(k_ovl_delta_st_flags, dim3(grid_for(n_st + 1)), dim3(BLOCK),
+                       0, ctx->stream, dl_s.as<sre_storage_entry>(), n_st,
+                       sdel.as<uint8_t>(), (uint64_t)Ndel, sd.as<uint32_t>(),
+                       err.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    uint32_t Sk = 0, Tk = 0;
+    DBuf esa(ctx), esd(ctx);
+    HIP_CHECK(ctx, esa.alloc((ns + 1) * 4));
+    HIP_CHECK(ctx, esd.alloc((n_st + 1) * 4));
+    if (scan_u32(ctx, sa.as<uint32_t>(), esa.as<uint32_t>(), ns + 1, &Sk))
+        return -1;
+    if (scan_u32(ctx, sd.as<uint32_t>(), esd.as<uint32_t>(), n_st + 1, &Tk))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>())) {
+        (void)hipFree(new_acct);
+        return -1;
+    }
+    uint64_t new_ns = Sk + Tk;
+    void *new_st = nullptr;
+    HIP_CHECK(ctx, hipMalloc(&new_st,
+                             (new_ns ? new_ns : 1) * sizeof(sre_storage_entry)));
+    gmax = ns > n_st ? ns : n_st;
+    if (gmax)
+        hipLaunchKernelGGL(k_ovl_scatter_st, dim3(grid_for(gmax)), dim3(BLOCK), 0,
+                           ctx->stream, ctx->d_st, ns, esa.as<uint32_t>(),
+                           dl_s.as<sre_storage_entry>(), n_st, esd.as<uint32_t>(),
+                           (sre_storage_entry *)new_st);
+    HIP_CHECK(ctx, hipGetLastError());
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+
+    // adopt the merged state
+    if (ctx->own_acct && ctx->d_acct)
+        (void)hipFree((void *)ctx->d_acct);
+    if (ctx->own_st && ctx->d_st)
+        (void)hipFree((void *)ctx->d_st);
+    ctx->d_acct = (const sre_account_entry *)new_acct;
+    ctx->na = new_na;
+    ctx->own_acct = true;
+    ctx->d_st = (const sre_storage_entry *)new_st;
+    ctx->ns = new_ns;
+    ctx->own_st = true;
+    return 0;
 }
 
 extern "C" int sre_root_with_updates(sre_ctx *ctx, uint8_t out_root[32])

@@ -34,6 +34,17 @@ STORAGE_DTYPE = np.dtype([
     ("value", np.uint8, 32),
 ])
 
+# sre_account_delta (include/sre.h): one overlay-delta account row
+DELTA_DTYPE = np.dtype([
+    ("key", np.uint8, 32),
+    ("nonce", np.uint64),
+    ("balance", np.uint8, 32),
+    ("code_hash", np.uint8, 32),
+    ("deleted", np.uint8),
+    ("pad", np.uint8, 7),
+])
+assert DELTA_DTYPE.itemsize == 112
+
 # sre_update_row (include/sre.h): one stored BranchNodeCompact + its path
 UPDATE_DTYPE = np.dtype([
     ("acct_key", np.uint8, 32),
@@ -149,6 +160,17 @@ class StateRootEngine:
         self._check(self._lib.sre_set_storage_device(
             ctypes.c_void_p(self._ctx), ctypes.c_void_p(st_u8.data_ptr()),
             st_u8.shape[0]))
+
+    def apply_delta(self, acct_delta: np.ndarray, st_delta: np.ndarray):
+        """Apply a HashedPostState overlay delta to the resident state
+        (post-state wins, zero value deletes a slot, deleted accounts wipe
+        their storage). A following root() computes the post-delta root —
+        the incremental-root entry (BASELINE configs[4])."""
+        assert acct_delta.dtype == DELTA_DTYPE
+        assert st_delta.dtype == STORAGE_DTYPE
+        self._check(self._lib.sre_apply_delta(
+            ctypes.c_void_p(self._ctx), _np_ptr(acct_delta), len(acct_delta),
+            _np_ptr(st_delta), len(st_delta)))
 
     # ---- compute ----
     def root(self) -> bytes:
