@@ -50,7 +50,13 @@ def main():
     ap.add_argument("--no-cpu-baseline", action="store_true")
     ap.add_argument("--check", action="store_true",
                     help="extra property check: sharded composition == root")
+    ap.add_argument("--incremental", action="store_true",
+                    help="BASELINE configs[4]: N-account resident base + "
+                         "delta-accounts overlay delta per step (N=1 only)")
+    ap.add_argument("--delta-accounts", type=int, default=5000)
     args = ap.parse_args()
+    if args.incremental:
+        args.slots = 0  # accounts-only base (see config note in the output)
 
     import torch
 
@@ -92,7 +98,54 @@ def main():
     # own allocations (records, level buffers) don't OOM at the 10M x 64 size
     torch.cuda.empty_cache()
 
+    # ---- incremental mode (configs[4]): build the idempotent overlay delta
+    # (absolute-valued modifications / inserts / deletes => every step
+    # produces the same post-delta root, so the determinism check holds)
+    delta = None
+    if args.incremental:
+        if world > 1:
+            print("--incremental is single-GPU (BASELINE configs[4])",
+                  file=sys.stderr)
+            sys.exit(2)
+        import numpy as np
+        from reth_amd.engine import DELTA_DTYPE, STORAGE_DTYPE
+        nd = args.delta_accounts
+        n_mod, n_del = (nd * 4) // 5, nd // 10
+        n_new = nd - n_mod - n_del
+        counters = list(range(n_mod))             + list(range(args.accounts, args.accounts + n_new))             + list(range(n_mod, n_mod + n_del))
+        msgs = torch.zeros((len(counters), 8), dtype=torch.uint8,
+                           device=f"cuda:{local_rank}")
+        for b in range(8):
+            msgs[:, b] = torch.tensor([(c >> (8 * b)) & 0xFF for c in counters],
+                                      dtype=torch.uint8)
+        dk = torch.empty((len(counters), 32), dtype=torch.uint8,
+                         device=f"cuda:{local_rank}")
+        eng.keccak_batch_device(msgs, 8, dk)
+        dk = dk.cpu().numpy()
+        ke = bytes.fromhex("c5d2460186f7233c927e7db2dcc703c0e500b653"
+                           "ca82273b7bfad8045d85a470")
+        rows = []
+        for i, c in enumerate(counters):
+            key = dk[i].tobytes()
+            if i < n_mod:            # absolute-valued modification
+                rows.append((key, c & 0xFFFF, 10**18 + c, ke, 0))
+            elif i < n_mod + n_new:  # insert
+                rows.append((key, 1, 5 * 10**17 + c, ke, 0))
+            else:                    # delete
+                rows.append((key, 0, 0, ke, 1))
+        d = np.zeros(len(rows), dtype=DELTA_DTYPE)
+        for i, (k, nn, b, ch, dead) in enumerate(sorted(rows)):
+            d[i]["key"] = np.frombuffer(k, np.uint8)
+            d[i]["nonce"] = nn
+            d[i]["balance"] = np.frombuffer(b.to_bytes(32, "big"), np.uint8)
+            d[i]["code_hash"] = np.frombuffer(ch, np.uint8)
+            d[i]["deleted"] = dead
+        delta = (d, np.zeros(0, dtype=STORAGE_DTYPE))
+
     def step():
+        if args.incremental:
+            eng.apply_delta(*delta)
+            return eng.root()
         if world > 1:
             refs, lens, roots, counts = eng.subtree_roots()
             m = sharding.all_gather_combine(refs, lens, roots, counts,
@@ -190,8 +243,14 @@ def main():
         "dtype": "u8",
         "data": "synthetic (SURVEY.md §8d, seed 0x5EED, generated on-device)",
         "config": {
-            "workload": f"{args.accounts} accounts x {args.slots} slots "
-                        "(BASELINE configs[3] shape; full job on every N)",
+            "workload": (f"incremental: {args.accounts}-account resident "
+                         f"base + {args.delta_accounts}-account overlay "
+                         "delta per step (BASELINE configs[4]; accounts-only "
+                         "base; apply_delta + full device recompute — "
+                         "dirty-path reuse is round-2 work)")
+            if args.incremental else
+            (f"{args.accounts} accounts x {args.slots} slots "
+             "(BASELINE configs[3] shape; full job on every N)"),
             "accounts": args.accounts,
             "slots_per_account": args.slots,
             "storage_leaves": int(total_storage_leaves),
