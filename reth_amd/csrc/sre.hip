@@ -1561,6 +1561,10 @@ struct sre_ctx {
     const sre_storage_entry *d_st = nullptr;
     uint64_t ns = 0;
     bool own_st = false;
+    // nonzero when the owned input array came from the pool (apply_delta
+    // replaces the state every call; pooling avoids a 21 GB hipMalloc+Free
+    // per delta at the 200M-account config)
+    size_t acct_pool_bytes = 0, st_pool_bytes = 0;
     sre_stats stats{};
     // TrieUpdates retention (sre_root_with_updates)
     bool retain_updates = false;
@@ -1601,6 +1605,30 @@ static void pool_put(sre_ctx *ctx, size_t bytes, void *p)
 {
     if (p)
         ctx->pool.emplace_back(pool_class(bytes), p);
+}
+
+static void release_acct(sre_ctx *ctx)
+{
+    if (ctx->own_acct && ctx->d_acct) {
+        if (ctx->acct_pool_bytes)
+            pool_put(ctx, ctx->acct_pool_bytes, (void *)ctx->d_acct);
+        else
+            (void)hipFree((void *)ctx->d_acct);
+    }
+    ctx->d_acct = nullptr;
+    ctx->acct_pool_bytes = 0;
+}
+
+static void release_st(sre_ctx *ctx)
+{
+    if (ctx->own_st && ctx->d_st) {
+        if (ctx->st_pool_bytes)
+            pool_put(ctx, ctx->st_pool_bytes, (void *)ctx->d_st);
+        else
+            (void)hipFree((void *)ctx->d_st);
+    }
+    ctx->d_st = nullptr;
+    ctx->st_pool_bytes = 0;
 }
 
 static std::string g_err;
@@ -1645,13 +1673,11 @@ extern "C" void sre_destroy(sre_ctx *ctx)
 {
     if (!ctx)
         return;
+    release_acct(ctx);
+    release_st(ctx);
     for (auto &e : ctx->pool)
         (void)hipFree(e.second);
     ctx->pool.clear();
-    if (ctx->own_acct && ctx->d_acct)
-        hipFree((void *)ctx->d_acct);
-    if (ctx->own_st && ctx->d_st)
-        hipFree((void *)ctx->d_st);
     if (ctx->stream)
         hipStreamDestroy(ctx->stream);
     delete ctx;
@@ -1661,9 +1687,7 @@ extern "C" int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entrie
                                    uint64_t n)
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
-    if (ctx->own_acct && ctx->d_acct)
-        hipFree((void *)ctx->d_acct);
-    ctx->d_acct = nullptr;
+    release_acct(ctx);
     void *p = nullptr;
     if (n) {
         HIP_CHECK(ctx, hipMalloc(&p, n * sizeof(sre_account_entry)));
@@ -1680,9 +1704,7 @@ extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries
                                   uint64_t n)
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
-    if (ctx->own_st && ctx->d_st)
-        hipFree((void *)ctx->d_st);
-    ctx->d_st = nullptr;
+    release_st(ctx);
     void *p = nullptr;
     if (n) {
         HIP_CHECK(ctx, hipMalloc(&p, n * sizeof(sre_storage_entry)));
@@ -1698,8 +1720,7 @@ extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries
 // Borrow device-resident inputs (zero-copy; caller keeps them alive).
 extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint64_t n)
 {
-    if (ctx->own_acct && ctx->d_acct)
-        hipFree((void *)ctx->d_acct);
+    release_acct(ctx);
     ctx->d_acct = (const sre_account_entry *)d_entries;
     ctx->na = n;
     ctx->own_acct = false;
@@ -1708,8 +1729,7 @@ extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint
 
 extern "C" int sre_set_storage_device(sre_ctx *ctx, const void *d_entries, uint64_t n)
 {
-    if (ctx->own_st && ctx->d_st)
-        hipFree((void *)ctx->d_st);
+    release_st(ctx);
     ctx->d_st = (const sre_storage_entry *)d_entries;
     ctx->ns = n;
     ctx->own_st = false;
@@ -2508,9 +2528,12 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
     if (check_err(ctx, err.as<uint32_t>()))
         return -1;
     uint64_t new_na = Bk + Dk;
-    void *new_acct = nullptr;
-    HIP_CHECK(ctx, hipMalloc(&new_acct,
-                             (new_na ? new_na : 1) * sizeof(sre_account_entry)));
+    size_t acct_bytes = (new_na ? new_na : 1) * sizeof(sre_account_entry);
+    void *new_acct = pool_get(ctx, acct_bytes);
+    if (!new_acct) {
+        set_err(ctx, "apply_delta: out of memory (accounts)");
+        return -1;
+    }
     uint64_t gmax = nb > n_acct ? nb : n_acct;
     if (gmax)
         hipLaunchKernelGGL(k_ovl_scatter_acct, dim3(grid_for(gmax)), dim3(BLOCK), 0,
@@ -2547,13 +2570,17 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
     if (scan_u32(ctx, sd.as<uint32_t>(), esd.as<uint32_t>(), n_st + 1, &Tk))
         return -1;
     if (check_err(ctx, err.as<uint32_t>())) {
-        (void)hipFree(new_acct);
+        pool_put(ctx, acct_bytes, new_acct);
         return -1;
     }
     uint64_t new_ns = Sk + Tk;
-    void *new_st = nullptr;
-    HIP_CHECK(ctx, hipMalloc(&new_st,
-                             (new_ns ? new_ns : 1) * sizeof(sre_storage_entry)));
+    size_t st_bytes = (new_ns ? new_ns : 1) * sizeof(sre_storage_entry);
+    void *new_st = pool_get(ctx, st_bytes);
+    if (!new_st) {
+        pool_put(ctx, acct_bytes, new_acct);
+        set_err(ctx, "apply_delta: out of memory (storage)");
+        return -1;
+    }
     gmax = ns > n_st ? ns : n_st;
     if (gmax)
         hipLaunchKernelGGL(k_ovl_scatter_st, dim3(grid_for(gmax)), dim3(BLOCK), 0,
@@ -2563,17 +2590,17 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
     HIP_CHECK(ctx, hipGetLastError());
     HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
 
-    // adopt the merged state
-    if (ctx->own_acct && ctx->d_acct)
-        (void)hipFree((void *)ctx->d_acct);
-    if (ctx->own_st && ctx->d_st)
-        (void)hipFree((void *)ctx->d_st);
+    // adopt the merged state (pool-backed; reused across repeated deltas)
+    release_acct(ctx);
+    release_st(ctx);
     ctx->d_acct = (const sre_account_entry *)new_acct;
     ctx->na = new_na;
     ctx->own_acct = true;
+    ctx->acct_pool_bytes = acct_bytes;
     ctx->d_st = (const sre_storage_entry *)new_st;
     ctx->ns = new_ns;
     ctx->own_st = true;
+    ctx->st_pool_bytes = st_bytes;
     return 0;
 }
 
