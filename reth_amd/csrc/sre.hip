@@ -2601,6 +2601,14 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
     ctx->ns = new_ns;
     ctx->own_st = true;
     ctx->st_pool_bytes = st_bytes;
+    // prewarm the ping-pong partner buffers: the NEXT apply_delta would
+    // otherwise pay a multi-hundred-ms first-time hipMalloc of this size
+    // class inside the caller's timed region
+    for (size_t b : {acct_bytes, st_bytes}) {
+        void *spare = pool_get(ctx, b);
+        if (spare)
+            pool_put(ctx, b, spare);
+    }
     return 0;
 }
 
