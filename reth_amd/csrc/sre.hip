@@ -537,7 +537,9 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
     const uint8_t *__restrict__ storage_roots, const int8_t *__restrict__ lcp,
     int subtree, node_rec *__restrict__ recs, uint8_t *__restrict__ depths,
     uint32_t *__restrict__ hist, uint8_t *__restrict__ roots,
-    uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens)
+    uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
+    const uint8_t *__restrict__ cell_dirty /* incremental: null = all */,
+    const uint8_t *__restrict__ covered /* incremental: seeded positions */)
 {
     __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_ACC + 66 * 4];
     uint32_t *hist_l = (uint32_t *)(lds + BLOCK * SLOT_ACC);
@@ -546,7 +548,16 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
     __syncthreads();
 
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i < na) {
+    bool active = i < na;
+    if (active && cell_dirty) {
+        uint32_t cell = ((uint32_t)acct[i].key[0] << 12) |
+                        ((uint32_t)acct[i].key[1] << 4) |
+                        ((uint32_t)acct[i].key[2] >> 4);
+        // recompute leaves of dirty cells and of positions no seeded
+        // cell-top interval covers; clean covered cells keep their seeds
+        active = cell_dirty[cell] != 0 || covered[i] == 0;
+    }
+    if (active) {
         uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_ACC;
         uint64_t *slot64 = (uint64_t *)slot;
 
@@ -598,8 +609,9 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
                     slot[p++] = bal[32 - blen + k];
             }
             slot[p++] = 0xa0;
-            for (int k = 0; k < 32; ++k)
-                slot[p++] = storage_roots[32 * i + k];
+            for (int k = 0; k < 32; ++k) // null => accounts-only state
+                slot[p++] = storage_roots ? storage_roots[32 * i + k]
+                                          : D_EMPTY_ROOT[k];
             slot[p++] = 0xa0;
             for (int k = 0; k < 32; ++k)
                 slot[p++] = acct[i].code_hash[k];
@@ -632,7 +644,7 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
             }
             memcpy(roots + 32ull * r.seg, hash, 32);
         }
-    }
+    } // active
     __syncthreads();
     if (threadIdx.x < 66 && hist_l[threadIdx.x])
         atomicAdd(&hist[threadIdx.x], hist_l[threadIdx.x]);
@@ -1275,6 +1287,26 @@ __device__ __forceinline__ uint64_t lb_keys(const uint8_t *base, uint64_t stride
     return lo;
 }
 
+// old->new position map for interval rebasing: map[i] = the new index of
+// base entry i (or of its successor when i was dropped); map[nb] = new_na.
+__global__ void k_ovl_posmap(const sre_account_entry *__restrict__ base,
+                             uint64_t nb, const uint32_t *__restrict__ bexcl,
+                             const sre_account_delta *__restrict__ dl,
+                             uint64_t nd, const uint32_t *__restrict__ dexcl,
+                             uint32_t *__restrict__ map)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i > nb)
+        return;
+    if (i == nb) {
+        map[i] = bexcl[nb] + dexcl[nd];
+        return;
+    }
+    uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_account_delta), nd,
+                         base[i].key, 32);
+    map[i] = bexcl[i] + dexcl[p];
+}
+
 // base account survives iff its key is absent from the delta
 __global__ void k_ovl_base_acct_flags(const sre_account_entry *__restrict__ base,
                                       uint64_t nb,
@@ -1418,6 +1450,119 @@ __global__ void k_ovl_scatter_st(const sre_storage_entry *__restrict__ base,
                              dl[j].acct_key, 64);
         out[dexcl[j] + bexcl[p]] = dl[j];
     }
+}
+
+// ---------------------------------------------------------------------------
+// dirty-path incremental: cell-top capture (CELL_NIBBLES-prefix cells)
+// ---------------------------------------------------------------------------
+
+#define CELL_NIBBLES 5
+#define N_CELLS (1u << (4 * CELL_NIBBLES))
+
+struct cap_row {
+    node_rec rec;    // interval, parent depth, ref — as of capture
+    uint32_t cell;   // 5-nibble prefix of the cell's keys
+    uint32_t pad_[3];
+};
+static_assert(sizeof(cap_row) == 64, "cap_row must be 64 bytes");
+
+__device__ __forceinline__ uint32_t cell_of_key(const uint8_t *key)
+{
+    return ((uint32_t)key[0] << 12) | ((uint32_t)key[1] << 4) |
+           ((uint32_t)key[2] >> 4);
+}
+
+// append every level-input record (the active nodes consumed at this level)
+// to the capture buffer: at levels < CELL_NIBBLES these are exactly the
+// cell-top nodes.
+__global__ void k_capture_L(const node_rec *__restrict__ L, uint64_t n,
+                            const uint8_t *__restrict__ keys,
+                            uint64_t key_stride, cap_row *__restrict__ out,
+                            uint32_t *__restrict__ cnt, uint64_t capacity,
+                            uint32_t *__restrict__ err)
+{
+    __shared__ uint32_t lds[BLOCK];
+    __shared__ uint32_t base;
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    bool act = j < n;
+    uint32_t excl, total;
+    block_scan(lds, act ? 1u : 0u, &excl, &total);
+    if (threadIdx.x == 0)
+        base = total ? atomicAdd(cnt, total) : 0;
+    __syncthreads();
+    if (!act)
+        return;
+    uint64_t slot = base + excl;
+    if (slot >= capacity) {
+        atomicOr(err, 1u << E_INTERNAL);
+        return;
+    }
+    cap_row r;
+    copy_rec(&r.rec, &L[j]);
+    r.cell = cell_of_key(keys + (uint64_t)L[j].s * key_stride);
+    r.pad_[0] = r.pad_[1] = r.pad_[2] = 0;
+    out[slot] = r;
+}
+
+// Revalidate retained cell-top rows against the merged state. Every row is
+// a depth-(CELL_NIBBLES-1) record: its keys share >= CELL_NIBBLES nibbles,
+// so any delta key inside (or joining) its interval lands in the SAME cell
+// and k_mark_delta_cells already dirtied it. A clean cell therefore has
+// unchanged interval content; only the junction can move, which the fresh
+// boundary-lcp check catches. Rebase: ns = map[s]; ne = map[e-1]+1 — the
+// latter excludes keys inserted between the interval and its right
+// neighbour (they share < CELL_NIBBLES nibbles, i.e. live outside the
+// subtree). Valid rows are seeded (recs[ns]+depths[ns]+hist) and mark
+// covered[ns..ne); invalid rows dirty their cell so their leaves rebuild.
+__global__ void k_revalidate_rows(const cap_row *__restrict__ rows, uint64_t n,
+                                  const uint32_t *__restrict__ old2new,
+                                  const int8_t *__restrict__ lcp,
+                                  uint8_t *__restrict__ cell_dirty,
+                                  node_rec *__restrict__ recs,
+                                  uint8_t *__restrict__ depths,
+                                  uint8_t *__restrict__ covered,
+                                  uint32_t *__restrict__ hist)
+{
+    __shared__ uint32_t hist_l[66];
+    if (threadIdx.x < 66)
+        hist_l[threadIdx.x] = 0;
+    __syncthreads();
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j < n) {
+        cap_row r = rows[j];
+        // one captured row per cell (disjoint): no other thread reads or
+        // writes this cell's flag concurrently
+        if (cell_dirty[r.cell] == 0) {
+            uint32_t ns_ = old2new[r.rec.s];
+            uint32_t ne_ = old2new[r.rec.e - 1] + 1;
+            int8_t l0 = lcp[ns_], l1 = lcp[ne_];
+            int8_t nd = l0 > l1 ? l0 : l1;
+            if (nd == r.rec.depth) {
+                r.rec.s = ns_;
+                r.rec.e = ne_;
+                copy_rec(&recs[ns_], &r.rec);
+                depths[ns_] = (uint8_t)(nd + 1);
+                atomicAdd(&hist_l[nd + 1], 1u);
+                for (uint32_t p = ns_; p < ne_; ++p)
+                    covered[p] = 1;
+            } else {
+                cell_dirty[r.cell] = 1; // junction moved: rebuild the cell
+            }
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x < 66 && hist_l[threadIdx.x])
+        atomicAdd(&hist[threadIdx.x], hist_l[threadIdx.x]);
+}
+
+// mark cells touched by delta keys
+__global__ void k_mark_delta_cells(const sre_account_delta *__restrict__ dl,
+                                   uint64_t nd, uint8_t *__restrict__ cell_dirty)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= nd)
+        return;
+    cell_dirty[cell_of_key(dl[j].key)] = 1;
 }
 
 // ---------------------------------------------------------------------------
@@ -1569,6 +1714,14 @@ struct sre_ctx {
     // TrieUpdates retention (sre_root_with_updates)
     bool retain_updates = false;
     std::vector<sre_update_row> updates;
+    // Dirty-path incremental retention (sre_root_retaining /
+    // sre_incremental_root): "cell-top" node records — the unique active
+    // node of each 5-nibble key-prefix cell, captured from the level inputs
+    // at depths < CELL_NIBBLES during a full account pass.
+    bool retain_cells = false;   // capture on the next root computation
+    bool cells_valid = false;    // retained rows match the resident state
+    void *d_cap_rows = nullptr;  // cap_row[cap_count]
+    uint64_t cap_count = 0, cap_capacity = 0;
     // size-class buffer pool: the level machinery allocates/frees dozens of
     // transient arrays per level; hipMalloc latency would dominate small
     // jobs. Freed buffers are cached by power-of-2 class and reused (also
@@ -1675,6 +1828,8 @@ extern "C" void sre_destroy(sre_ctx *ctx)
         return;
     release_acct(ctx);
     release_st(ctx);
+    if (ctx->d_cap_rows)
+        (void)hipFree(ctx->d_cap_rows);
     for (auto &e : ctx->pool)
         (void)hipFree(e.second);
     ctx->pool.clear();
@@ -1686,6 +1841,7 @@ extern "C" void sre_destroy(sre_ctx *ctx)
 extern "C" int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entries,
                                    uint64_t n)
 {
+    ctx->cells_valid = false;
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     release_acct(ctx);
     void *p = nullptr;
@@ -1703,6 +1859,7 @@ extern "C" int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entrie
 extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries,
                                   uint64_t n)
 {
+    ctx->cells_valid = false;
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     release_st(ctx);
     void *p = nullptr;
@@ -1720,6 +1877,7 @@ extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries
 // Borrow device-resident inputs (zero-copy; caller keeps them alive).
 extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint64_t n)
 {
+    ctx->cells_valid = false;
     release_acct(ctx);
     ctx->d_acct = (const sre_account_entry *)d_entries;
     ctx->na = n;
@@ -1729,6 +1887,7 @@ extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint
 
 extern "C" int sre_set_storage_device(sre_ctx *ctx, const void *d_entries, uint64_t n)
 {
+    ctx->cells_valid = false;
     release_st(ctx);
     ctx->d_st = (const sre_storage_entry *)d_entries;
     ctx->ns = n;
@@ -1854,7 +2013,14 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                       const int8_t *d_lcp, const uint8_t *d_keys, uint64_t key_stride,
                       const uint32_t *hist_host, int subtree, uint8_t *d_seg_roots,
                       uint8_t *d_child_refs, uint8_t *d_child_lens, uint32_t *d_err,
-                      pass_out *po, int updates_kind, uint8_t *d_bhash)
+                      pass_out *po, int updates_kind, uint8_t *d_bhash,
+                      // cell-top capture (dirty-path incremental): when
+                      // capture_depth > 0, every level input at depth
+                      // capture_depth-1 is appended to d_cap (each such
+                      // record's keys share >= capture_depth nibbles: exactly
+                      // one cell, disjoint from every other captured row)
+                      int capture_depth = 0, cap_row *d_cap = nullptr,
+                      uint32_t *d_cap_cnt = nullptr, uint64_t cap_capacity = 0)
 {
     int maxd = -1;
     for (int d = 63; d >= 0; --d)
@@ -1936,6 +2102,12 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                carries[d].as<node_rec>(), nB, Lbuf.as<node_rec>());
             HIP_CHECK(ctx, hipGetLastError());
             L = Lbuf.as<node_rec>();
+        }
+        if (capture_depth > 0 && d == capture_depth - 1) {
+            hipLaunchKernelGGL(k_capture_L, dim3(grid_for(n_level)), dim3(BLOCK),
+                               0, ctx->stream, L, n_level, d_keys, key_stride,
+                               d_cap, d_cap_cnt, cap_capacity, d_err);
+            HIP_CHECK(ctx, hipGetLastError());
         }
         // 3. group flags + scan
         HIP_CHECK(ctx, flags.alloc(n_level * 4));
@@ -2203,7 +2375,10 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
 
 static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int subtree,
                             uint8_t *d_roots, uint8_t *d_child_refs,
-                            uint8_t *d_child_lens, pass_out *po, uint32_t *d_err)
+                            uint8_t *d_child_lens, pass_out *po, uint32_t *d_err,
+                            int capture_depth = 0, cap_row *d_cap = nullptr,
+                            uint32_t *d_cap_cnt = nullptr,
+                            uint64_t cap_capacity = 0)
 {
     uint64_t na = ctx->na;
     DBuf lcp(ctx), recs(ctx), depths(ctx), hist(ctx);
@@ -2227,7 +2402,7 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
     hipLaunchKernelGGL(k_leaf_account, dim3(grid_for(na)), dim3(BLOCK), 0, ctx->stream,
                        ctx->d_acct, na, d_storage_roots, lcp.as<int8_t>(), subtree,
                        recs.as<node_rec>(), depths.as<uint8_t>(), hist.as<uint32_t>(),
-                       d_roots, d_child_refs, d_child_lens);
+                       d_roots, d_child_refs, d_child_lens, nullptr, nullptr);
     HIP_CHECK(ctx, hipGetLastError());
     hipEventRecord(ev1, ctx->stream);
     HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
@@ -2250,7 +2425,8 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
     if (run_levels(ctx, na, recs.as<node_rec>(), depths.as<uint8_t>(),
                    lcp.as<int8_t>(), keys, sizeof(sre_account_entry), hist_host,
                    subtree, d_roots, d_child_refs, d_child_lens, d_err, po,
-                   ctx->retain_updates ? 0 : -1, bhash.as<uint8_t>()))
+                   ctx->retain_updates ? 0 : -1, bhash.as<uint8_t>(),
+                   capture_depth, d_cap, d_cap_cnt, cap_capacity))
         return -1;
     if (ctx->retain_updates) {
         for (size_t r = upd_start; r < ctx->updates.size(); ++r)
@@ -2480,10 +2656,10 @@ static bool row_less(const sre_update_row &a, const sre_update_row &b)
     return a.path_len < b.path_len;
 }
 
-extern "C" int sre_apply_delta(sre_ctx *ctx,
-                               const sre_account_delta *acct_delta,
-                               uint64_t n_acct,
-                               const sre_storage_entry *st_delta, uint64_t n_st)
+static int apply_delta_impl(sre_ctx *ctx, const sre_account_delta *acct_delta,
+                            uint64_t n_acct, const sre_storage_entry *st_delta,
+                            uint64_t n_st, DBuf *map_out /* optional: old->new
+                            positions, (old na)+1 u32 */)
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     uint64_t nb = ctx->na, ns = ctx->ns;
@@ -2541,6 +2717,14 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
                            dl_a.as<sre_account_delta>(), n_acct, ed.as<uint32_t>(),
                            (sre_account_entry *)new_acct);
     HIP_CHECK(ctx, hipGetLastError());
+    if (map_out) {
+        HIP_CHECK(ctx, map_out->alloc((nb + 1) * 4));
+        hipLaunchKernelGGL(k_ovl_posmap, dim3(grid_for(nb + 1)), dim3(BLOCK), 0,
+                           ctx->stream, ctx->d_acct, nb, ea.as<uint32_t>(),
+                           dl_a.as<sre_account_delta>(), n_acct,
+                           ed.as<uint32_t>(), map_out->as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+    }
     HIP_CHECK(ctx, sdel.alloc((Ndel ? Ndel : 1) * 32));
     if (n_acct)
         hipLaunchKernelGGL(k_ovl_gather_deleted, dim3(grid_for(n_acct)),
@@ -2609,6 +2793,203 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
         if (spare)
             pool_put(ctx, b, spare);
     }
+    return 0;
+}
+
+extern "C" int sre_apply_delta(sre_ctx *ctx,
+                               const sre_account_delta *acct_delta,
+                               uint64_t n_acct,
+                               const sre_storage_entry *st_delta, uint64_t n_st)
+{
+    ctx->cells_valid = false; // a plain apply invalidates cell retention
+    return apply_delta_impl(ctx, acct_delta, n_acct, st_delta, n_st, nullptr);
+}
+
+// Full root over an accounts-only state, retaining the cell-top records
+// for subsequent sre_incremental_root calls (dirty-path incremental).
+extern "C" int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32])
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    if (ctx->ns != 0) {
+        set_err(ctx, "sre_root_retaining: accounts-only states (v1)");
+        return -1;
+    }
+    memset(&ctx->stats, 0, sizeof(ctx->stats));
+    ctx->cells_valid = false;
+    uint64_t want_cap = 2 * (ctx->na < (uint64_t)N_CELLS ? ctx->na
+                                                         : (uint64_t)N_CELLS) +
+                        8192;
+    if (ctx->cap_capacity < want_cap) {
+        if (ctx->d_cap_rows)
+            (void)hipFree(ctx->d_cap_rows);
+        ctx->d_cap_rows = nullptr;
+        HIP_CHECK(ctx, hipMalloc(&ctx->d_cap_rows, want_cap * sizeof(cap_row)));
+        ctx->cap_capacity = want_cap;
+    }
+    if (ctx->na == 0) {
+        ctx->cap_count = 0;
+        ctx->cells_valid = true;
+        memcpy(out_root, EMPTY_ROOT_H, 32);
+        return 0;
+    }
+    DBuf err(ctx), roots(ctx), capcnt(ctx);
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, roots.alloc(32));
+    HIP_CHECK(ctx, capcnt.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(capcnt.p, 0, 4, ctx->stream));
+    pass_out po;
+    if (run_account_pass(ctx, nullptr, 0, roots.as<uint8_t>(), nullptr, nullptr,
+                         &po, err.as<uint32_t>(), CELL_NIBBLES,
+                         (cap_row *)ctx->d_cap_rows, capcnt.as<uint32_t>(),
+                         ctx->cap_capacity))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    uint32_t cnt = 0;
+    HIP_CHECK(ctx, hipMemcpy(&cnt, capcnt.p, 4, hipMemcpyDeviceToHost));
+    ctx->cap_count = cnt;
+    ctx->cells_valid = true;
+    HIP_CHECK(ctx, hipMemcpy(out_root, roots.p, 32, hipMemcpyDeviceToHost));
+    return 0;
+}
+
+// Dirty-path incremental root: apply an accounts-only overlay delta and
+// recompute only the 5-nibble cells it touches, reusing every clean
+// cell-top record (the §3b walker-skip semantics expressed in this
+// engine's level machinery). Requires a prior sre_root_retaining.
+extern "C" int sre_incremental_root(sre_ctx *ctx,
+                                    const sre_account_delta *acct_delta,
+                                    uint64_t n_delta, uint8_t out_root[32])
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    if (!ctx->cells_valid || ctx->ns != 0) {
+        set_err(ctx, "sre_incremental_root: needs sre_root_retaining on an "
+                     "accounts-only state first");
+        return -1;
+    }
+    memset(&ctx->stats, 0, sizeof(ctx->stats));
+    hipEvent_t t0, t1;
+    hipEventCreate(&t0);
+    hipEventCreate(&t1);
+    hipEventRecord(t0, ctx->stream);
+
+    DBuf map(ctx);
+    if (apply_delta_impl(ctx, acct_delta, n_delta, nullptr, 0, &map))
+        return -1;
+    uint64_t na = ctx->na;
+    if (na == 0) {
+        ctx->cells_valid = false;
+        memcpy(out_root, EMPTY_ROOT_H, 32);
+        return 0;
+    }
+    DBuf err(ctx), bitmap(ctx), dl(ctx), lcp(ctx), recs(ctx), depths(ctx),
+        hist(ctx), roots(ctx), capcnt(ctx);
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, bitmap.alloc(N_CELLS));
+    HIP_CHECK(ctx, hipMemsetAsync(bitmap.p, 0, N_CELLS, ctx->stream));
+    HIP_CHECK(ctx, dl.alloc((n_delta ? n_delta : 1) * sizeof(sre_account_delta)));
+    if (n_delta) {
+        HIP_CHECK(ctx, hipMemcpyAsync(dl.p, acct_delta,
+                                      n_delta * sizeof(sre_account_delta),
+                                      hipMemcpyHostToDevice, ctx->stream));
+        hipLaunchKernelGGL(k_mark_delta_cells, dim3(grid_for(n_delta)),
+                           dim3(BLOCK), 0, ctx->stream,
+                           dl.as<sre_account_delta>(), n_delta,
+                           bitmap.as<uint8_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+    }
+    HIP_CHECK(ctx, lcp.alloc(na + 1));
+    hipLaunchKernelGGL(k_lcp_account, dim3(grid_for(na + 1)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_acct, na, 0, lcp.as<int8_t>(),
+                       err.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    DBuf covered(ctx);
+    HIP_CHECK(ctx, recs.alloc(na * sizeof(node_rec)));
+    HIP_CHECK(ctx, depths.alloc(na));
+    HIP_CHECK(ctx, covered.alloc(na));
+    HIP_CHECK(ctx, hist.alloc(66 * 4));
+    HIP_CHECK(ctx, roots.alloc(32));
+    HIP_CHECK(ctx, hipMemsetAsync(depths.p, 0xFF, na, ctx->stream));
+    HIP_CHECK(ctx, hipMemsetAsync(covered.p, 0, na, ctx->stream));
+    HIP_CHECK(ctx, hipMemsetAsync(hist.p, 0, 66 * 4, ctx->stream));
+    // seed clean cell-tops; anything invalid dirties its cell
+    if (ctx->cap_count)
+        hipLaunchKernelGGL(k_revalidate_rows, dim3(grid_for(ctx->cap_count)),
+                           dim3(BLOCK), 0, ctx->stream,
+                           (const cap_row *)ctx->d_cap_rows, ctx->cap_count,
+                           map.as<uint32_t>(), lcp.as<int8_t>(),
+                           bitmap.as<uint8_t>(), recs.as<node_rec>(),
+                           depths.as<uint8_t>(), covered.as<uint8_t>(),
+                           hist.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    // rehash leaves of dirty cells and of positions no seed covers
+    pass_out po;
+    hipEvent_t ev0, ev1;
+    hipEventCreate(&ev0);
+    hipEventCreate(&ev1);
+    hipEventRecord(ev0, ctx->stream);
+    hipLaunchKernelGGL(k_leaf_account, dim3(grid_for(na)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_acct, na, nullptr, lcp.as<int8_t>(),
+                       0, recs.as<node_rec>(), depths.as<uint8_t>(),
+                       hist.as<uint32_t>(), roots.as<uint8_t>(), nullptr,
+                       nullptr, bitmap.as<uint8_t>(), covered.as<uint8_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    hipEventRecord(ev1, ctx->stream);
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    po.leaf_ms += ms;
+    hipEventDestroy(ev0);
+    hipEventDestroy(ev1);
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    uint32_t hist_host[66];
+    HIP_CHECK(ctx, hipMemcpy(hist_host, hist.p, 66 * 4, hipMemcpyDeviceToHost));
+    // fresh capture for the next delta; the retained rows were consumed by
+    // k_revalidate_rows above, so capturing over the same buffer is safe —
+    // unless na grew past its capacity, in which case swap in a bigger one.
+    uint64_t want_cap = 2 * (na < (uint64_t)N_CELLS ? na : (uint64_t)N_CELLS) +
+                        8192;
+    if (ctx->cap_capacity < want_cap) {
+        void *bigger = nullptr;
+        HIP_CHECK(ctx, hipMalloc(&bigger, want_cap * sizeof(cap_row)));
+        HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+        (void)hipFree(ctx->d_cap_rows);
+        ctx->d_cap_rows = bigger;
+        ctx->cap_capacity = want_cap;
+    }
+    HIP_CHECK(ctx, capcnt.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(capcnt.p, 0, 4, ctx->stream));
+    const uint8_t *keys =
+        (const uint8_t *)ctx->d_acct + offsetof(sre_account_entry, key);
+    if (run_levels(ctx, na, recs.as<node_rec>(), depths.as<uint8_t>(),
+                   lcp.as<int8_t>(), keys, sizeof(sre_account_entry), hist_host,
+                   0, roots.as<uint8_t>(), nullptr, nullptr, err.as<uint32_t>(),
+                   &po, -1, nullptr, CELL_NIBBLES, (cap_row *)ctx->d_cap_rows,
+                   capcnt.as<uint32_t>(), ctx->cap_capacity))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    uint32_t cnt = 0;
+    HIP_CHECK(ctx, hipMemcpy(&cnt, capcnt.p, 4, hipMemcpyDeviceToHost));
+    ctx->cap_count = cnt;
+    ctx->cells_valid = true;
+
+    hipEventRecord(t1, ctx->stream);
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    float total = 0;
+    hipEventElapsedTime(&total, t0, t1);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    ctx->stats.total_ms = total;
+    ctx->stats.leaf_hash_ms = po.leaf_ms;
+    ctx->stats.branch_hash_ms = po.branch_ms;
+    ctx->stats.branch_count = po.branch_count;
+    ctx->stats.levels = po.levels;
+
+    HIP_CHECK(ctx, hipMemcpy(out_root, roots.p, 32, hipMemcpyDeviceToHost));
     return 0;
 }
 

@@ -173,6 +173,27 @@ class StateRootEngine:
             _np_ptr(st_delta), len(st_delta)))
 
     # ---- compute ----
+    def root_retaining(self) -> bytes:
+        """Full root that also retains cell-top trie records so following
+        incremental_root() calls recompute only dirty 5-nibble cells
+        (accounts-only states). See include/sre.h sre_root_retaining."""
+        out = (ctypes.c_uint8 * 32)()
+        self._check(self._lib.sre_root_retaining(
+            ctypes.c_void_p(self._ctx), out))
+        return bytes(out)
+
+    def incremental_root(self, acct_delta: np.ndarray) -> bytes:
+        """Apply an accounts-only overlay delta and recompute the root along
+        dirty paths only (requires a prior root_retaining). The resident
+        state becomes the merged result and retention is refreshed, so
+        deltas chain."""
+        assert acct_delta.dtype == DELTA_DTYPE
+        out = (ctypes.c_uint8 * 32)()
+        self._check(self._lib.sre_incremental_root(
+            ctypes.c_void_p(self._ctx), _np_ptr(acct_delta),
+            len(acct_delta), out))
+        return bytes(out)
+
     def root(self) -> bytes:
         out = (ctypes.c_uint8 * 32)()
         self._check(self._lib.sre_root(ctypes.c_void_p(self._ctx), out))

@@ -136,3 +136,168 @@ def test_overlay_rejects_slot_for_deleted_account(eng):
     s[0]["value"] = np.frombuffer((1).to_bytes(32, "big"), np.uint8)
     with pytest.raises(RuntimeError):
         eng.apply_delta(d, s)
+
+
+# ---------------------------------------------------------------------------
+# Dirty-path incremental (sre_root_retaining / sre_incremental_root):
+# recompute only the 5-nibble cells a delta touches, reusing retained
+# cell-top records. Parity vs the CPU oracle on dict-merged state across
+# chained randomized delta sequences.
+# ---------------------------------------------------------------------------
+
+def _acct_dict(acct):
+    return {bytes(a["key"]): [int(a["nonce"]),
+                              int.from_bytes(bytes(a["balance"]), "big"),
+                              bytes(a["code_hash"]), {}] for a in acct}
+
+
+def _apply_dict(accounts, rows):
+    for (k, n, b, ch, dead) in rows:
+        if dead:
+            accounts.pop(k, None)
+        else:
+            accounts[k] = [n, b, ch, {}]
+
+
+def _want_root(accounts):
+    return bind.state_root(*_arrays_of(accounts))
+
+
+def test_incremental_chained_random_deltas(eng):
+    rng = np.random.default_rng(1234)
+    acct, _ = gen.gen_state_numpy(5000, 0, bind.keccak256_batch)
+    accounts = _acct_dict(acct)
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    assert eng.root_retaining() == _want_root(accounts)
+    ke = bind.keccak256(b"")
+    for step in range(6):
+        keys = sorted(accounts)
+        rows = []
+        # modify a random subset
+        for i in rng.choice(len(keys), size=40, replace=False):
+            k = keys[int(i)]
+            v = accounts[k]
+            rows.append((k, v[0] + 1, v[1] + 7, v[2], 0))
+        # delete a random subset (disjoint keys)
+        dels = rng.choice(len(keys), size=15, replace=False)
+        for i in dels:
+            k = keys[int(i)]
+            if any(r[0] == k for r in rows):
+                continue
+            rows.append((k, 0, 0, ke, 1))
+        # insert fresh accounts
+        for i in range(20):
+            nk = bind.keccak256(b"inc" + bytes([step, i]))
+            rows.append((nk, 3, 1000 + step * 100 + i, ke, 0))
+        rows = sorted(set(rows))
+        _apply_dict(accounts, rows)
+        d, _ = _mk_delta(rows, [])
+        assert eng.incremental_root(d) == _want_root(accounts), f"step {step}"
+
+
+def test_incremental_adjacent_keys_same_cell(eng):
+    # keys sharing long prefixes stress cell-boundary logic: several keys in
+    # ONE 5-nibble cell, deltas that split/merge nodes inside it, plus a
+    # neighbour cell that must be reused untouched.
+    ke = bind.keccak256(b"")
+    base = bytearray(bind.keccak256(b"cellbase"))
+    keys = []
+    for i in range(8):
+        k = bytearray(base)
+        k[31] = i  # same first 62 nibbles
+        keys.append(bytes(k))
+    other = bytearray(base)
+    other[2] ^= 0x40  # different cell (3rd nibble differs)
+    keys.append(bytes(other))
+    accounts = {k: [1, 10 + i, ke, {}] for i, k in enumerate(sorted(keys))}
+    eng.upload(*_arrays_of(accounts))
+    assert eng.root_retaining() == _want_root(accounts)
+    # delete half of the clustered keys, modify one, insert one deeper twin
+    rows = []
+    for k in sorted(keys)[:4]:
+        rows.append((k, 0, 0, ke, 1))
+    k5 = sorted(keys)[5]
+    rows.append((k5, 9, 9999, ke, 0))
+    twin = bytearray(base)
+    twin[31] = 0xEE
+    rows.append((bytes(twin), 2, 5, ke, 0))
+    rows = sorted(rows)
+    _apply_dict(accounts, rows)
+    d, _ = _mk_delta(rows, [])
+    assert eng.incremental_root(d) == _want_root(accounts)
+    # second chained delta: re-insert one deleted key
+    back = sorted(keys)[0]
+    rows2 = [(back, 4, 44, ke, 0)]
+    _apply_dict(accounts, rows2)
+    d2, _ = _mk_delta(rows2, [])
+    assert eng.incremental_root(d2) == _want_root(accounts)
+
+
+def test_incremental_edge_cases(eng):
+    ke = bind.keccak256(b"")
+    acct, _ = gen.gen_state_numpy(300, 0, bind.keccak256_batch)
+    accounts = _acct_dict(acct)
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    r0 = eng.root_retaining()
+    assert r0 == _want_root(accounts)
+    # empty delta: same root
+    assert eng.incremental_root(np.zeros(0, DELTA_DTYPE)) == r0
+    # idempotent delta (rewrite same values)
+    keys = sorted(accounts)
+    rows = [(k, accounts[k][0], accounts[k][1], accounts[k][2], 0)
+            for k in keys[:10]]
+    d, _ = _mk_delta(rows, [])
+    assert eng.incremental_root(d) == r0
+    # delete everything -> EMPTY_ROOT, then refill from empty
+    rows = [(k, 0, 0, ke, 1) for k in keys]
+    d, _ = _mk_delta(rows, [])
+    empty_root = bytes.fromhex(
+        "56e81f171bcc55a6ff8345e692c0f86e5b48e01b996cadc001622fb5e363b421")
+    assert eng.incremental_root(d) == empty_root
+    # retention invalidated at na==0; re-arm and continue
+    assert eng.root_retaining() == empty_root
+    rows = [(bind.keccak256(b"solo"), 1, 2, ke, 0)]
+    d, _ = _mk_delta(rows, [])
+    accounts = {rows[0][0]: [1, 2, ke, {}]}
+    assert eng.incremental_root(d) == _want_root(accounts)
+    # single-account state (root is a leaf — coverage-hole fallback path)
+    rows2 = [(bind.keccak256(b"solo2"), 7, 8, ke, 0)]
+    accounts[rows2[0][0]] = [7, 8, ke, {}]
+    d2, _ = _mk_delta(rows2, [])
+    assert eng.incremental_root(d2) == _want_root(accounts)
+
+
+def test_incremental_requires_retention(eng):
+    acct, _ = gen.gen_state_numpy(100, 0, bind.keccak256_batch)
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    # no root_retaining yet -> error (upload invalidates retention)
+    with pytest.raises(RuntimeError):
+        eng.incremental_root(np.zeros(0, DELTA_DTYPE))
+    # storage present -> root_retaining refuses (v1 accounts-only)
+    acct2, st2 = gen.gen_state_numpy(100, 2, bind.keccak256_batch)
+    eng.upload(acct2, st2)
+    with pytest.raises(RuntimeError):
+        eng.root_retaining()
+
+
+def test_incremental_matches_apply_delta_path(eng):
+    # same delta through (apply_delta + root) and incremental_root must agree
+    rng = np.random.default_rng(77)
+    acct, _ = gen.gen_state_numpy(2000, 0, bind.keccak256_batch)
+    ke = bind.keccak256(b"")
+    keys = sorted(bytes(a["key"]) for a in acct)
+    rows = []
+    for i in rng.choice(len(keys), size=25, replace=False):
+        rows.append((keys[int(i)], 42, 4242, ke, 0))
+    for i in range(5):
+        rows.append((bind.keccak256(b"cmp" + bytes([i])), 1, i, ke, 0))
+    rows = sorted(set(rows))
+    d, _ = _mk_delta(rows, [])
+
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    eng.apply_delta(d, np.zeros(0, bind.STORAGE_DTYPE))
+    want = eng.root()
+
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    eng.root_retaining()
+    assert eng.incremental_root(d) == want
