@@ -246,8 +246,10 @@ def main():
             "workload": (f"incremental: {args.accounts}-account resident "
                          f"base + {args.delta_accounts}-account overlay "
                          "delta per step (BASELINE configs[4]; accounts-only "
-                         "base; apply_delta + full device recompute — "
-                         "dirty-path reuse is round-2 work)")
+                         "base; " +
+                         ("dirty-path recompute via retained cell tops"
+                          if args.dirty else
+                          "apply_delta + full device recompute") + ")")
             if args.incremental else
             (f"{args.accounts} accounts x {args.slots} slots "
              "(BASELINE configs[3] shape; full job on every N)"),
