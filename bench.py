@@ -54,6 +54,9 @@ def main():
                     help="BASELINE configs[4]: N-account resident base + "
                          "delta-accounts overlay delta per step (N=1 only)")
     ap.add_argument("--delta-accounts", type=int, default=5000)
+    ap.add_argument("--dirty", action="store_true",
+                    help="with --incremental: dirty-path recompute "
+                         "(sre_incremental_root) instead of merge+full root")
     args = ap.parse_args()
     if args.incremental:
         args.slots = 0  # accounts-only base (see config note in the output)
@@ -143,6 +146,8 @@ def main():
         delta = (d, np.zeros(0, dtype=STORAGE_DTYPE))
 
     def step():
+        if args.incremental and args.dirty:
+            return eng.incremental_root(delta[0])
         if args.incremental:
             eng.apply_delta(*delta)
             return eng.root()
@@ -159,6 +164,8 @@ def main():
         torch.cuda.synchronize()
 
     # ---- warmup ----
+    if args.incremental and args.dirty:
+        eng.root_retaining()  # arm cell-top retention (untimed, once)
     root0 = None
     for _ in range(max(args.warmup, 1)):
         root0 = step()
