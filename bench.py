@@ -278,6 +278,14 @@ def main():
                     "kernel time. The kernel is integer-VALU bound, not "
                     "HBM bound: chip Keccak-f ceiling ~13.6 GH/s at 78.6 "
                     "Tops/s u32 VALU peak (DESIGN.md §roofline).",
+        } if not (args.incremental and args.dirty) else {
+            "bound": "hbm", "achieved": None, "peak": 8000.0,
+            "unit": "GB/s", "frac": None, "traffic": None,
+            "note": "dirty-path incremental: the leaf kernel touches only "
+                    "delta-dirty cells, so the full-job algorithmic-bytes "
+                    "formula does not apply; dominant cost is the O(na) "
+                    "merge/lcp/revalidate scans (see engine_stats and the "
+                    "full-rebuild bench line for the kernel roofline).",
         },
         "cpu_baseline": cpu_baseline,
         "keccak_ghs_leaf_kernel": round(keccak_ghs, 2),
