@@ -837,6 +837,15 @@ __global__ void k_group_flags(const node_rec *__restrict__ L, uint64_t n,
 // ---------------------------------------------------------------------------
 
 #define SLOT_BR 552 // >= 4*136 zero-padded branch RLP slot in global scratch
+// Branch scratch layout: 1 = row-major (each group's padded RLP in its own
+// 576-B row, 9 cache lines, per-lane dense stores/loads served by L2
+// locality), 0 = column-major words (per-instruction coalescing, but
+// divergent flush timing scatters it). Measured choice — see profiles/.
+#ifndef SRE_SCRATCH_ROWMAJOR
+#define SRE_SCRATCH_ROWMAJOR 1
+#endif
+#define SLOT_BR_ROW 576 // row stride, 64-B aligned
+
 
 // per-group metadata produced by the assemble kernel
 struct br_meta {
@@ -885,11 +894,19 @@ struct byte_appender {
         stride = s;
         g = gg;
     }
+    __device__ __forceinline__ uint64_t *slot64(int w)
+    {
+#if SRE_SCRATCH_ROWMAJOR
+        return base + (uint64_t)g * (SLOT_BR_ROW / 8) + w;
+#else
+        return base + (uint64_t)w * stride + g;
+#endif
+    }
     __device__ __forceinline__ void put(uint8_t v)
     {
         cur |= (uint64_t)v << (8 * pos);
         if (++pos == 8) {
-            base[(uint64_t)widx * stride + g] = cur;
+            *slot64(widx) = cur;
             widx++;
             pos = 0;
             cur = 0;
@@ -901,27 +918,77 @@ struct byte_appender {
     {
         int last = nb * 17 - 1;
         while (widx < last) {
-            base[(uint64_t)widx * stride + g] = cur;
+            *slot64(widx) = cur;
             widx++;
             cur = 0;
             pos = 0;
         }
         cur |= 0x8000000000000000ULL;
-        base[(uint64_t)last * stride + g] = cur;
+        *slot64(last) = cur;
     }
 };
+
+// Branch-size class of a group with nmem members: guaranteed RLP block
+// count 1/2/3/4 (payload <= 1 + (16-nmem) + 33*nmem, so nmem<=3 -> <=115 B
+// -> 1 block, <=7 -> 2, <=11 -> 3, else 4). Groups are partitioned by class
+// per chunk so every wave of k_branch_assemble / k_branch_hash is
+// block-count uniform: the hash absorb runs lane-sum (not wave-max) keccak,
+// and assemble's column-major flush addresses stay within one class's word
+// range instead of scattering over 0..68.
+__device__ __forceinline__ int nmem_class(uint32_t nmem)
+{
+    return nmem <= 3 ? 0 : nmem <= 7 ? 1 : nmem <= 11 ? 2 : 3;
+}
+
+#define CLS_BLOCK 256
+// per-block class histogram, laid out cnts[c*nblk + b] so ONE exclusive
+// scan over 4*nblk yields every (class, block) partition offset
+__global__ void k_class_hist(const uint32_t *__restrict__ gs, uint32_t n_groups,
+                             uint32_t nblk, uint32_t *__restrict__ cnts)
+{
+    __shared__ uint32_t c_l[4];
+    if (threadIdx.x < 4)
+        c_l[threadIdx.x] = 0;
+    __syncthreads();
+    uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g < n_groups)
+        atomicAdd(&c_l[nmem_class(gs[g + 1] - gs[g])], 1u);
+    __syncthreads();
+    if (threadIdx.x < 4)
+        cnts[threadIdx.x * nblk + blockIdx.x] = c_l[threadIdx.x];
+}
+
+// scatter: perm[offs[c*nblk + b] + rank] = g (rank via LDS bump; stability
+// is unnecessary — records are written back at their original group index)
+__global__ void k_class_scatter(const uint32_t *__restrict__ gs,
+                                uint32_t n_groups, uint32_t nblk,
+                                const uint32_t *__restrict__ offs,
+                                uint32_t *__restrict__ perm)
+{
+    __shared__ uint32_t base_l[4];
+    if (threadIdx.x < 4)
+        base_l[threadIdx.x] = offs[threadIdx.x * nblk + blockIdx.x];
+    __syncthreads();
+    uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g >= n_groups)
+        return;
+    int c = nmem_class(gs[g + 1] - gs[g]);
+    perm[atomicAdd(&base_l[c], 1u)] = g;
+}
 
 __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     const node_rec *__restrict__ L, const uint32_t *__restrict__ gs,
     uint32_t n_groups, const int8_t *__restrict__ lcp,
     const uint8_t *__restrict__ keys, uint64_t key_stride, int d,
     uint8_t *__restrict__ scratch, uint64_t scratch_stride,
-    br_meta *__restrict__ meta, uint32_t *__restrict__ err)
+    br_meta *__restrict__ meta, const uint32_t *__restrict__ perm,
+    uint32_t *__restrict__ err)
 {
     uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
     if (g >= n_groups)
         return;
-    uint64_t j = gs[g], jend = gs[g + 1];
+    uint32_t gg = perm ? perm[g] : g; // meta/scratch by thread slot g
+    uint64_t j = gs[gg], jend = gs[gg + 1];
     br_meta mt;
     mt.s = L[j].s;
     mt.e = L[jend - 1].e;
@@ -960,6 +1027,10 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     }
     int h = rlp_list_hdr_len(payload);
     int br_len = h + payload;
+    // NOTE: the keccak pad must land at the message's own rate boundary —
+    // absorbing extra zero blocks up to the class bound would change the
+    // digest, so block count stays per-lane (waves are still class-uniform
+    // in the typical all-hashed-refs case).
     int nb = br_len / 136 + 1;
 
     byte_appender ap;
@@ -1010,7 +1081,8 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     const uint8_t *__restrict__ scratch, uint64_t scratch_stride,
     const br_meta *__restrict__ meta,
     uint32_t n_groups, const uint8_t *__restrict__ keys, uint64_t key_stride,
-    int subtree, node_rec *__restrict__ out, uint8_t *__restrict__ seg_roots,
+    int subtree, node_rec *__restrict__ out, const uint32_t *__restrict__ perm,
+    uint8_t *__restrict__ seg_roots,
     uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
     uint32_t *__restrict__ pending, uint32_t *__restrict__ err,
     uint8_t *__restrict__ bhash_by_s, /* updates mode: branch hash by
@@ -1031,8 +1103,8 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     br_meta mt{};
     node_rec *r = nullptr;
     if (active) {
-        mt = meta[g];
-        r = &out[g]; // write the record in place (no scratch spill)
+        mt = meta[g]; // meta/scratch by thread slot; record by group index
+        r = &out[perm ? perm[g] : g];
         r->s = mt.s;
         r->e = mt.e;
         r->seg = mt.seg;
@@ -1053,10 +1125,17 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
 #pragma unroll
         for (int i = 0; i < 25; ++i)
             s[i] = 0;
+#if SRE_SCRATCH_ROWMAJOR
+        const uint64_t *row = scr64 + (uint64_t)g * (SLOT_BR_ROW / 8);
+#endif
         for (int blk = 0; blk < nblocks; ++blk) {
 #pragma unroll
             for (int i = 0; i < 17; ++i)
+#if SRE_SCRATCH_ROWMAJOR
+                s[i] ^= row[blk * 17 + i];
+#else
                 s[i] ^= scr64[(uint64_t)(blk * 17 + i) * scratch_stride + g];
+#endif
             keccak_f(s);
         }
 #pragma unroll
@@ -1070,8 +1149,13 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
 #pragma unroll
         for (int k = 0; k < 31; ++k)
             if (k < br_ref_len)
-                br_ref[k] = (uint8_t)(scr64[(uint64_t)(k >> 3) * scratch_stride + g]
+    #if SRE_SCRATCH_ROWMAJOR
+            br_ref[k] = (uint8_t)(scr64[(uint64_t)g * (SLOT_BR_ROW / 8) + (k >> 3)]
                                       >> (8 * (k & 7)));
+#else
+            br_ref[k] = (uint8_t)(scr64[(uint64_t)(k >> 3) * scratch_stride + g]
+                                      >> (8 * (k & 7)));
+#endif
     } else {
         br_ref_len = 33;
         br_ref[0] = 0xa0;
@@ -1180,7 +1264,8 @@ __global__ void __launch_bounds__(BLOCK) k_emit_updates(
     const uint8_t *__restrict__ keys, uint64_t key_stride,
     const uint8_t *__restrict__ bhash_by_s, int kind,
     sre_update_row *__restrict__ rows, uint32_t *__restrict__ row_counter,
-    uint32_t *__restrict__ rowidx /* per group: row slot or ~0 */)
+    const uint32_t *__restrict__ perm,
+    uint32_t *__restrict__ rowidx /* per thread slot: row slot or ~0 */)
 {
     __shared__ uint32_t lds[BLOCK];
     __shared__ uint32_t base;
@@ -1197,7 +1282,8 @@ __global__ void __launch_bounds__(BLOCK) k_emit_updates(
         return;
     br_meta mt = meta[g];
     sre_update_row *row = &rows[base + excl];
-    uint64_t j = gs[g], jend = gs[g + 1];
+    uint32_t gg = perm ? perm[g] : g;
+    uint64_t j = gs[gg], jend = gs[gg + 1];
     uint16_t state = 0, tree = 0, hashm = 0;
     int nh = 0;
     for (uint64_t m = j; m < jend; ++m) {
@@ -2125,7 +2211,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         HIP_CHECK(ctx, newn.alloc((uint64_t)n_groups * sizeof(node_rec)));
         HIP_CHECK(ctx, gs.alloc(((uint64_t)n_groups + 1) * 4));
         uint64_t chunk = n_groups < BR_CHUNK ? n_groups : BR_CHUNK;
-        HIP_CHECK(ctx, scratch.alloc(chunk * SLOT_BR));
+        HIP_CHECK(ctx, scratch.alloc(chunk * (SRE_SCRATCH_ROWMAJOR
+                                              ? SLOT_BR_ROW : SLOT_BR)));
         HIP_CHECK(ctx, meta.alloc(chunk * sizeof(br_meta)));
         if (updates_kind >= 0) {
             // per-level: capacity must cover THIS level's chunk
@@ -2139,14 +2226,41 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         uint32_t n_level32 = (uint32_t)n_level;
         HIP_CHECK(ctx, hipMemcpyAsync(gs.as<uint32_t>() + n_groups, &n_level32, 4,
                                       hipMemcpyHostToDevice, ctx->stream));
+        // class partition buffers (see nmem_class): one perm per chunk makes
+        // assemble/hash waves block-count uniform
+        DBuf perm(ctx), ccnt(ctx), coff(ctx);
+        const uint32_t CLS_MIN = 1u << 14; // below this the win is noise
+        if (n_groups >= CLS_MIN) {
+            HIP_CHECK(ctx, perm.alloc(chunk * 4));
+            uint32_t nblk_max = (uint32_t)((chunk + CLS_BLOCK - 1) / CLS_BLOCK);
+            HIP_CHECK(ctx, ccnt.alloc((uint64_t)4 * nblk_max * 4));
+            HIP_CHECK(ctx, coff.alloc((uint64_t)4 * nblk_max * 4));
+        }
         hipEventRecord(ev0, ctx->stream);
         for (uint64_t g0 = 0; g0 < n_groups; g0 += chunk) {
             uint32_t gc = (uint32_t)(n_groups - g0 < chunk ? n_groups - g0 : chunk);
+            uint32_t *d_perm = nullptr;
+            if (n_groups >= CLS_MIN) {
+                d_perm = perm.as<uint32_t>();
+                uint32_t nblk = (gc + CLS_BLOCK - 1) / CLS_BLOCK;
+                hipLaunchKernelGGL(k_class_hist, dim3(nblk), dim3(CLS_BLOCK), 0,
+                                   ctx->stream, gs.as<uint32_t>() + g0, gc, nblk,
+                                   ccnt.as<uint32_t>());
+                HIP_CHECK(ctx, hipGetLastError());
+                uint32_t tot = 0;
+                if (scan_u32(ctx, ccnt.as<uint32_t>(), coff.as<uint32_t>(),
+                             (uint64_t)4 * nblk, &tot))
+                    return -1;
+                hipLaunchKernelGGL(k_class_scatter, dim3(nblk), dim3(CLS_BLOCK),
+                                   0, ctx->stream, gs.as<uint32_t>() + g0, gc,
+                                   nblk, coff.as<uint32_t>(), d_perm);
+                HIP_CHECK(ctx, hipGetLastError());
+            }
             hipLaunchKernelGGL(k_branch_assemble,
                                dim3((gc + BLOCK_A - 1) / BLOCK_A), dim3(BLOCK_A),
                                0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
                                d_lcp, d_keys, key_stride, d, scratch.as<uint8_t>(),
-                               chunk, meta.as<br_meta>(), d_err);
+                               chunk, meta.as<br_meta>(), d_perm, d_err);
             HIP_CHECK(ctx, hipGetLastError());
             if (updates_kind >= 0) {
                 // emit BEFORE hashing: children's bhash entries must not yet
@@ -2156,14 +2270,16 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                    0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
                                    meta.as<br_meta>(), d_keys, key_stride, d_bhash,
                                    updates_kind, urows.as<sre_update_row>(),
-                                   urow_cnt.as<uint32_t>(), urowidx.as<uint32_t>());
+                                   urow_cnt.as<uint32_t>(), d_perm,
+                                   urowidx.as<uint32_t>());
                 HIP_CHECK(ctx, hipGetLastError());
             }
             hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(gc)), dim3(BLOCK), 0,
                                ctx->stream, scratch.as<uint8_t>(), chunk,
                                meta.as<br_meta>(),
                                gc, d_keys, key_stride, subtree,
-                               newn.as<node_rec>() + g0, d_seg_roots, d_child_refs,
+                               newn.as<node_rec>() + g0, d_perm,
+                               d_seg_roots, d_child_refs,
                                d_child_lens, pend.as<uint32_t>(), d_err, d_bhash,
                                updates_kind >= 0 ? urows.as<sre_update_row>()
                                                  : nullptr,
