@@ -842,7 +842,7 @@ __global__ void k_group_flags(const node_rec *__restrict__ L, uint64_t n,
 // locality), 0 = column-major words (per-instruction coalescing, but
 // divergent flush timing scatters it). Measured choice — see profiles/.
 #ifndef SRE_SCRATCH_ROWMAJOR
-#define SRE_SCRATCH_ROWMAJOR 1
+#define SRE_SCRATCH_ROWMAJOR 0 // measured: col 444.7 ms vs row 456.6 ms @10Mx64
 #endif
 #define SLOT_BR_ROW 576 // row stride, 64-B aligned
 
