@@ -674,6 +674,62 @@ __device__ __forceinline__ void block_scan(uint32_t *lds, uint32_t v,
     __syncthreads();
 }
 
+// One-shot stable 66-way bucket of leaf records by parent depth, replacing
+// the per-level k_sel_count/k_sel_gather scans: every record is moved ONCE
+// into depth-major order (doff from the host-side hist prefix), after which
+// each level's fresh-leaf input is a contiguous slice. Stability (position
+// order within a depth) comes from wave-ballot ranks + per-wave LDS
+// offsets + the [depth][block] scan order. Records with depths > 65 (dead /
+// non-seeded incremental positions) are skipped.
+__global__ void k_depth_hist66(const uint8_t *__restrict__ depths, uint64_t n,
+                               uint32_t nblk, uint32_t *__restrict__ cnts)
+{
+    __shared__ uint32_t c_l[66];
+    if (threadIdx.x < 66)
+        c_l[threadIdx.x] = 0;
+    __syncthreads();
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n && depths[i] <= 65)
+        atomicAdd(&c_l[depths[i]], 1u);
+    __syncthreads();
+    if (threadIdx.x < 66)
+        cnts[(uint64_t)threadIdx.x * nblk + blockIdx.x] = c_l[threadIdx.x];
+}
+
+__global__ void k_depth_scatter66(const uint8_t *__restrict__ depths,
+                                  const node_rec *__restrict__ recs, uint64_t n,
+                                  uint32_t nblk,
+                                  const uint32_t *__restrict__ offs,
+                                  node_rec *__restrict__ out)
+{
+    __shared__ uint32_t wh[66][BLOCK / 64]; // per-wave per-depth counts
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    if (threadIdx.x < 66)
+#pragma unroll
+        for (int w = 0; w < BLOCK / 64; ++w)
+            wh[threadIdx.x][w] = 0;
+    __syncthreads();
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int key = (i < n) ? depths[i] : 255;
+    uint32_t rank_in_wave = 0;
+    // 66 uniform ballots: each lane keeps the rank for its own key
+    for (int k = 0; k < 66; ++k) {
+        uint64_t m = __ballot(key == k);
+        if (key == k)
+            rank_in_wave = __popcll(m & ((1ull << lane) - 1));
+        if (lane == 0 && m)
+            wh[k][wid] = (uint32_t)__popcll(m);
+    }
+    __syncthreads();
+    if (key > 65)
+        return;
+    uint32_t before = 0;
+    for (int w = 0; w < wid; ++w)
+        before += wh[key][w];
+    uint32_t pos = offs[(uint64_t)key * nblk + blockIdx.x] + before + rank_in_wave;
+    copy_rec(&out[pos], &recs[i]);
+}
+
 __global__ void k_sel_count(const uint8_t *__restrict__ depths, uint64_t n,
                             uint32_t want, uint32_t *__restrict__ block_cnt)
 {
@@ -1785,6 +1841,7 @@ __global__ void k_scan_add(uint32_t *__restrict__ out, uint64_t n,
 struct sre_ctx {
     int device = 0;
     hipStream_t stream = nullptr;
+    hipStream_t stream2 = nullptr; // branch-assemble pipeline stage
     std::string err;
     const sre_account_entry *d_acct = nullptr;
     uint64_t na = 0;
@@ -1900,7 +1957,8 @@ extern "C" sre_ctx *sre_create(int device)
     }
     sre_ctx *ctx = new sre_ctx();
     ctx->device = device;
-    if (hipStreamCreate(&ctx->stream) != hipSuccess) {
+    if (hipStreamCreate(&ctx->stream) != hipSuccess ||
+        hipStreamCreate(&ctx->stream2) != hipSuccess) {
         set_err(nullptr, "sre_create: hipStreamCreate failed");
         delete ctx;
         return nullptr;
@@ -1921,6 +1979,8 @@ extern "C" void sre_destroy(sre_ctx *ctx)
     ctx->pool.clear();
     if (ctx->stream)
         hipStreamDestroy(ctx->stream);
+    if (ctx->stream2)
+        hipStreamDestroy(ctx->stream2);
     delete ctx;
 }
 
@@ -2146,6 +2206,32 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
     hipEventCreate(&ev0);
     hipEventCreate(&ev1);
 
+    // 1. (once) stable depth-major bucket of the leaf records: doff[v] =
+    // start of the depth-v slice; every level's fresh input is then a slice.
+    DBuf dsorted(ctx);
+    uint64_t doff[67];
+    {
+        HIP_CHECK(ctx, dsorted.alloc(n * sizeof(node_rec)));
+        uint32_t nblk = (uint32_t)((n + BLOCK - 1) / BLOCK);
+        DBuf dc(ctx), do_(ctx);
+        HIP_CHECK(ctx, dc.alloc((uint64_t)66 * nblk * 4));
+        HIP_CHECK(ctx, do_.alloc((uint64_t)66 * nblk * 4));
+        hipLaunchKernelGGL(k_depth_hist66, dim3(nblk), dim3(BLOCK), 0,
+                           ctx->stream, d_depths, n, nblk, dc.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        uint32_t tot = 0;
+        if (scan_u32(ctx, dc.as<uint32_t>(), do_.as<uint32_t>(),
+                     (uint64_t)66 * nblk, &tot))
+            return -1;
+        hipLaunchKernelGGL(k_depth_scatter66, dim3(nblk), dim3(BLOCK), 0,
+                           ctx->stream, d_depths, d_recs, n, nblk,
+                           do_.as<uint32_t>(), dsorted.as<node_rec>());
+        HIP_CHECK(ctx, hipGetLastError());
+        doff[0] = 0;
+        for (int v = 0; v < 66; ++v)
+            doff[v + 1] = doff[v] + hist_host[v];
+    }
+
     for (int d = maxd; d >= 0; --d) {
         uint64_t nA = hist_host[d + 1];
         uint64_t nB = carry_cnt[d];
@@ -2154,37 +2240,21 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         po->levels++;
         uint64_t n_level = nA + nB;
 
-        // 1. select leaves with parent depth == d
-        if (nA) {
-            HIP_CHECK(ctx, Lsel.alloc(nA * sizeof(node_rec)));
-            uint32_t g = sel_grid_for(n);
-            hipLaunchKernelGGL(k_sel_count, dim3(g), dim3(BLOCK), 0, ctx->stream,
-                               d_depths, n, (uint32_t)(d + 1), blk_a.as<uint32_t>());
-            uint32_t tot = 0;
-            if (scan_u32(ctx, blk_a.as<uint32_t>(), off_a.as<uint32_t>(), g, &tot))
-                return -1;
-            if (tot != nA) {
-                set_err(ctx, "internal: leaf histogram mismatch");
-                return -1;
-            }
-            hipLaunchKernelGGL(k_sel_gather, dim3(g), dim3(BLOCK), 0, ctx->stream,
-                               d_depths, d_recs, n, (uint32_t)(d + 1),
-                               off_a.as<uint32_t>(), Lsel.as<node_rec>());
-            HIP_CHECK(ctx, hipGetLastError());
-        }
+        // 1. this level's fresh leaves: the depth-(d+1) slice of dsorted
+        node_rec *Lslice = dsorted.as<node_rec>() + doff[d + 1];
         // 2. merge leaves with this depth's carry into the level input L
         node_rec *L;
         if (nB == 0) {
-            L = Lsel.as<node_rec>();
+            L = Lslice;
         } else if (nA == 0) {
             L = carries[d].as<node_rec>();
         } else {
             HIP_CHECK(ctx, Lbuf.alloc(n_level * sizeof(node_rec)));
             hipLaunchKernelGGL(k_merge_a, dim3(grid_for(nA)), dim3(BLOCK), 0,
-                               ctx->stream, Lsel.as<node_rec>(), nA,
+                               ctx->stream, Lslice, nA,
                                carries[d].as<node_rec>(), nB, Lbuf.as<node_rec>());
             hipLaunchKernelGGL(k_merge_b, dim3(grid_for(nB)), dim3(BLOCK), 0,
-                               ctx->stream, Lsel.as<node_rec>(), nA,
+                               ctx->stream, Lslice, nA,
                                carries[d].as<node_rec>(), nB, Lbuf.as<node_rec>());
             HIP_CHECK(ctx, hipGetLastError());
             L = Lbuf.as<node_rec>();
@@ -2226,22 +2296,22 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         uint32_t n_level32 = (uint32_t)n_level;
         HIP_CHECK(ctx, hipMemcpyAsync(gs.as<uint32_t>() + n_groups, &n_level32, 4,
                                       hipMemcpyHostToDevice, ctx->stream));
-        // class partition buffers (see nmem_class): one perm per chunk makes
-        // assemble/hash waves block-count uniform
-        DBuf perm(ctx), ccnt(ctx), coff(ctx);
+        // class partition (see nmem_class): one perm per chunk makes
+        // assemble/hash waves block-count uniform. All chunks' perms are
+        // built upfront on the main stream; the per-chunk assemble (memory
+        // heavy, stream2) is then pipelined against the hash (VALU heavy,
+        // main stream) with ping-pong scratch/meta buffers.
+        DBuf perm(ctx), ccnt(ctx), coff(ctx), scratch2(ctx), meta2(ctx);
         const uint32_t CLS_MIN = 1u << 14; // below this the win is noise
-        if (n_groups >= CLS_MIN) {
-            HIP_CHECK(ctx, perm.alloc(chunk * 4));
+        bool use_cls = n_groups >= CLS_MIN;
+        if (use_cls) {
+            HIP_CHECK(ctx, perm.alloc((uint64_t)n_groups * 4));
             uint32_t nblk_max = (uint32_t)((chunk + CLS_BLOCK - 1) / CLS_BLOCK);
             HIP_CHECK(ctx, ccnt.alloc((uint64_t)4 * nblk_max * 4));
             HIP_CHECK(ctx, coff.alloc((uint64_t)4 * nblk_max * 4));
-        }
-        hipEventRecord(ev0, ctx->stream);
-        for (uint64_t g0 = 0; g0 < n_groups; g0 += chunk) {
-            uint32_t gc = (uint32_t)(n_groups - g0 < chunk ? n_groups - g0 : chunk);
-            uint32_t *d_perm = nullptr;
-            if (n_groups >= CLS_MIN) {
-                d_perm = perm.as<uint32_t>();
+            for (uint64_t g0 = 0; g0 < n_groups; g0 += chunk) {
+                uint32_t gc = (uint32_t)(n_groups - g0 < chunk ? n_groups - g0
+                                                               : chunk);
                 uint32_t nblk = (gc + CLS_BLOCK - 1) / CLS_BLOCK;
                 hipLaunchKernelGGL(k_class_hist, dim3(nblk), dim3(CLS_BLOCK), 0,
                                    ctx->stream, gs.as<uint32_t>() + g0, gc, nblk,
@@ -2253,30 +2323,64 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                     return -1;
                 hipLaunchKernelGGL(k_class_scatter, dim3(nblk), dim3(CLS_BLOCK),
                                    0, ctx->stream, gs.as<uint32_t>() + g0, gc,
-                                   nblk, coff.as<uint32_t>(), d_perm);
+                                   nblk, coff.as<uint32_t>(),
+                                   perm.as<uint32_t>() + g0);
                 HIP_CHECK(ctx, hipGetLastError());
             }
+        }
+        bool pipe2 = n_groups > chunk; // >1 chunk: overlap pays for 2nd buf
+        if (pipe2) {
+            HIP_CHECK(ctx, scratch2.alloc(chunk * (SRE_SCRATCH_ROWMAJOR
+                                                   ? SLOT_BR_ROW : SLOT_BR)));
+            HIP_CHECK(ctx, meta2.alloc(chunk * sizeof(br_meta)));
+        }
+        hipEvent_t ev_asm[2], ev_hash[2];
+        for (int b = 0; b < 2; ++b) {
+            hipEventCreateWithFlags(&ev_asm[b], hipEventDisableTiming);
+            hipEventCreateWithFlags(&ev_hash[b], hipEventDisableTiming);
+        }
+        // stream2 must not outrun state main-stream work this level depends
+        // on (L, gs, perm are ready once the partition above completes)
+        hipEventRecord(ev_hash[0], ctx->stream);
+        hipEventRecord(ev_hash[1], ctx->stream);
+        hipStreamWaitEvent(ctx->stream2, ev_hash[0], 0);
+        int chunk_i = 0;
+        hipEventRecord(ev0, ctx->stream);
+        for (uint64_t g0 = 0; g0 < n_groups; g0 += chunk, ++chunk_i) {
+            uint32_t gc = (uint32_t)(n_groups - g0 < chunk ? n_groups - g0 : chunk);
+            int buf = chunk_i & 1;
+            uint8_t *scr = (pipe2 && buf) ? scratch2.as<uint8_t>()
+                                          : scratch.as<uint8_t>();
+            br_meta *mt = (pipe2 && buf) ? meta2.as<br_meta>()
+                                         : meta.as<br_meta>();
+            uint32_t *d_perm = use_cls ? perm.as<uint32_t>() + g0 : nullptr;
+            hipStream_t s_asm = pipe2 ? ctx->stream2 : ctx->stream;
+            if (pipe2) // wait until the hash consuming this buffer finished
+                hipStreamWaitEvent(s_asm, ev_hash[buf], 0);
             hipLaunchKernelGGL(k_branch_assemble,
                                dim3((gc + BLOCK_A - 1) / BLOCK_A), dim3(BLOCK_A),
-                               0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
-                               d_lcp, d_keys, key_stride, d, scratch.as<uint8_t>(),
-                               chunk, meta.as<br_meta>(), d_perm, d_err);
+                               0, s_asm, L, gs.as<uint32_t>() + g0, gc,
+                               d_lcp, d_keys, key_stride, d, scr,
+                               chunk, mt, d_perm, d_err);
             HIP_CHECK(ctx, hipGetLastError());
+            if (pipe2) {
+                hipEventRecord(ev_asm[buf], s_asm);
+                hipStreamWaitEvent(ctx->stream, ev_asm[buf], 0);
+            }
             if (updates_kind >= 0) {
                 // emit BEFORE hashing: children's bhash entries must not yet
                 // be overwritten by this level's nodes (shared leftmost s)
                 HIP_CHECK(ctx, hipMemsetAsync(urow_cnt.p, 0, 4, ctx->stream));
                 hipLaunchKernelGGL(k_emit_updates, dim3(grid_for(gc)), dim3(BLOCK),
                                    0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
-                                   meta.as<br_meta>(), d_keys, key_stride, d_bhash,
+                                   mt, d_keys, key_stride, d_bhash,
                                    updates_kind, urows.as<sre_update_row>(),
                                    urow_cnt.as<uint32_t>(), d_perm,
                                    urowidx.as<uint32_t>());
                 HIP_CHECK(ctx, hipGetLastError());
             }
             hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(gc)), dim3(BLOCK), 0,
-                               ctx->stream, scratch.as<uint8_t>(), chunk,
-                               meta.as<br_meta>(),
+                               ctx->stream, scr, chunk, mt,
                                gc, d_keys, key_stride, subtree,
                                newn.as<node_rec>() + g0, d_perm,
                                d_seg_roots, d_child_refs,
@@ -2286,6 +2390,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                updates_kind >= 0 ? urowidx.as<uint32_t>()
                                                  : nullptr);
             HIP_CHECK(ctx, hipGetLastError());
+            if (pipe2)
+                hipEventRecord(ev_hash[buf], ctx->stream);
             if (updates_kind >= 0) {
                 uint32_t nrows = 0;
                 HIP_CHECK(ctx, hipMemcpy(&nrows, urow_cnt.p, 4,
@@ -2299,6 +2405,10 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                              hipMemcpyDeviceToHost));
                 }
             }
+        }
+        for (int b = 0; b < 2; ++b) {
+            hipEventDestroy(ev_asm[b]);
+            hipEventDestroy(ev_hash[b]);
         }
         hipEventRecord(ev1, ctx->stream);
         HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
