@@ -730,109 +730,60 @@ __global__ void k_depth_scatter66(const uint8_t *__restrict__ depths,
     copy_rec(&out[pos], &recs[i]);
 }
 
-__global__ void k_sel_count(const uint8_t *__restrict__ depths, uint64_t n,
-                            uint32_t want, uint32_t *__restrict__ block_cnt)
+
+
+// depth-major bucket of branch OUTPUT records (key = rec.depth + 1, always
+// 0..64 here; bucket 0 = segment roots, left in place and never
+// distributed). Same stable ballot-rank scheme as k_depth_scatter66.
+__global__ void k_depth_hist66_rec(const node_rec *__restrict__ recs,
+                                   uint64_t n, uint32_t nblk,
+                                   uint32_t *__restrict__ cnts)
 {
-    __shared__ uint32_t lds[BLOCK];
-    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SEL_ITEMS +
-                    (uint64_t)threadIdx.x * SEL_ITEMS;
-    uint32_t c = 0;
-    for (int k = 0; k < SEL_ITEMS; ++k) {
-        uint64_t i = base + k;
-        if (i < n && depths[i] == want)
-            c++;
-    }
-    uint32_t excl, total;
-    block_scan(lds, c, &excl, &total);
-    if (threadIdx.x == 0)
-        block_cnt[blockIdx.x] = total;
+    __shared__ uint32_t c_l[66];
+    if (threadIdx.x < 66)
+        c_l[threadIdx.x] = 0;
+    __syncthreads();
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < n)
+        atomicAdd(&c_l[recs[i].depth + 1], 1u);
+    __syncthreads();
+    if (threadIdx.x < 66)
+        cnts[(uint64_t)threadIdx.x * nblk + blockIdx.x] = c_l[threadIdx.x];
 }
 
-__global__ void k_sel_gather(const uint8_t *__restrict__ depths,
-                             const node_rec *__restrict__ recs, uint64_t n,
-                             uint32_t want, const uint32_t *__restrict__ block_off,
-                             node_rec *__restrict__ out)
+__global__ void k_depth_scatter66_rec(const node_rec *__restrict__ recs,
+                                      uint64_t n, uint32_t nblk,
+                                      const uint32_t *__restrict__ offs,
+                                      node_rec *__restrict__ out)
 {
-    __shared__ uint32_t lds[BLOCK];
-    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SEL_ITEMS +
-                    (uint64_t)threadIdx.x * SEL_ITEMS;
-    uint32_t c = 0;
-    for (int k = 0; k < SEL_ITEMS; ++k) {
-        uint64_t i = base + k;
-        if (i < n && depths[i] == want)
-            c++;
+    __shared__ uint32_t wh[66][BLOCK / 64];
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    if (threadIdx.x < 66)
+#pragma unroll
+        for (int w = 0; w < BLOCK / 64; ++w)
+            wh[threadIdx.x][w] = 0;
+    __syncthreads();
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    int key = (i < n) ? recs[i].depth + 1 : 255;
+    uint32_t rank_in_wave = 0;
+    for (int k = 0; k < 66; ++k) {
+        uint64_t m = __ballot(key == k);
+        if (key == k)
+            rank_in_wave = __popcll(m & ((1ull << lane) - 1));
+        if (lane == 0 && m)
+            wh[k][wid] = (uint32_t)__popcll(m);
     }
-    uint32_t excl, total;
-    block_scan(lds, c, &excl, &total);
-    uint32_t pos = block_off[blockIdx.x] + excl;
-    for (int k = 0; k < SEL_ITEMS; ++k) {
-        uint64_t i = base + k;
-        if (i < n && depths[i] == want)
-            copy_rec(&out[pos++], &recs[i]);
-    }
+    __syncthreads();
+    if (key > 65)
+        return;
+    uint32_t before = 0;
+    for (int w = 0; w < wid; ++w)
+        before += wh[key][w];
+    uint32_t pos = offs[(uint64_t)key * nblk + blockIdx.x] + before + rank_in_wave;
+    copy_rec(&out[pos], &recs[i]);
 }
 
-// carry partition by record depth (sel = depth==want, rest = others)
-__global__ void k_part_count(const node_rec *__restrict__ recs, uint64_t n,
-                             int want_depth, uint32_t *__restrict__ bc_sel,
-                             uint32_t *__restrict__ bc_rest)
-{
-    __shared__ uint32_t lds[BLOCK];
-    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SEL_ITEMS +
-                    (uint64_t)threadIdx.x * SEL_ITEMS;
-    uint32_t cs = 0, cr = 0;
-    for (int k = 0; k < SEL_ITEMS; ++k) {
-        uint64_t i = base + k;
-        if (i < n) {
-            if (recs[i].depth == want_depth)
-                cs++;
-            else
-                cr++;
-        }
-    }
-    uint32_t excl, tot;
-    block_scan(lds, cs, &excl, &tot);
-    if (threadIdx.x == 0)
-        bc_sel[blockIdx.x] = tot;
-    block_scan(lds, cr, &excl, &tot);
-    if (threadIdx.x == 0)
-        bc_rest[blockIdx.x] = tot;
-}
 
-__global__ void k_part_gather(const node_rec *__restrict__ recs, uint64_t n,
-                              int want_depth, const uint32_t *__restrict__ off_sel,
-                              const uint32_t *__restrict__ off_rest,
-                              node_rec *__restrict__ out_sel,
-                              node_rec *__restrict__ out_rest)
-{
-    __shared__ uint32_t lds[BLOCK];
-    uint64_t base = (uint64_t)blockIdx.x * BLOCK * SEL_ITEMS +
-                    (uint64_t)threadIdx.x * SEL_ITEMS;
-    uint32_t cs = 0, cr = 0;
-    for (int k = 0; k < SEL_ITEMS; ++k) {
-        uint64_t i = base + k;
-        if (i < n) {
-            if (recs[i].depth == want_depth)
-                cs++;
-            else
-                cr++;
-        }
-    }
-    uint32_t es, er, tot;
-    block_scan(lds, cs, &es, &tot);
-    block_scan(lds, cr, &er, &tot);
-    uint32_t ps = off_sel[blockIdx.x] + es;
-    uint32_t pr = off_rest[blockIdx.x] + er;
-    for (int k = 0; k < SEL_ITEMS; ++k) {
-        uint64_t i = base + k;
-        if (i < n) {
-            if (recs[i].depth == want_depth)
-                copy_rec(&out_sel[ps++], &recs[i]);
-            else
-                copy_rec(&out_rest[pr++], &recs[i]);
-        }
-    }
-}
 
 // merge two record arrays sorted by .s (distinct keys)
 __global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
@@ -2187,16 +2138,10 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         carries.emplace_back(ctx);
     uint64_t carry_cnt[64] = {0};
 
-    DBuf Lsel(ctx), Lbuf(ctx), newn(ctx), trash(ctx), newp(ctx), cmerge(ctx);
+    DBuf Lbuf(ctx), newn(ctx), newp(ctx), cmerge(ctx);
     DBuf flags(ctx), gidx(ctx), pend(ctx);
     DBuf gs(ctx), scratch(ctx), meta(ctx), urows(ctx), urow_cnt(ctx),
         urowidx(ctx);
-    DBuf blk_a(ctx), blk_b(ctx), off_a(ctx), off_b(ctx);
-    uint64_t max_blocks = sel_grid_for(n) + 2;
-    HIP_CHECK(ctx, blk_a.alloc(max_blocks * 4));
-    HIP_CHECK(ctx, blk_b.alloc(max_blocks * 4));
-    HIP_CHECK(ctx, off_a.alloc(max_blocks * 4));
-    HIP_CHECK(ctx, off_b.alloc(max_blocks * 4));
     HIP_CHECK(ctx, pend.alloc(66 * 4));
     HIP_CHECK(ctx, hipMemsetAsync(pend.p, 0, 66 * 4, ctx->stream));
 
@@ -2430,34 +2375,54 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         HIP_CHECK(ctx, hipMemsetAsync((uint8_t *)pend.p + 4 * (d + 1), 0, 4,
                                       ctx->stream));
         po->branch_blocks = branch_blocks_base + pending_host[65];
+        // one stable depth-major scatter of the outputs; per-depth slices
+        // then merge (or copy) into their carries — no per-depth rescans.
+        uint64_t fresh_cnt[64] = {0};
+        uint64_t slice_off[66];
+        bool any_fresh = false;
+        {
+            uint64_t roots_n = n_groups;
+            for (int p = 0; p < d; ++p) {
+                fresh_cnt[p] = pending_host[p + 1] - prev_pending[p + 1];
+                roots_n -= fresh_cnt[p];
+                any_fresh |= fresh_cnt[p] != 0;
+            }
+            slice_off[0] = 0;
+            slice_off[1] = roots_n; // bucket 0: segment roots, stay put
+            for (int v = 1; v < 65; ++v)
+                slice_off[v + 1] = slice_off[v] + (v - 1 < d ? fresh_cnt[v - 1]
+                                                             : 0);
+        }
+        if (any_fresh) {
+            HIP_CHECK(ctx, newp.alloc((uint64_t)n_groups * sizeof(node_rec)));
+            uint32_t nblk = (uint32_t)((n_groups + BLOCK - 1) / BLOCK);
+            DBuf dc2(ctx), do2(ctx);
+            HIP_CHECK(ctx, dc2.alloc((uint64_t)66 * nblk * 4));
+            HIP_CHECK(ctx, do2.alloc((uint64_t)66 * nblk * 4));
+            hipLaunchKernelGGL(k_depth_hist66_rec, dim3(nblk), dim3(BLOCK), 0,
+                               ctx->stream, newn.as<node_rec>(), n_groups,
+                               nblk, dc2.as<uint32_t>());
+            HIP_CHECK(ctx, hipGetLastError());
+            uint32_t tot = 0;
+            if (scan_u32(ctx, dc2.as<uint32_t>(), do2.as<uint32_t>(),
+                         (uint64_t)66 * nblk, &tot))
+                return -1;
+            hipLaunchKernelGGL(k_depth_scatter66_rec, dim3(nblk), dim3(BLOCK),
+                               0, ctx->stream, newn.as<node_rec>(), n_groups,
+                               nblk, do2.as<uint32_t>(), newp.as<node_rec>());
+            HIP_CHECK(ctx, hipGetLastError());
+        }
         for (int p = d - 1; p >= 0; --p) {
-            uint64_t fresh = pending_host[p + 1] - prev_pending[p + 1];
+            uint64_t fresh = fresh_cnt[p];
             if (fresh == 0)
                 continue;
-            // extract depth-p nodes out of newn (stable -> sorted by s)
-            HIP_CHECK(ctx, newp.alloc(fresh * sizeof(node_rec)));
-            HIP_CHECK(ctx, trash.alloc((uint64_t)n_groups * sizeof(node_rec)));
-            uint32_t g = sel_grid_for(n_groups);
-            hipLaunchKernelGGL(k_part_count, dim3(g), dim3(BLOCK), 0, ctx->stream,
-                               newn.as<node_rec>(), n_groups, p,
-                               blk_a.as<uint32_t>(), blk_b.as<uint32_t>());
-            uint32_t ts = 0;
-            if (scan_u32(ctx, blk_a.as<uint32_t>(), off_a.as<uint32_t>(), g, &ts))
-                return -1;
-            uint32_t tr = 0;
-            if (scan_u32(ctx, blk_b.as<uint32_t>(), off_b.as<uint32_t>(), g, &tr))
-                return -1;
-            if (ts != fresh) {
-                set_err(ctx, "internal: fresh-node count mismatch");
-                return -1;
-            }
-            hipLaunchKernelGGL(k_part_gather, dim3(g), dim3(BLOCK), 0, ctx->stream,
-                               newn.as<node_rec>(), n_groups, p,
-                               off_a.as<uint32_t>(), off_b.as<uint32_t>(),
-                               newp.as<node_rec>(), trash.as<node_rec>());
-            HIP_CHECK(ctx, hipGetLastError());
+            node_rec *slice = newp.as<node_rec>() + slice_off[p + 1];
             if (carry_cnt[p] == 0) {
-                swap_bufs(carries[p], newp);
+                HIP_CHECK(ctx, carries[p].alloc(fresh * sizeof(node_rec)));
+                HIP_CHECK(ctx, hipMemcpyAsync(carries[p].p, slice,
+                                              fresh * sizeof(node_rec),
+                                              hipMemcpyDeviceToDevice,
+                                              ctx->stream));
                 carry_cnt[p] = fresh;
             } else {
                 uint64_t total = carry_cnt[p] + fresh;
@@ -2465,14 +2430,12 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                 hipLaunchKernelGGL(k_merge_a, dim3(grid_for(carry_cnt[p])),
                                    dim3(BLOCK), 0, ctx->stream,
                                    carries[p].as<node_rec>(), carry_cnt[p],
-                                   newp.as<node_rec>(), fresh,
-                                   cmerge.as<node_rec>());
+                                   slice, fresh, cmerge.as<node_rec>());
                 hipLaunchKernelGGL(k_merge_b, dim3(grid_for(fresh)), dim3(BLOCK), 0,
                                    ctx->stream, carries[p].as<node_rec>(),
-                                   carry_cnt[p], newp.as<node_rec>(), fresh,
+                                   carry_cnt[p], slice, fresh,
                                    cmerge.as<node_rec>());
                 HIP_CHECK(ctx, hipGetLastError());
-                HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
                 swap_bufs(carries[p], cmerge);
                 carry_cnt[p] = total;
             }
