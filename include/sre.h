@@ -190,6 +190,26 @@ int sre_incremental_root(sre_ctx *ctx,
                          const sre_account_delta *acct_delta,
                          uint64_t n_delta, uint8_t out_root[32]);
 
+/* Account multiproof — the surface of Proof::account_proof /
+ * Proof::multiproof restricted to account targets
+ * (crates/trie/trie/src/proof/mod.rs:59-137 `multiproof`, collecting the
+ * RLP of every node on each target's path root-first, the ProofRetainer
+ * semantics of the alloy-trie HashBuilder at trie.rs:292-300). v1 proves
+ * PRESENT keys only: a target absent from the uploaded state fails with an
+ * error (exclusion-proof trimming is follow-up work). Nodes are returned
+ * root-first per target, concatenated: out_nodes holds the RLP bytes
+ * back-to-back, out_lens one length per node, out_counts one node count
+ * per target (in target order). cap_* are capacities in bytes / entries;
+ * fails loudly when exceeded. Account-trie proofs never contain inline
+ * (<32 B) nodes: every account leaf RLP is >= 70 B, so every referenced
+ * node on the path is hashed. */
+int sre_account_proof(sre_ctx *ctx,
+                      const uint8_t *targets /* n x 32, hashed keys */,
+                      uint64_t n_targets,
+                      uint8_t *out_nodes, uint64_t cap_nodes,
+                      uint32_t *out_lens, uint64_t cap_lens,
+                      uint32_t *out_counts);
+
 /* Compute the state root AND retain the stored trie nodes (TrieUpdates) —
  * the surface of StateRootProvider::state_root_with_updates
  * (crates/storage/storage-api/src/trie.rs:30) / StateRoot::root_with_updates

@@ -227,3 +227,59 @@ def state_root(accounts: dict) -> bytes:
         sr = storage_root(slots)
         items[k] = account_value(nonce, balance, sr, code_hash)
     return trie_root(items)
+
+
+def _collect_proof(items, pos, target, is_root, nodes):
+    """Walk the _build recursion along `target` (nibble tuple), appending
+    every node RLP that is hash-referenced (>= 32 B) or the root — the
+    standard eth_getProof node-list semantics (the reference's
+    ProofRetainer keyed by path prefix, crates/trie/trie/src/proof/mod.rs)."""
+    rlp = _build(items, pos)
+    if len(items) == 1:
+        if is_root or len(rlp) >= 32:
+            nodes.append(rlp)
+        return
+    first, last = items[0][0], items[-1][0]
+    p = pos
+    while p < len(first) and p < len(last) and first[p] == last[p]:
+        p += 1
+    if p > pos:
+        # extension node, then its branch child
+        if is_root or len(rlp) >= 32:
+            nodes.append(rlp)
+        assert target[pos:p] == first[pos:p], "target diverges (absent key)"
+        brlp = _build(items, p)
+        if len(brlp) >= 32:
+            nodes.append(brlp)
+        _descend_branch(items, p, target, nodes)
+        return
+    if is_root or len(rlp) >= 32:
+        nodes.append(rlp)
+    _descend_branch(items, pos, target, nodes)
+
+
+def _descend_branch(items, pos, target, nodes):
+    i = 0
+    for nib in range(16):
+        j = i
+        while j < len(items) and items[j][0][pos] == nib:
+            j += 1
+        if nib == target[pos]:
+            assert j > i, "target child absent"
+            _collect_proof(items[i:j], pos + 1, target, False, nodes)
+            return
+        i = j
+    raise AssertionError("unreachable")
+
+
+def account_proof(accounts: dict, hashed_key: bytes):
+    """Proof node list (root-first RLPs) for a PRESENT hashed account key."""
+    items_d = {}
+    for k, (nonce, balance, code_hash, slots) in accounts.items():
+        sr = storage_root(slots)
+        items_d[k] = account_value(nonce, balance, sr, code_hash)
+    items = sorted((tuple(nibbles_of(k)), v) for k, v in items_d.items())
+    target = tuple(nibbles_of(hashed_key))
+    nodes = []
+    _collect_proof(items, 0, target, True, nodes)
+    return nodes

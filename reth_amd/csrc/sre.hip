@@ -786,6 +786,47 @@ __global__ void k_depth_scatter66_rec(const node_rec *__restrict__ recs,
 
 
 // merge two record arrays sorted by .s (distinct keys)
+// (merge-path variant measured 392.9 vs 390.1 ms same-box at 10Mx64 —
+// the carry side is small and L2-resident, so per-element binary search
+// wins; keeping the simple two-kernel merge.)
+__global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
+                          const node_rec *__restrict__ B, uint64_t nB,
+                          node_rec *__restrict__ out)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= nA)
+        return;
+    uint32_t key = A[i].s;
+    uint64_t lo = 0, hi = nB;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2;
+        if (B[mid].s < key)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    copy_rec(&out[i + lo], &A[i]);
+}
+
+__global__ void k_merge_b(const node_rec *__restrict__ A, uint64_t nA,
+                          const node_rec *__restrict__ B, uint64_t nB,
+                          node_rec *__restrict__ out)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= nB)
+        return;
+    uint32_t key = B[j].s;
+    uint64_t lo = 0, hi = nA;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2;
+        if (A[mid].s < key)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    copy_rec(&out[j + lo], &B[j]);
+}
+
 __global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
                           const node_rec *__restrict__ B, uint64_t nB,
                           node_rec *__restrict__ out)
@@ -970,7 +1011,8 @@ __global__ void k_class_hist(const uint32_t *__restrict__ gs, uint32_t n_groups,
 __global__ void k_class_scatter(const uint32_t *__restrict__ gs,
                                 uint32_t n_groups, uint32_t nblk,
                                 const uint32_t *__restrict__ offs,
-                                uint32_t *__restrict__ perm)
+                                uint32_t *__restrict__ perm,
+                                uint32_t *__restrict__ inv /* nullable */)
 {
     __shared__ uint32_t base_l[4];
     if (threadIdx.x < 4)
@@ -980,7 +1022,10 @@ __global__ void k_class_scatter(const uint32_t *__restrict__ gs,
     if (g >= n_groups)
         return;
     int c = nmem_class(gs[g + 1] - gs[g]);
-    perm[atomicAdd(&base_l[c], 1u)] = g;
+    uint32_t slot = atomicAdd(&base_l[c], 1u);
+    perm[slot] = g;
+    if (inv)
+        inv[g] = slot;
 }
 
 __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
@@ -1260,6 +1305,7 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         atomicAdd(&pending[threadIdx.x], hist_l[threadIdx.x]);
 }
 
+
 // TrieUpdates emission (updates mode): one row per STORED branch
 // (hash_mask != 0 — semantics in sre.h, pinned by the reference tests).
 // Runs per level chunk after k_branch_hash; row slots via LDS-aggregated
@@ -1378,6 +1424,94 @@ __device__ __forceinline__ uint64_t lb_keys(const uint8_t *base, uint64_t stride
             hi = mid;
     }
     return lo;
+}
+
+// ---- account multiproof capture (sre_account_proof; proof/mod.rs:59-137
+// semantics). A path node for target index ti is exactly the branch group
+// whose member interval contains ti, so per level chunk one thread per
+// target binary-searches the groups and, on containment, copies the
+// assembled branch RLP out of the scratch slot before it is reused.
+struct proof_row {
+    uint32_t target;
+    int16_t d, P;
+    uint32_t br_len;
+    uint8_t rlp[536];
+};
+static_assert(sizeof(proof_row) == 548, "proof_row layout");
+
+// target key -> index in the sorted account array (+ presence flag)
+__global__ void k_proof_ti(const sre_account_entry *__restrict__ acct,
+                           uint64_t na, const uint8_t *__restrict__ targets,
+                           uint32_t n_t, uint32_t *__restrict__ ti,
+                           uint32_t *__restrict__ present)
+{
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= n_t)
+        return;
+    const uint8_t *key = targets + 32ull * t;
+    uint64_t lo = 0, hi = na;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2;
+        if (cmp_key32(acct[mid].key, key) < 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    ti[t] = (uint32_t)lo;
+    present[t] = (lo < na && cmp_key32(acct[lo].key, key) == 0) ? 1u : 0u;
+}
+
+__global__ void k_proof_grab(const node_rec *__restrict__ L,
+                             const uint32_t *__restrict__ gs, uint32_t n_groups,
+                             const br_meta *__restrict__ meta,
+                             const uint8_t *__restrict__ scratch,
+                             uint64_t scratch_stride,
+                             const uint32_t *__restrict__ inv,
+                             const uint32_t *__restrict__ ti, uint32_t n_t,
+                             proof_row *__restrict__ rows,
+                             uint32_t *__restrict__ cnt, uint32_t cap,
+                             uint32_t *__restrict__ err)
+{
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= n_t)
+        return;
+    uint32_t pos = ti[t];
+    // largest group g with interval start <= pos
+    uint32_t lo = 0, hi = n_groups;
+    while (lo < hi) {
+        uint32_t mid = (lo + hi) / 2;
+        if (L[gs[mid]].s <= pos)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    if (lo == 0)
+        return;
+    uint32_t g = lo - 1;
+    if (pos >= L[gs[g + 1] - 1].e)
+        return; // this chunk's groups do not cover the target
+    uint32_t slot = inv ? inv[g] : g;
+    br_meta mt = meta[slot];
+    if (mt.br_len == 0)
+        return;
+    uint32_t r = atomicAdd(cnt, 1u);
+    if (r >= cap) {
+        atomicOr(err, 1u << E_INTERNAL);
+        return;
+    }
+    rows[r].target = t;
+    rows[r].d = (int16_t)mt.d;
+    rows[r].P = (int16_t)mt.P;
+    rows[r].br_len = mt.br_len;
+    const uint64_t *scr64 = (const uint64_t *)scratch;
+    uint64_t *dst = (uint64_t *)rows[r].rlp;
+    int nw = (mt.br_len + 7) / 8;
+    for (int w = 0; w < nw; ++w)
+#if SRE_SCRATCH_ROWMAJOR
+        dst[w] = scr64[(uint64_t)slot * (SLOT_BR_ROW / 8) + w];
+#else
+        dst[w] = scr64[(uint64_t)w * scratch_stride + slot];
+#endif
 }
 
 // old->new position map for interval rebasing: map[i] = the new index of
@@ -2117,7 +2251,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                       // record's keys share >= capture_depth nibbles: exactly
                       // one cell, disjoint from every other captured row)
                       int capture_depth = 0, cap_row *d_cap = nullptr,
-                      uint32_t *d_cap_cnt = nullptr, uint64_t cap_capacity = 0)
+                      uint32_t *d_cap_cnt = nullptr, uint64_t cap_capacity = 0,
+                      // account multiproof capture (sre_account_proof)
+                      const uint32_t *d_pti = nullptr, uint32_t n_pt = 0,
+                      proof_row *d_prows = nullptr,
+                      uint32_t *d_prow_cnt = nullptr, uint32_t prow_cap = 0)
 {
     int maxd = -1;
     for (int d = 63; d >= 0; --d)
@@ -2249,6 +2387,9 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         DBuf perm(ctx), ccnt(ctx), coff(ctx), scratch2(ctx), meta2(ctx);
         const uint32_t CLS_MIN = 1u << 14; // below this the win is noise
         bool use_cls = n_groups >= CLS_MIN;
+        DBuf cinv(ctx);
+        if (use_cls && n_pt)
+            HIP_CHECK(ctx, cinv.alloc((uint64_t)n_groups * 4));
         if (use_cls) {
             HIP_CHECK(ctx, perm.alloc((uint64_t)n_groups * 4));
             uint32_t nblk_max = (uint32_t)((chunk + CLS_BLOCK - 1) / CLS_BLOCK);
@@ -2269,7 +2410,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                 hipLaunchKernelGGL(k_class_scatter, dim3(nblk), dim3(CLS_BLOCK),
                                    0, ctx->stream, gs.as<uint32_t>() + g0, gc,
                                    nblk, coff.as<uint32_t>(),
-                                   perm.as<uint32_t>() + g0);
+                                   perm.as<uint32_t>() + g0,
+                                   n_pt ? cinv.as<uint32_t>() + g0 : nullptr);
                 HIP_CHECK(ctx, hipGetLastError());
             }
         }
@@ -2311,6 +2453,16 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
             if (pipe2) {
                 hipEventRecord(ev_asm[buf], s_asm);
                 hipStreamWaitEvent(ctx->stream, ev_asm[buf], 0);
+            }
+            if (n_pt) { // multiproof: copy path-node RLPs out of scratch
+                hipLaunchKernelGGL(k_proof_grab,
+                                   dim3((n_pt + BLOCK - 1) / BLOCK), dim3(BLOCK),
+                                   0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
+                                   mt, scr, chunk,
+                                   use_cls ? cinv.as<uint32_t>() + g0 : nullptr,
+                                   d_pti, n_pt, d_prows, d_prow_cnt, prow_cap,
+                                   d_err);
+                HIP_CHECK(ctx, hipGetLastError());
             }
             if (updates_kind >= 0) {
                 // emit BEFORE hashing: children's bhash entries must not yet
@@ -2431,8 +2583,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                    dim3(BLOCK), 0, ctx->stream,
                                    carries[p].as<node_rec>(), carry_cnt[p],
                                    slice, fresh, cmerge.as<node_rec>());
-                hipLaunchKernelGGL(k_merge_b, dim3(grid_for(fresh)), dim3(BLOCK), 0,
-                                   ctx->stream, carries[p].as<node_rec>(),
+                hipLaunchKernelGGL(k_merge_b, dim3(grid_for(fresh)), dim3(BLOCK),
+                                   0, ctx->stream, carries[p].as<node_rec>(),
                                    carry_cnt[p], slice, fresh,
                                    cmerge.as<node_rec>());
                 HIP_CHECK(ctx, hipGetLastError());
@@ -2567,7 +2719,11 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
                             uint8_t *d_child_lens, pass_out *po, uint32_t *d_err,
                             int capture_depth = 0, cap_row *d_cap = nullptr,
                             uint32_t *d_cap_cnt = nullptr,
-                            uint64_t cap_capacity = 0)
+                            uint64_t cap_capacity = 0,
+                            const uint32_t *d_pti = nullptr, uint32_t n_pt = 0,
+                            proof_row *d_prows = nullptr,
+                            uint32_t *d_prow_cnt = nullptr,
+                            uint32_t prow_cap = 0)
 {
     uint64_t na = ctx->na;
     DBuf lcp(ctx), recs(ctx), depths(ctx), hist(ctx);
@@ -2615,7 +2771,8 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
                    lcp.as<int8_t>(), keys, sizeof(sre_account_entry), hist_host,
                    subtree, d_roots, d_child_refs, d_child_lens, d_err, po,
                    ctx->retain_updates ? 0 : -1, bhash.as<uint8_t>(),
-                   capture_depth, d_cap, d_cap_cnt, cap_capacity))
+                   capture_depth, d_cap, d_cap_cnt, cap_capacity,
+                   d_pti, n_pt, d_prows, d_prow_cnt, prow_cap))
         return -1;
     if (ctx->retain_updates) {
         for (size_t r = upd_start; r < ctx->updates.size(); ++r)
@@ -2651,6 +2808,332 @@ static const uint8_t EMPTY_ROOT_H[32] = {
     0x92, 0xc0, 0xf8, 0x6e, 0x5b, 0x48, 0xe0, 0x1b, 0x99, 0x6c, 0xad, 0xc0,
     0x01, 0x62, 0x2f, 0xb5, 0xe3, 0x63, 0xb4, 0x21,
 };
+
+// ---------------------------------------------------------------------------
+// account multiproof (sre_account_proof)
+// ---------------------------------------------------------------------------
+// Host-side Keccak-256 (FIPS-202 restatement, independent of oracle/) —
+// used only to derive extension-node child references while assembling
+// proof node lists host-side; all bulk hashing stays on the GPU.
+static void h_keccak_f(uint64_t s[25])
+{
+    static const uint64_t RC[24] = {
+        0x0000000000000001ULL, 0x0000000000008082ULL, 0x800000000000808aULL,
+        0x8000000080008000ULL, 0x000000000000808bULL, 0x0000000080000001ULL,
+        0x8000000080008081ULL, 0x8000000000008009ULL, 0x000000000000008aULL,
+        0x0000000000000088ULL, 0x0000000080008009ULL, 0x000000008000000aULL,
+        0x000000008000808bULL, 0x800000000000008bULL, 0x8000000000008089ULL,
+        0x8000000000008003ULL, 0x8000000000008002ULL, 0x8000000000000080ULL,
+        0x000000000000800aULL, 0x800000008000000aULL, 0x8000000080008081ULL,
+        0x8000000000008080ULL, 0x0000000080000001ULL, 0x8000000080008008ULL};
+    static const int ROT[25] = {0,  1,  62, 28, 27, 36, 44, 6,  55, 20, 3, 10, 43,
+                                25, 39, 41, 45, 15, 21, 8,  18, 2,  61, 56, 14};
+    auto rotl = [](uint64_t v, int r) {
+        return r ? (v << r) | (v >> (64 - r)) : v;
+    };
+    for (int r = 0; r < 24; ++r) {
+        uint64_t c[5], dd[5], b[25];
+        for (int x = 0; x < 5; ++x)
+            c[x] = s[x] ^ s[x + 5] ^ s[x + 10] ^ s[x + 15] ^ s[x + 20];
+        for (int x = 0; x < 5; ++x)
+            dd[x] = c[(x + 4) % 5] ^ rotl(c[(x + 1) % 5], 1);
+        for (int i = 0; i < 25; ++i)
+            s[i] ^= dd[i % 5];
+        for (int x = 0; x < 5; ++x)
+            for (int y = 0; y < 5; ++y)
+                b[y + 5 * ((2 * x + 3 * y) % 5)] = rotl(s[x + 5 * y],
+                                                        ROT[x + 5 * y]);
+        for (int y = 0; y < 5; ++y)
+            for (int x = 0; x < 5; ++x)
+                s[x + 5 * y] = b[x + 5 * y] ^
+                               ((~b[(x + 1) % 5 + 5 * y]) &
+                                b[(x + 2) % 5 + 5 * y]);
+        s[0] ^= RC[r];
+    }
+}
+
+static void h_keccak256(const uint8_t *msg, size_t len, uint8_t out[32])
+{
+    uint64_t s[25] = {0};
+    uint8_t blk[136];
+    size_t off = 0;
+    while (len - off >= 136) {
+        memcpy(blk, msg + off, 136);
+        for (int i = 0; i < 17; ++i) {
+            uint64_t w;
+            memcpy(&w, blk + 8 * i, 8);
+            s[i] ^= w;
+        }
+        h_keccak_f(s);
+        off += 136;
+    }
+    memset(blk, 0, 136);
+    memcpy(blk, msg + off, len - off);
+    blk[len - off] = 0x01;
+    blk[135] |= 0x80;
+    for (int i = 0; i < 17; ++i) {
+        uint64_t w;
+        memcpy(&w, blk + 8 * i, 8);
+        s[i] ^= w;
+    }
+    h_keccak_f(s);
+    memcpy(out, s, 32);
+}
+
+static inline int h_nib(const uint8_t *key, int i)
+{
+    return (i & 1) ? (key[i >> 1] & 0xF) : (key[i >> 1] >> 4);
+}
+
+// RLP string item of the hex-prefix-encoded path key[from..to) (HP per the
+// yellow paper; in-repo shape proof_v2/node.rs:30-68). Returns bytes written.
+static int h_hp_item(uint8_t *dst, const uint8_t *key, int from, int to,
+                     int leaf)
+{
+    int n = to - from;
+    int hl = 1 + n / 2;
+    uint8_t hp[34];
+    int odd = n & 1;
+    hp[0] = (uint8_t)((leaf ? 0x20 : 0x00) | (odd ? 0x10 | h_nib(key, from) : 0));
+    int p = 1, i = from + odd;
+    for (; i < to; i += 2)
+        hp[p++] = (uint8_t)((h_nib(key, i) << 4) | h_nib(key, i + 1));
+    int w = 0;
+    if (hl == 1 && hp[0] < 0x80) {
+        dst[w++] = hp[0];
+    } else {
+        dst[w++] = (uint8_t)(0x80 + hl);
+        memcpy(dst + w, hp, hl);
+        w += hl;
+    }
+    return w;
+}
+
+// RLP of a big-endian unsigned integer (nonce / balance), minimal form.
+static int h_rlp_uint(uint8_t *dst, const uint8_t *be, int len)
+{
+    int s = 0;
+    while (s < len && be[s] == 0)
+        s++;
+    int n = len - s;
+    if (n == 0) {
+        dst[0] = 0x80;
+        return 1;
+    }
+    if (n == 1 && be[s] < 0x80) {
+        dst[0] = be[s];
+        return 1;
+    }
+    dst[0] = (uint8_t)(0x80 + n);
+    memcpy(dst + 1, be + s, n);
+    return 1 + n;
+}
+
+static int h_rlp_list_hdr(uint8_t *dst, int payload)
+{
+    if (payload < 56) {
+        dst[0] = (uint8_t)(0xc0 + payload);
+        return 1;
+    }
+    dst[0] = 0xf8;
+    dst[1] = (uint8_t)payload;
+    return 2;
+}
+
+// Account leaf node RLP: [HP(short,1), RLP_string(RLP([nonce,balance,
+// storage_root,code_hash]))] (trie.rs:472-476; proof_v2/value.rs:55-139).
+static int h_leaf_node(uint8_t *dst, const uint8_t *key, int from,
+                       const sre_account_entry *a,
+                       const uint8_t storage_root[32])
+{
+    uint8_t val[120];
+    uint8_t nb[8];
+    for (int i = 0; i < 8; ++i)
+        nb[i] = (uint8_t)(a->nonce >> (8 * (7 - i)));
+    int vp = 0;
+    uint8_t body[112];
+    int bp = 0;
+    bp += h_rlp_uint(body + bp, nb, 8);
+    bp += h_rlp_uint(body + bp, a->balance, 32);
+    body[bp++] = 0xa0;
+    memcpy(body + bp, storage_root, 32);
+    bp += 32;
+    body[bp++] = 0xa0;
+    memcpy(body + bp, a->code_hash, 32);
+    bp += 32;
+    vp += h_rlp_list_hdr(val + vp, bp);
+    memcpy(val + vp, body, bp);
+    vp += bp;
+    // leaf list: [hp, value-as-string]
+    uint8_t hp[40];
+    int hl = h_hp_item(hp, key, from, 64, 1);
+    int pay = hl + (vp < 56 ? 1 : 2) + vp;
+    int w = h_rlp_list_hdr(dst, pay);
+    memcpy(dst + w, hp, hl);
+    w += hl;
+    if (vp < 56) {
+        dst[w++] = (uint8_t)(0x80 + vp);
+    } else {
+        dst[w++] = 0xb8;
+        dst[w++] = (uint8_t)vp;
+    }
+    memcpy(dst + w, val, vp);
+    return w + vp;
+}
+
+extern "C" int sre_account_proof(sre_ctx *ctx, const uint8_t *targets,
+                                 uint64_t n_targets, uint8_t *out_nodes,
+                                 uint64_t cap_nodes, uint32_t *out_lens,
+                                 uint64_t cap_lens, uint32_t *out_counts)
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    if (ctx->na == 0) {
+        set_err(ctx, "sre_account_proof: empty state (target cannot be present)");
+        return -1;
+    }
+    if (n_targets == 0 || n_targets > 4096) {
+        set_err(ctx, "sre_account_proof: 1..4096 targets");
+        return -1;
+    }
+    uint32_t n_t = (uint32_t)n_targets;
+    DBuf err(ctx), acct_roots(ctx), root(ctx), dtgt(ctx), dti(ctx), dpres(ctx),
+        prows(ctx), prowc(ctx);
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, acct_roots.alloc(ctx->na * 32));
+    HIP_CHECK(ctx, root.alloc(32));
+    HIP_CHECK(ctx, dtgt.alloc(32ull * n_t));
+    HIP_CHECK(ctx, hipMemcpyAsync(dtgt.p, targets, 32ull * n_t,
+                                  hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHECK(ctx, dti.alloc(4ull * n_t));
+    HIP_CHECK(ctx, dpres.alloc(4ull * n_t));
+    hipLaunchKernelGGL(k_proof_ti, dim3(grid_for(n_t)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_acct, ctx->na, dtgt.as<uint8_t>(),
+                       n_t, dti.as<uint32_t>(), dpres.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    std::vector<uint32_t> ti(n_t), pres(n_t);
+    HIP_CHECK(ctx, hipMemcpy(ti.data(), dti.p, 4ull * n_t,
+                             hipMemcpyDeviceToHost));
+    HIP_CHECK(ctx, hipMemcpy(pres.data(), dpres.p, 4ull * n_t,
+                             hipMemcpyDeviceToHost));
+    for (uint32_t t = 0; t < n_t; ++t)
+        if (!pres[t]) {
+            set_err(ctx, "sre_account_proof: target key not present "
+                         "(exclusion proofs are not supported in v1)");
+            return -1;
+        }
+    uint32_t cap_rows = n_t * 130 + 64;
+    HIP_CHECK(ctx, prows.alloc((uint64_t)cap_rows * sizeof(proof_row)));
+    HIP_CHECK(ctx, prowc.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(prowc.p, 0, 4, ctx->stream));
+
+    pass_out po;
+    if (run_storage_pass(ctx, acct_roots.as<uint8_t>(), &po, err.as<uint32_t>()))
+        return -1;
+    if (run_account_pass(ctx, acct_roots.as<uint8_t>(), 0, root.as<uint8_t>(),
+                         nullptr, nullptr, &po, err.as<uint32_t>(), 0, nullptr,
+                         nullptr, 0, dti.as<uint32_t>(), n_t,
+                         prows.as<proof_row>(), prowc.as<uint32_t>(), cap_rows))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    uint32_t nrows = 0;
+    HIP_CHECK(ctx, hipMemcpy(&nrows, prowc.p, 4, hipMemcpyDeviceToHost));
+    std::vector<proof_row> rows(nrows);
+    if (nrows)
+        HIP_CHECK(ctx, hipMemcpy(rows.data(), prows.p,
+                                 (uint64_t)nrows * sizeof(proof_row),
+                                 hipMemcpyDeviceToHost));
+    uint8_t engine_root[32];
+    HIP_CHECK(ctx, hipMemcpy(engine_root, root.p, 32, hipMemcpyDeviceToHost));
+
+    // per-target path rows, root-first (ascending branch depth)
+    std::vector<std::vector<const proof_row *>> per(n_t);
+    for (const auto &r : rows)
+        per[r.target].push_back(&r);
+    for (auto &v : per)
+        std::sort(v.begin(), v.end(),
+                  [](const proof_row *a, const proof_row *b) {
+                      return a->d < b->d;
+                  });
+
+    uint64_t nb = 0, nl = 0;
+    for (uint32_t t = 0; t < n_t; ++t) {
+        const uint8_t *key = targets + 32ull * t;
+        uint32_t cnt = 0;
+        auto emit = [&](const uint8_t *node, int len) -> int {
+            if (nl >= cap_lens || nb + (uint64_t)len > cap_nodes) {
+                set_err(ctx, "sre_account_proof: output capacity exceeded");
+                return -1;
+            }
+            memcpy(out_nodes + nb, node, len);
+            nb += len;
+            out_lens[nl++] = (uint32_t)len;
+            cnt++;
+            return 0;
+        };
+        bool first = true;
+        for (const proof_row *r : per[t]) {
+            if (r->d > r->P + 1) { // extension above this branch
+                uint8_t ext[80];
+                uint8_t cref[33];
+                int crl;
+                if (r->br_len >= 32) {
+                    cref[0] = 0xa0;
+                    h_keccak256(r->rlp, r->br_len, cref + 1);
+                    crl = 33;
+                } else { // inline child (never in account tries; defensive)
+                    memcpy(cref, r->rlp, r->br_len);
+                    crl = (int)r->br_len;
+                }
+                uint8_t hp[40];
+                int hl = h_hp_item(hp, key, r->P + 1, r->d, 0);
+                int pay = hl + crl;
+                int w = h_rlp_list_hdr(ext, pay);
+                memcpy(ext + w, hp, hl);
+                memcpy(ext + w + hl, cref, crl);
+                // inline (<32 B, non-root) nodes are embedded in their
+                // parent and are not separate proof entries (never hit in
+                // account tries; defensive exactness)
+                if ((w + pay >= 32 || r->P < 0) && emit(ext, w + pay))
+                    return -1;
+            }
+            if ((r->br_len >= 32 || (r->P < 0 && r->d == r->P + 1)) &&
+                emit(r->rlp, (int)r->br_len))
+                return -1;
+            first = false;
+        }
+        (void)first;
+        // the account leaf itself
+        int dmax = per[t].empty() ? -1 : per[t].back()->d;
+        sre_account_entry ae;
+        uint8_t sroot[32];
+        HIP_CHECK(ctx, hipMemcpy(&ae,
+                                 (const uint8_t *)ctx->d_acct +
+                                     (uint64_t)ti[t] * sizeof(sre_account_entry),
+                                 sizeof(ae), hipMemcpyDeviceToHost));
+        HIP_CHECK(ctx, hipMemcpy(sroot,
+                                 acct_roots.as<uint8_t>() + 32ull * ti[t], 32,
+                                 hipMemcpyDeviceToHost));
+        uint8_t leaf[200];
+        int ll = h_leaf_node(leaf, key, dmax + 1, &ae, sroot);
+        if (emit(leaf, ll))
+            return -1;
+        out_counts[t] = cnt;
+        // sanity: the first node must hash to the engine's own root
+        uint8_t h[32];
+        uint64_t firstoff = nb;
+        uint32_t firstlen = out_lens[nl - cnt];
+        for (uint32_t k2 = nl - cnt; k2 < nl; ++k2)
+            firstoff -= out_lens[k2];
+        h_keccak256(out_nodes + firstoff, firstlen, h);
+        if (memcmp(h, engine_root, 32) != 0) {
+            set_err(ctx, "sre_account_proof: internal root mismatch");
+            return -1;
+        }
+    }
+    return 0;
+}
 
 extern "C" int sre_root(sre_ctx *ctx, uint8_t out_root[32])
 {

@@ -213,6 +213,32 @@ class StateRootEngine:
                 ctypes.c_void_p(self._ctx), _np_ptr(rows), n))
         return bytes(out), rows
 
+    def account_proof(self, targets) -> list:
+        """Account multiproof for PRESENT hashed keys: per target, the list
+        of proof node RLPs root-first (Proof::multiproof account surface,
+        include/sre.h sre_account_proof). targets: list of 32-byte keys.
+        Returns list-of-list-of-bytes in target order."""
+        n = len(targets)
+        tarr = np.frombuffer(b"".join(targets), dtype=np.uint8).reshape(n, 32)
+        tarr = np.ascontiguousarray(tarr)
+        cap_nodes = n * 130 * 560 + 4096
+        cap_lens = n * 132
+        nodes = np.zeros(cap_nodes, dtype=np.uint8)
+        lens = np.zeros(cap_lens, dtype=np.uint32)
+        counts = np.zeros(n, dtype=np.uint32)
+        self._check(self._lib.sre_account_proof(
+            ctypes.c_void_p(self._ctx), _np_ptr(tarr), ctypes.c_uint64(n),
+            _np_ptr(nodes), ctypes.c_uint64(cap_nodes),
+            _np_ptr(lens), ctypes.c_uint64(cap_lens), _np_ptr(counts)))
+        out, off, li = [], 0, 0
+        for t in range(n):
+            tl = []
+            for _ in range(int(counts[t])):
+                ln = int(lens[li]); li += 1
+                tl.append(nodes[off:off + ln].tobytes()); off += ln
+            out.append(tl)
+        return out
+
     def storage_roots(self, n) -> np.ndarray:
         out = np.empty((n, 32), dtype=np.uint8)
         self._check(self._lib.sre_storage_roots(
