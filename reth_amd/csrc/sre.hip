@@ -786,46 +786,6 @@ __global__ void k_depth_scatter66_rec(const node_rec *__restrict__ recs,
 
 
 // merge two record arrays sorted by .s (distinct keys)
-// (merge-path variant measured 392.9 vs 390.1 ms same-box at 10Mx64 —
-// the carry side is small and L2-resident, so per-element binary search
-// wins; keeping the simple two-kernel merge.)
-__global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
-                          const node_rec *__restrict__ B, uint64_t nB,
-                          node_rec *__restrict__ out)
-{
-    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= nA)
-        return;
-    uint32_t key = A[i].s;
-    uint64_t lo = 0, hi = nB;
-    while (lo < hi) {
-        uint64_t mid = (lo + hi) / 2;
-        if (B[mid].s < key)
-            lo = mid + 1;
-        else
-            hi = mid;
-    }
-    copy_rec(&out[i + lo], &A[i]);
-}
-
-__global__ void k_merge_b(const node_rec *__restrict__ A, uint64_t nA,
-                          const node_rec *__restrict__ B, uint64_t nB,
-                          node_rec *__restrict__ out)
-{
-    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (j >= nB)
-        return;
-    uint32_t key = B[j].s;
-    uint64_t lo = 0, hi = nA;
-    while (lo < hi) {
-        uint64_t mid = (lo + hi) / 2;
-        if (A[mid].s < key)
-            lo = mid + 1;
-        else
-            hi = mid;
-    }
-    copy_rec(&out[j + lo], &B[j]);
-}
 
 __global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
                           const node_rec *__restrict__ B, uint64_t nB,
