@@ -210,6 +210,20 @@ int sre_account_proof(sre_ctx *ctx,
                       uint32_t *out_lens, uint64_t cap_lens,
                       uint32_t *out_counts);
 
+/* Storage multiproof — StorageProof::storage_multiproof
+ * (crates/trie/trie/src/proof/mod.rs) for PRESENT (acct_key, slot_key)
+ * pairs: per target the account's storage root (out_roots, 32 B each) and
+ * the root-first node list of its storage trie, same output layout and v1
+ * limits as sre_account_proof. Storage tries CAN contain inline (<32 B)
+ * nodes; those are embedded in their parents and not emitted. */
+int sre_storage_proof(sre_ctx *ctx,
+                      const uint8_t *acct_keys /* n x 32 */,
+                      const uint8_t *slot_keys /* n x 32 */,
+                      uint64_t n_targets, uint8_t *out_roots,
+                      uint8_t *out_nodes, uint64_t cap_nodes,
+                      uint32_t *out_lens, uint64_t cap_lens,
+                      uint32_t *out_counts);
+
 /* Compute the state root AND retain the stored trie nodes (TrieUpdates) —
  * the surface of StateRootProvider::state_root_with_updates
  * (crates/storage/storage-api/src/trie.rs:30) / StateRoot::root_with_updates

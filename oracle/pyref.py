@@ -283,3 +283,16 @@ def account_proof(accounts: dict, hashed_key: bytes):
     nodes = []
     _collect_proof(items, 0, target, True, nodes)
     return nodes
+
+
+def storage_proof(accounts: dict, acct_key: bytes, slot_key: bytes):
+    """(storage_root, proof-node list root-first) for a PRESENT slot of a
+    present account — StorageProof::storage_multiproof semantics."""
+    slots = accounts[acct_key][3]
+    items = sorted((tuple(nibbles_of(k)), rlp_int(v))
+                   for k, v in slots.items() if v != 0)
+    target = tuple(nibbles_of(slot_key))
+    nodes = []
+    _collect_proof(items, 0, target, True, nodes)
+    root = keccak256(_build(items, 0))
+    return root, nodes

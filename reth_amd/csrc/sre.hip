@@ -1421,6 +1421,30 @@ __global__ void k_proof_ti(const sre_account_entry *__restrict__ acct,
     present[t] = (lo < na && cmp_key32(acct[lo].key, key) == 0) ? 1u : 0u;
 }
 
+// (acct_key, slot_key) -> index in the sorted storage array (+ presence)
+__global__ void k_proof_ti64(const sre_storage_entry *__restrict__ st,
+                             uint64_t ns, const uint8_t *__restrict__ targets,
+                             uint32_t n_t, uint32_t *__restrict__ ti,
+                             uint32_t *__restrict__ present)
+{
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= n_t)
+        return;
+    const uint8_t *key = targets + 64ull * t;
+    uint64_t lo = 0, hi = ns;
+    while (lo < hi) {
+        uint64_t mid = (lo + hi) / 2;
+        if (cmp_key64((const uint8_t *)&st[mid], key) < 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    ti[t] = (uint32_t)lo;
+    present[t] = (lo < ns && cmp_key64((const uint8_t *)&st[lo], key) == 0)
+                     ? 1u
+                     : 0u;
+}
+
 __global__ void k_proof_grab(const node_rec *__restrict__ L,
                              const uint32_t *__restrict__ gs, uint32_t n_groups,
                              const br_meta *__restrict__ meta,
@@ -2568,7 +2592,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 // ---------------------------------------------------------------------------
 
 static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
-                            uint32_t *d_err)
+                            uint32_t *d_err,
+                            const uint32_t *d_pti = nullptr, uint32_t n_pt = 0,
+                            proof_row *d_prows = nullptr,
+                            uint32_t *d_prow_cnt = nullptr,
+                            uint32_t prow_cap = 0)
 {
     uint64_t ns = ctx->ns, na = ctx->na;
     hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(na)), dim3(BLOCK), 0,
@@ -2645,7 +2673,9 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
     if (run_levels(ctx, ns, recs.as<node_rec>(), depths.as<uint8_t>(),
                    lcp.as<int8_t>(), keys, sizeof(sre_storage_entry), hist_host, 0,
                    seg_roots.as<uint8_t>(), nullptr, nullptr, d_err, po,
-                   ctx->retain_updates ? 1 : -1, bhash.as<uint8_t>()))
+                   ctx->retain_updates ? 1 : -1, bhash.as<uint8_t>(),
+                   0, nullptr, nullptr, 0, d_pti, n_pt, d_prows, d_prow_cnt,
+                   prow_cap))
         return -1;
     if (ctx->retain_updates && ctx->updates.size() > upd_start) {
         // patch acct_key from the stashed seg ids: seg -> account index ->
@@ -3089,6 +3119,186 @@ extern "C" int sre_account_proof(sre_ctx *ctx, const uint8_t *targets,
         h_keccak256(out_nodes + firstoff, firstlen, h);
         if (memcmp(h, engine_root, 32) != 0) {
             set_err(ctx, "sre_account_proof: internal root mismatch");
+            return -1;
+        }
+    }
+    return 0;
+}
+
+// Storage leaf node RLP: [HP(short,1), RLP_string(RLP(U256 value))]
+// (trie.rs:819-825 encode_fixed_size; proof_v2/value.rs:55).
+static int h_storage_leaf(uint8_t *dst, const uint8_t *slot_key, int from,
+                          const uint8_t value_be[32])
+{
+    uint8_t val[40];
+    int vp = h_rlp_uint(val, value_be, 32);
+    uint8_t hp[40];
+    int hl = h_hp_item(hp, slot_key, from, 64, 1);
+    int pay = hl + 1 + vp; // value string header is always 1 byte (vp <= 33)
+    int w = h_rlp_list_hdr(dst, pay);
+    memcpy(dst + w, hp, hl);
+    w += hl;
+    dst[w++] = (uint8_t)(0x80 + vp);
+    memcpy(dst + w, val, vp);
+    return w + vp;
+}
+
+/* Storage multiproof: per (acct_key, slot_key) PRESENT pair, the storage
+ * root and the root-first node list of that account's storage trie —
+ * StorageProof::storage_multiproof (crates/trie/trie/src/proof/mod.rs)
+ * restricted to present slots, same v1 limits as sre_account_proof. */
+extern "C" int sre_storage_proof(sre_ctx *ctx, const uint8_t *acct_keys,
+                                 const uint8_t *slot_keys, uint64_t n_targets,
+                                 uint8_t *out_roots, uint8_t *out_nodes,
+                                 uint64_t cap_nodes, uint32_t *out_lens,
+                                 uint64_t cap_lens, uint32_t *out_counts)
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    if (ctx->na == 0 || ctx->ns == 0) {
+        set_err(ctx, "sre_storage_proof: no storage resident");
+        return -1;
+    }
+    if (n_targets == 0 || n_targets > 4096) {
+        set_err(ctx, "sre_storage_proof: 1..4096 targets");
+        return -1;
+    }
+    uint32_t n_t = (uint32_t)n_targets;
+    std::vector<uint8_t> pairs(64ull * n_t);
+    for (uint32_t t = 0; t < n_t; ++t) {
+        memcpy(pairs.data() + 64ull * t, acct_keys + 32ull * t, 32);
+        memcpy(pairs.data() + 64ull * t + 32, slot_keys + 32ull * t, 32);
+    }
+    DBuf err(ctx), acct_roots(ctx), dtgt(ctx), dti(ctx), dpres(ctx),
+        dacct(ctx), dtia(ctx), dpra(ctx), prows(ctx), prowc(ctx);
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, acct_roots.alloc(ctx->na * 32));
+    HIP_CHECK(ctx, dtgt.alloc(64ull * n_t));
+    HIP_CHECK(ctx, hipMemcpyAsync(dtgt.p, pairs.data(), 64ull * n_t,
+                                  hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHECK(ctx, dti.alloc(4ull * n_t));
+    HIP_CHECK(ctx, dpres.alloc(4ull * n_t));
+    hipLaunchKernelGGL(k_proof_ti64, dim3(grid_for(n_t)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_st, ctx->ns, dtgt.as<uint8_t>(),
+                       n_t, dti.as<uint32_t>(), dpres.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    // account indices (for the per-target storage root)
+    HIP_CHECK(ctx, dacct.alloc(32ull * n_t));
+    HIP_CHECK(ctx, hipMemcpyAsync(dacct.p, acct_keys, 32ull * n_t,
+                                  hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHECK(ctx, dtia.alloc(4ull * n_t));
+    HIP_CHECK(ctx, dpra.alloc(4ull * n_t));
+    hipLaunchKernelGGL(k_proof_ti, dim3(grid_for(n_t)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_acct, ctx->na, dacct.as<uint8_t>(),
+                       n_t, dtia.as<uint32_t>(), dpra.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    std::vector<uint32_t> ti(n_t), pres(n_t), tia(n_t), presa(n_t);
+    HIP_CHECK(ctx, hipMemcpy(ti.data(), dti.p, 4ull * n_t,
+                             hipMemcpyDeviceToHost));
+    HIP_CHECK(ctx, hipMemcpy(pres.data(), dpres.p, 4ull * n_t,
+                             hipMemcpyDeviceToHost));
+    HIP_CHECK(ctx, hipMemcpy(tia.data(), dtia.p, 4ull * n_t,
+                             hipMemcpyDeviceToHost));
+    HIP_CHECK(ctx, hipMemcpy(presa.data(), dpra.p, 4ull * n_t,
+                             hipMemcpyDeviceToHost));
+    for (uint32_t t = 0; t < n_t; ++t)
+        if (!pres[t] || !presa[t]) {
+            set_err(ctx, "sre_storage_proof: (account, slot) not present "
+                         "(exclusion proofs are not supported in v1)");
+            return -1;
+        }
+    uint32_t cap_rows = n_t * 130 + 64;
+    HIP_CHECK(ctx, prows.alloc((uint64_t)cap_rows * sizeof(proof_row)));
+    HIP_CHECK(ctx, prowc.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(prowc.p, 0, 4, ctx->stream));
+
+    pass_out po;
+    if (run_storage_pass(ctx, acct_roots.as<uint8_t>(), &po, err.as<uint32_t>(),
+                         dti.as<uint32_t>(), n_t, prows.as<proof_row>(),
+                         prowc.as<uint32_t>(), cap_rows))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    uint32_t nrows = 0;
+    HIP_CHECK(ctx, hipMemcpy(&nrows, prowc.p, 4, hipMemcpyDeviceToHost));
+    std::vector<proof_row> rows(nrows);
+    if (nrows)
+        HIP_CHECK(ctx, hipMemcpy(rows.data(), prows.p,
+                                 (uint64_t)nrows * sizeof(proof_row),
+                                 hipMemcpyDeviceToHost));
+    std::vector<std::vector<const proof_row *>> per(n_t);
+    for (const auto &r : rows)
+        per[r.target].push_back(&r);
+    for (auto &v : per)
+        std::sort(v.begin(), v.end(),
+                  [](const proof_row *a, const proof_row *b) {
+                      return a->d < b->d;
+                  });
+
+    uint64_t nb = 0, nl = 0;
+    for (uint32_t t = 0; t < n_t; ++t) {
+        const uint8_t *key = slot_keys + 32ull * t;
+        uint8_t sroot[32];
+        HIP_CHECK(ctx, hipMemcpy(sroot,
+                                 acct_roots.as<uint8_t>() + 32ull * tia[t], 32,
+                                 hipMemcpyDeviceToHost));
+        memcpy(out_roots + 32ull * t, sroot, 32);
+        uint32_t cnt = 0;
+        auto emit = [&](const uint8_t *node, int len) -> int {
+            if (nl >= cap_lens || nb + (uint64_t)len > cap_nodes) {
+                set_err(ctx, "sre_storage_proof: output capacity exceeded");
+                return -1;
+            }
+            memcpy(out_nodes + nb, node, len);
+            nb += len;
+            out_lens[nl++] = (uint32_t)len;
+            cnt++;
+            return 0;
+        };
+        for (const proof_row *r : per[t]) {
+            if (r->d > r->P + 1) {
+                uint8_t ext[80];
+                uint8_t cref[33];
+                int crl;
+                if (r->br_len >= 32) {
+                    cref[0] = 0xa0;
+                    h_keccak256(r->rlp, r->br_len, cref + 1);
+                    crl = 33;
+                } else {
+                    memcpy(cref, r->rlp, r->br_len);
+                    crl = (int)r->br_len;
+                }
+                uint8_t hp[40];
+                int hl = h_hp_item(hp, key, r->P + 1, r->d, 0);
+                int pay = hl + crl;
+                int w = h_rlp_list_hdr(ext, pay);
+                memcpy(ext + w, hp, hl);
+                memcpy(ext + w + hl, cref, crl);
+                if ((w + pay >= 32 || r->P < 0) && emit(ext, w + pay))
+                    return -1;
+            }
+            if ((r->br_len >= 32 || (r->P < 0 && r->d == r->P + 1)) &&
+                emit(r->rlp, (int)r->br_len))
+                return -1;
+        }
+        int dmax = per[t].empty() ? -1 : per[t].back()->d;
+        sre_storage_entry se;
+        HIP_CHECK(ctx, hipMemcpy(&se,
+                                 (const uint8_t *)ctx->d_st +
+                                     (uint64_t)ti[t] * sizeof(sre_storage_entry),
+                                 sizeof(se), hipMemcpyDeviceToHost));
+        uint8_t leaf[80];
+        int ll = h_storage_leaf(leaf, key, dmax + 1, se.value);
+        if ((ll >= 32 || per[t].empty()) && emit(leaf, ll))
+            return -1;
+        out_counts[t] = cnt;
+        uint64_t firstoff = nb;
+        for (uint32_t k2 = nl - cnt; k2 < nl; ++k2)
+            firstoff -= out_lens[k2];
+        uint8_t h[32];
+        h_keccak256(out_nodes + firstoff, out_lens[nl - cnt], h);
+        if (memcmp(h, sroot, 32) != 0) {
+            set_err(ctx, "sre_storage_proof: internal root mismatch");
             return -1;
         }
     }

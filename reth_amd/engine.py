@@ -239,6 +239,36 @@ class StateRootEngine:
             out.append(tl)
         return out
 
+    def storage_proof(self, acct_keys, slot_keys):
+        """Storage multiproof for PRESENT (account, slot) pairs: returns
+        (roots, proofs) — per target the account's storage root (bytes) and
+        the root-first node-RLP list of its storage trie."""
+        n = len(acct_keys)
+        assert len(slot_keys) == n
+        aarr = np.ascontiguousarray(
+            np.frombuffer(b"".join(acct_keys), dtype=np.uint8).reshape(n, 32))
+        sarr = np.ascontiguousarray(
+            np.frombuffer(b"".join(slot_keys), dtype=np.uint8).reshape(n, 32))
+        cap_nodes = n * 130 * 560 + 4096
+        cap_lens = n * 132
+        roots = np.zeros((n, 32), dtype=np.uint8)
+        nodes = np.zeros(cap_nodes, dtype=np.uint8)
+        lens = np.zeros(cap_lens, dtype=np.uint32)
+        counts = np.zeros(n, dtype=np.uint32)
+        self._check(self._lib.sre_storage_proof(
+            ctypes.c_void_p(self._ctx), _np_ptr(aarr), _np_ptr(sarr),
+            ctypes.c_uint64(n), _np_ptr(roots),
+            _np_ptr(nodes), ctypes.c_uint64(cap_nodes),
+            _np_ptr(lens), ctypes.c_uint64(cap_lens), _np_ptr(counts)))
+        out, off, li = [], 0, 0
+        for t in range(n):
+            tl = []
+            for _ in range(int(counts[t])):
+                ln = int(lens[li]); li += 1
+                tl.append(nodes[off:off + ln].tobytes()); off += ln
+            out.append(tl)
+        return [bytes(r) for r in roots], out
+
     def storage_roots(self, n) -> np.ndarray:
         out = np.empty((n, 32), dtype=np.uint8)
         self._check(self._lib.sre_storage_roots(

@@ -151,3 +151,56 @@ def test_proof_absent_key_rejected(eng):
     eng.upload(acct, st)
     with pytest.raises(RuntimeError):
         eng.account_proof([bind.keccak256(b"definitely-absent")])
+
+
+def test_storage_proof_parity(eng):
+    acct, st = gen.gen_state_numpy(500, 24, bind.keccak256_batch)
+    accounts = _dict_of(acct, st)
+    eng.upload(acct, st)
+    root = bind.state_root(acct, st)
+    assert eng.root() == root
+    rng = np.random.default_rng(7)
+    # pick (account, slot) pairs across several accounts
+    aks, sks = [], []
+    keys = sorted(accounts)
+    for i in rng.choice(len(keys), 6, replace=False):
+        ak = keys[int(i)]
+        slots = sorted(accounts[ak][3])
+        sks.append(slots[int(rng.integers(len(slots)))])
+        aks.append(ak)
+    roots, proofs = eng.storage_proof(aks, sks)
+    for ak, sk, sr, nodes in zip(aks, sks, roots, proofs):
+        want_root, want_nodes = pyref.storage_proof(accounts, ak, sk)
+        assert sr == want_root
+        assert nodes == want_nodes, f"storage proof mismatch {ak.hex()}/{sk.hex()}"
+        # replay: walk from the storage root to the slot leaf
+        val = _replay(nodes, sk, sr)
+        assert val == pyref.rlp_int(accounts[ak][3][sk])
+
+
+def test_storage_proof_small_values_inline_nodes(eng):
+    # tiny values force inline (<32 B) deep nodes: exactly the case where
+    # proof lists must SKIP embedded nodes
+    ke = bind.keccak256(b"")
+    ak = bind.keccak256(b"acct")
+    slots = {}
+    for i in range(40):
+        slots[bind.keccak256(b"s" + bytes([i]))] = i + 1  # 1-byte values
+    accounts = {ak: (1, 1, ke, slots)}
+    acct, st = to_arrays(accounts)
+    eng.upload(acct, st)
+    assert eng.root() == bind.state_root(acct, st)
+    sks = sorted(slots)[:5]
+    roots, proofs = eng.storage_proof([ak] * 5, sks)
+    for sk, sr, nodes in zip(sks, roots, proofs):
+        want_root, want_nodes = pyref.storage_proof(accounts, ak, sk)
+        assert sr == want_root
+        assert nodes == want_nodes
+
+
+def test_storage_proof_absent_rejected(eng):
+    acct, st = gen.gen_state_numpy(50, 2, bind.keccak256_batch)
+    eng.upload(acct, st)
+    ak = bytes(acct[0]["key"])
+    with pytest.raises(RuntimeError):
+        eng.storage_proof([ak], [bind.keccak256(b"missing-slot")])
