@@ -3132,13 +3132,16 @@ static int h_storage_leaf(uint8_t *dst, const uint8_t *slot_key, int from,
 {
     uint8_t val[40];
     int vp = h_rlp_uint(val, value_be, 32);
+    // string-wrap the encoded integer: a single byte < 0x80 stays bare
+    int sh = (vp == 1 && val[0] < 0x80) ? 0 : 1;
     uint8_t hp[40];
     int hl = h_hp_item(hp, slot_key, from, 64, 1);
-    int pay = hl + 1 + vp; // value string header is always 1 byte (vp <= 33)
+    int pay = hl + sh + vp;
     int w = h_rlp_list_hdr(dst, pay);
     memcpy(dst + w, hp, hl);
     w += hl;
-    dst[w++] = (uint8_t)(0x80 + vp);
+    if (sh)
+        dst[w++] = (uint8_t)(0x80 + vp);
     memcpy(dst + w, val, vp);
     return w + vp;
 }
