@@ -113,7 +113,10 @@ __device__ void keccak_f(uint64_t s[25])
         0, 1, 62, 28, 27, 36, 44, 6, 55, 20, 3, 10, 43, 25, 39,
         41, 45, 15, 21, 8, 18, 2, 61, 56, 14,
     };
-#pragma unroll 2
+#ifndef KECCAK_UNROLL
+#define KECCAK_UNROLL 2
+#endif
+#pragma unroll KECCAK_UNROLL
     for (int r = 0; r < 24; ++r) {
         uint64_t c[5], d[5], b[25];
 #pragma unroll
