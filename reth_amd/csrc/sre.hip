@@ -856,6 +856,9 @@ __global__ void k_group_flags(const node_rec *__restrict__ L, uint64_t n,
 #define SRE_SCRATCH_ROWMAJOR 0 // measured: col 444.7 ms vs row 456.6 ms @10Mx64
 #endif
 #define SLOT_BR_ROW 576 // row stride, 64-B aligned
+#ifndef SRE_BR_CHUNK_MB
+#define SRE_BR_CHUNK_MB 16 // branch-pipeline chunk, in Mi-groups
+#endif
 
 
 // per-group metadata produced by the assemble kernel
@@ -2347,7 +2350,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
             return -1;
         // 4. branch pipeline: group starts -> assemble -> hash, chunked so
         // the 552-B scratch slots stay bounded.
-        const uint64_t BR_CHUNK = 16ull << 20;
+        const uint64_t BR_CHUNK =
+            (uint64_t)(SRE_BR_CHUNK_MB) << 20;
         HIP_CHECK(ctx, newn.alloc((uint64_t)n_groups * sizeof(node_rec)));
         HIP_CHECK(ctx, gs.alloc(((uint64_t)n_groups + 1) * 4));
         uint64_t chunk = n_groups < BR_CHUNK ? n_groups : BR_CHUNK;
