@@ -194,9 +194,11 @@ int sre_incremental_root(sre_ctx *ctx,
  * Proof::multiproof restricted to account targets
  * (crates/trie/trie/src/proof/mod.rs:59-137 `multiproof`, collecting the
  * RLP of every node on each target's path root-first, the ProofRetainer
- * semantics of the alloy-trie HashBuilder at trie.rs:292-300). v1 proves
- * PRESENT keys only: a target absent from the uploaded state fails with an
- * error (exclusion-proof trimming is follow-up work). Nodes are returned
+ * semantics of the alloy-trie HashBuilder at trie.rs:292-300). Present
+ * targets yield the path to their leaf; ABSENT targets yield the
+ * exclusion proof — the lookup-path nodes ending at the proven divergence
+ * (empty branch slot or mismatching extension/leaf path); an empty state
+ * yields an empty list. Nodes are returned
  * root-first per target, concatenated: out_nodes holds the RLP bytes
  * back-to-back, out_lens one length per node, out_counts one node count
  * per target (in target order). cap_* are capacities in bytes / entries;
@@ -211,11 +213,14 @@ int sre_account_proof(sre_ctx *ctx,
                       uint32_t *out_counts);
 
 /* Storage multiproof — StorageProof::storage_multiproof
- * (crates/trie/trie/src/proof/mod.rs) for PRESENT (acct_key, slot_key)
- * pairs: per target the account's storage root (out_roots, 32 B each) and
- * the root-first node list of its storage trie, same output layout and v1
- * limits as sre_account_proof. Storage tries CAN contain inline (<32 B)
- * nodes; those are embedded in their parents and not emitted. */
+ * (crates/trie/trie/src/proof/mod.rs) for (acct_key, slot_key)
+ * pairs (account must be present; the slot may be absent — exclusion
+ * semantics as in sre_account_proof; a storage-less account yields
+ * EMPTY_ROOT_HASH and an empty list): per target the account's storage
+ * root (out_roots, 32 B each) and the root-first node list of its storage
+ * trie, same output layout as sre_account_proof. Storage tries CAN
+ * contain inline (<32 B) nodes; those are embedded in their parents and
+ * not emitted. */
 int sre_storage_proof(sre_ctx *ctx,
                       const uint8_t *acct_keys /* n x 32 */,
                       const uint8_t *slot_keys /* n x 32 */,

@@ -247,7 +247,8 @@ def _collect_proof(items, pos, target, is_root, nodes):
         # extension node, then its branch child
         if is_root or len(rlp) >= 32:
             nodes.append(rlp)
-        assert target[pos:p] == first[pos:p], "target diverges (absent key)"
+        if target[pos:p] != first[pos:p]:
+            return  # extension path diverges: absence proven here
         brlp = _build(items, p)
         if len(brlp) >= 32:
             nodes.append(brlp)
@@ -265,7 +266,8 @@ def _descend_branch(items, pos, target, nodes):
         while j < len(items) and items[j][0][pos] == nib:
             j += 1
         if nib == target[pos]:
-            assert j > i, "target child absent"
+            if j == i:
+                return  # empty child slot: absence proven at this branch
             _collect_proof(items[i:j], pos + 1, target, False, nodes)
             return
         i = j
@@ -273,7 +275,9 @@ def _descend_branch(items, pos, target, nodes):
 
 
 def account_proof(accounts: dict, hashed_key: bytes):
-    """Proof node list (root-first RLPs) for a PRESENT hashed account key."""
+    """Proof node list (root-first RLPs) for a hashed account key — the
+    lookup-path nodes; for an absent key the list ends at the divergence
+    (exclusion proof). Empty state => empty list."""
     items_d = {}
     for k, (nonce, balance, code_hash, slots) in accounts.items():
         sr = storage_root(slots)
@@ -281,7 +285,8 @@ def account_proof(accounts: dict, hashed_key: bytes):
     items = sorted((tuple(nibbles_of(k)), v) for k, v in items_d.items())
     target = tuple(nibbles_of(hashed_key))
     nodes = []
-    _collect_proof(items, 0, target, True, nodes)
+    if items:
+        _collect_proof(items, 0, target, True, nodes)
     return nodes
 
 
@@ -293,6 +298,9 @@ def storage_proof(accounts: dict, acct_key: bytes, slot_key: bytes):
                    for k, v in slots.items() if v != 0)
     target = tuple(nibbles_of(slot_key))
     nodes = []
-    _collect_proof(items, 0, target, True, nodes)
-    root = keccak256(_build(items, 0))
+    if items:
+        _collect_proof(items, 0, target, True, nodes)
+        root = keccak256(_build(items, 0))
+    else:
+        root = keccak256(b"\x80")  # EMPTY_ROOT_HASH
     return root, nodes

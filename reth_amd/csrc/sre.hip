@@ -42,6 +42,8 @@
 #include <algorithm>
 #include <string>
 #include <vector>
+#include <map>
+#include <array>
 
 #include "../../include/sre.h"
 
@@ -1402,9 +1404,11 @@ struct proof_row {
     uint32_t target;
     int16_t d, P;
     uint32_t br_len;
+    uint32_t s, e;      // member interval (leaf index range)
+    uint8_t key0[32];   // first key of the interval (ext path nibbles)
     uint8_t rlp[536];
 };
-static_assert(sizeof(proof_row) == 548, "proof_row layout");
+static_assert(sizeof(proof_row) == 588, "proof_row layout");
 
 // target key -> index in the sorted account array (+ presence flag)
 __global__ void k_proof_ti(const sre_account_entry *__restrict__ acct,
@@ -1452,35 +1456,12 @@ __global__ void k_proof_ti64(const sre_storage_entry *__restrict__ st,
                      : 0u;
 }
 
-__global__ void k_proof_grab(const node_rec *__restrict__ L,
-                             const uint32_t *__restrict__ gs, uint32_t n_groups,
-                             const br_meta *__restrict__ meta,
-                             const uint8_t *__restrict__ scratch,
-                             uint64_t scratch_stride,
-                             const uint32_t *__restrict__ inv,
-                             const uint32_t *__restrict__ ti, uint32_t n_t,
-                             proof_row *__restrict__ rows,
-                             uint32_t *__restrict__ cnt, uint32_t cap,
-                             uint32_t *__restrict__ err)
+__device__ __forceinline__ void proof_grab_one(
+    const node_rec *L, const uint32_t *gs, const br_meta *meta,
+    const uint8_t *scratch, uint64_t scratch_stride, const uint32_t *inv,
+    const uint8_t *keys, uint64_t key_stride, uint32_t t, uint32_t g,
+    proof_row *rows, uint32_t *cnt, uint32_t cap, uint32_t *err)
 {
-    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
-    if (t >= n_t)
-        return;
-    uint32_t pos = ti[t];
-    // largest group g with interval start <= pos
-    uint32_t lo = 0, hi = n_groups;
-    while (lo < hi) {
-        uint32_t mid = (lo + hi) / 2;
-        if (L[gs[mid]].s <= pos)
-            lo = mid + 1;
-        else
-            hi = mid;
-    }
-    if (lo == 0)
-        return;
-    uint32_t g = lo - 1;
-    if (pos >= L[gs[g + 1] - 1].e)
-        return; // this chunk's groups do not cover the target
     uint32_t slot = inv ? inv[g] : g;
     br_meta mt = meta[slot];
     if (mt.br_len == 0)
@@ -1494,6 +1475,11 @@ __global__ void k_proof_grab(const node_rec *__restrict__ L,
     rows[r].d = (int16_t)mt.d;
     rows[r].P = (int16_t)mt.P;
     rows[r].br_len = mt.br_len;
+    rows[r].s = mt.s;
+    rows[r].e = mt.e;
+    const uint8_t *k0 = keys + (uint64_t)mt.s * key_stride;
+    for (int k = 0; k < 32; ++k)
+        rows[r].key0[k] = k0[k];
     const uint64_t *scr64 = (const uint64_t *)scratch;
     uint64_t *dst = (uint64_t *)rows[r].rlp;
     int nw = (mt.br_len + 7) / 8;
@@ -1503,6 +1489,55 @@ __global__ void k_proof_grab(const node_rec *__restrict__ L,
 #else
         dst[w] = scr64[(uint64_t)w * scratch_stride + slot];
 #endif
+}
+
+// find the chunk-local group covering leaf index pos, or ~0u
+__device__ __forceinline__ uint32_t proof_find_group(const node_rec *L,
+                                                     const uint32_t *gs,
+                                                     uint32_t n_groups,
+                                                     uint32_t pos)
+{
+    uint32_t lo = 0, hi = n_groups;
+    while (lo < hi) {
+        uint32_t mid = (lo + hi) / 2;
+        if (L[gs[mid]].s <= pos)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    if (lo == 0)
+        return ~0u;
+    uint32_t g = lo - 1;
+    if (pos >= L[gs[g + 1] - 1].e)
+        return ~0u;
+    return g;
+}
+
+__global__ void k_proof_grab(const node_rec *__restrict__ L,
+                             const uint32_t *__restrict__ gs, uint32_t n_groups,
+                             const br_meta *__restrict__ meta,
+                             const uint8_t *__restrict__ scratch,
+                             uint64_t scratch_stride,
+                             const uint32_t *__restrict__ inv,
+                             const uint8_t *__restrict__ keys,
+                             uint64_t key_stride,
+                             const uint32_t *__restrict__ ti,
+                             const uint32_t *__restrict__ ti2, uint32_t n_t,
+                             proof_row *__restrict__ rows,
+                             uint32_t *__restrict__ cnt, uint32_t cap,
+                             uint32_t *__restrict__ err)
+{
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= n_t)
+        return;
+    uint32_t g1 = proof_find_group(L, gs, n_groups, ti[t]);
+    uint32_t g2 = ti2 ? proof_find_group(L, gs, n_groups, ti2[t]) : g1;
+    if (g1 != ~0u)
+        proof_grab_one(L, gs, meta, scratch, scratch_stride, inv, keys,
+                       key_stride, t, g1, rows, cnt, cap, err);
+    if (g2 != ~0u && g2 != g1)
+        proof_grab_one(L, gs, meta, scratch, scratch_stride, inv, keys,
+                       key_stride, t, g2, rows, cnt, cap, err);
 }
 
 // old->new position map for interval rebasing: map[i] = the new index of
@@ -2246,7 +2281,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                       // account multiproof capture (sre_account_proof)
                       const uint32_t *d_pti = nullptr, uint32_t n_pt = 0,
                       proof_row *d_prows = nullptr,
-                      uint32_t *d_prow_cnt = nullptr, uint32_t prow_cap = 0)
+                      uint32_t *d_prow_cnt = nullptr, uint32_t prow_cap = 0,
+                      const uint32_t *d_pti2 = nullptr)
 {
     int maxd = -1;
     for (int d = 63; d >= 0; --d)
@@ -2452,8 +2488,9 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                    0, ctx->stream, L, gs.as<uint32_t>() + g0, gc,
                                    mt, scr, chunk,
                                    use_cls ? cinv.as<uint32_t>() + g0 : nullptr,
-                                   d_pti, n_pt, d_prows, d_prow_cnt, prow_cap,
-                                   d_err);
+                                   d_keys, key_stride,
+                                   d_pti, d_pti2, n_pt, d_prows, d_prow_cnt,
+                                   prow_cap, d_err);
                 HIP_CHECK(ctx, hipGetLastError());
             }
             if (updates_kind >= 0) {
@@ -2604,7 +2641,8 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
                             const uint32_t *d_pti = nullptr, uint32_t n_pt = 0,
                             proof_row *d_prows = nullptr,
                             uint32_t *d_prow_cnt = nullptr,
-                            uint32_t prow_cap = 0)
+                            uint32_t prow_cap = 0,
+                            const uint32_t *d_pti2 = nullptr)
 {
     uint64_t ns = ctx->ns, na = ctx->na;
     hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(na)), dim3(BLOCK), 0,
@@ -2683,7 +2721,7 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
                    seg_roots.as<uint8_t>(), nullptr, nullptr, d_err, po,
                    ctx->retain_updates ? 1 : -1, bhash.as<uint8_t>(),
                    0, nullptr, nullptr, 0, d_pti, n_pt, d_prows, d_prow_cnt,
-                   prow_cap))
+                   prow_cap, d_pti2))
         return -1;
     if (ctx->retain_updates && ctx->updates.size() > upd_start) {
         // patch acct_key from the stashed seg ids: seg -> account index ->
@@ -2721,7 +2759,8 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
                             const uint32_t *d_pti = nullptr, uint32_t n_pt = 0,
                             proof_row *d_prows = nullptr,
                             uint32_t *d_prow_cnt = nullptr,
-                            uint32_t prow_cap = 0)
+                            uint32_t prow_cap = 0,
+                            const uint32_t *d_pti2 = nullptr)
 {
     uint64_t na = ctx->na;
     DBuf lcp(ctx), recs(ctx), depths(ctx), hist(ctx);
@@ -2770,7 +2809,7 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
                    subtree, d_roots, d_child_refs, d_child_lens, d_err, po,
                    ctx->retain_updates ? 0 : -1, bhash.as<uint8_t>(),
                    capture_depth, d_cap, d_cap_cnt, cap_capacity,
-                   d_pti, n_pt, d_prows, d_prow_cnt, prow_cap))
+                   d_pti, n_pt, d_prows, d_prow_cnt, prow_cap, d_pti2))
         return -1;
     if (ctx->retain_updates) {
         for (size_t r = upd_start; r < ctx->updates.size(); ++r)
@@ -2979,19 +3018,221 @@ static int h_leaf_node(uint8_t *dst, const uint8_t *key, int from,
     return w + vp;
 }
 
+// ---- proof assembly: RLP list parser + semantic walk -----------------
+// The emitted proof is exactly the node chain a verifier walks from the
+// root by the target's nibbles (eth_getProof / ProofRetainer semantics):
+// present keys end at the target leaf, absent keys end at the first
+// divergence (empty branch slot, mismatching extension/leaf path). Inline
+// (<32 B) nodes are traversed in place and never emitted.
+struct h_item {
+    const uint8_t *pay;
+    int len;
+    bool is_list;
+    const uint8_t *raw;
+    int raw_len;
+};
+
+static int h_rlp_list_items(const uint8_t *b, int len, std::vector<h_item> &out)
+{
+    if (len < 1 || b[0] < 0xc0)
+        return -1;
+    int pl, off;
+    if (b[0] < 0xf8) {
+        pl = b[0] - 0xc0;
+        off = 1;
+    } else {
+        int n = b[0] - 0xf7;
+        pl = 0;
+        for (int k = 0; k < n; ++k)
+            pl = (pl << 8) | b[1 + k];
+        off = 1 + n;
+    }
+    if (off + pl != len)
+        return -1;
+    int i = off, end = off + pl;
+    while (i < end) {
+        uint8_t c = b[i];
+        h_item it{};
+        if (c < 0x80) {
+            it.pay = b + i;
+            it.len = 1;
+            i += 1;
+        } else if (c < 0xb8) {
+            it.pay = b + i + 1;
+            it.len = c - 0x80;
+            i += 1 + it.len;
+        } else if (c < 0xc0) {
+            int n = c - 0xb7;
+            int ln = 0;
+            for (int k = 0; k < n; ++k)
+                ln = (ln << 8) | b[i + 1 + k];
+            it.pay = b + i + 1 + n;
+            it.len = ln;
+            i += 1 + n + ln;
+        } else {
+            int hn, ln;
+            if (c < 0xf8) {
+                hn = 1;
+                ln = c - 0xc0;
+            } else {
+                hn = 1 + (c - 0xf7);
+                ln = 0;
+                for (int k = 0; k < hn - 1; ++k)
+                    ln = (ln << 8) | b[i + 1 + k];
+            }
+            it.is_list = true;
+            it.raw = b + i;
+            it.raw_len = hn + ln;
+            i += hn + ln;
+        }
+        if (i > end)
+            return -1;
+        out.push_back(it);
+    }
+    return 0;
+}
+
+typedef std::array<uint8_t, 32> hkey;
+typedef std::map<hkey, std::pair<const uint8_t *, int>> node_map;
+
+// returns 0 = reached the target leaf (present), 1 = divergence proven
+// (absent), -1 = internal error. Emits each hash-referenced node visited.
+template <typename EmitFn>
+static int proof_walk_emit(sre_ctx *ctx, const uint8_t root[32],
+                           const node_map &byhash, const uint8_t *key,
+                           EmitFn &&emit)
+{
+    uint8_t nib[64];
+    for (int k = 0; k < 32; ++k) {
+        nib[2 * k] = key[k] >> 4;
+        nib[2 * k + 1] = key[k] & 0xF;
+    }
+    hkey rk;
+    memcpy(rk.data(), root, 32);
+    auto it = byhash.find(rk);
+    if (it == byhash.end()) {
+        set_err(ctx, "proof: root node missing from capture");
+        return -1;
+    }
+    const uint8_t *cur = it->second.first;
+    int clen = it->second.second;
+    bool emit_cur = true;
+    int pos = 0;
+    for (int guard = 0; guard < 200; ++guard) {
+        if (emit_cur && emit(cur, clen))
+            return -1;
+        std::vector<h_item> items;
+        if (h_rlp_list_items(cur, clen, items)) {
+            set_err(ctx, "proof: malformed node");
+            return -1;
+        }
+        const h_item *next = nullptr;
+        if (items.size() == 17) {
+            if (pos >= 64) {
+                set_err(ctx, "proof: branch below leaf depth");
+                return -1;
+            }
+            const h_item &c = items[nib[pos]];
+            if (!c.is_list && c.len == 0)
+                return 1; // empty child slot: absence proven
+            pos++;
+            next = &c;
+        } else if (items.size() == 2) {
+            const uint8_t *hp = items[0].pay;
+            int hl = items[0].len;
+            int flag = hp[0] >> 4;
+            bool is_leaf = (flag & 2) != 0;
+            int np = (hl - 1) * 2 + ((flag & 1) ? 1 : 0);
+            bool match = pos + np <= 64;
+            for (int k = 0; match && k < np; ++k) {
+                int v = ((flag & 1) == 0)
+                            ? ((k & 1) ? hp[1 + k / 2] & 0xF
+                                       : hp[1 + k / 2] >> 4)
+                            : (k == 0 ? hp[0] & 0xF
+                                      : ((k & 1) ? hp[1 + (k - 1) / 2] >> 4
+                                                 : hp[1 + (k - 1) / 2] & 0xF));
+                match = v == nib[pos + k];
+            }
+            if (!match)
+                return 1; // path mismatch: absence proven
+            pos += np;
+            if (is_leaf) {
+                if (pos != 64) {
+                    set_err(ctx, "proof: leaf at wrong depth");
+                    return -1;
+                }
+                return 0; // target leaf reached
+            }
+            next = &items[1];
+        } else {
+            set_err(ctx, "proof: unexpected node arity");
+            return -1;
+        }
+        if (!next->is_list && next->len == 32) {
+            hkey k;
+            memcpy(k.data(), next->pay, 32);
+            auto jt = byhash.find(k);
+            if (jt == byhash.end()) {
+                set_err(ctx, "proof: child node missing from capture");
+                return -1;
+            }
+            cur = jt->second.first;
+            clen = jt->second.second;
+            emit_cur = true;
+        } else { // inline embedded node: traverse, do not emit
+            cur = next->is_list ? next->raw : next->pay;
+            clen = next->is_list ? next->raw_len : next->len;
+            emit_cur = false;
+        }
+    }
+    set_err(ctx, "proof: walk did not terminate");
+    return -1;
+}
+
+// build the ext-wrapper bytes for a captured row (path nibbles from the
+// row's own first key — valid for on-path and sibling rows alike)
+static void h_row_ext(const proof_row *r, std::vector<uint8_t> &out)
+{
+    uint8_t cref[33];
+    int crl;
+    if (r->br_len >= 32) {
+        cref[0] = 0xa0;
+        h_keccak256(r->rlp, r->br_len, cref + 1);
+        crl = 33;
+    } else {
+        memcpy(cref, r->rlp, r->br_len);
+        crl = (int)r->br_len;
+    }
+    uint8_t hp[40];
+    int hl = h_hp_item(hp, r->key0, r->P + 1, r->d, 0);
+    out.resize(2 + hl + crl);
+    int w = h_rlp_list_hdr(out.data(), hl + crl);
+    memcpy(out.data() + w, hp, hl);
+    memcpy(out.data() + w + hl, cref, crl);
+    out.resize(w + hl + crl);
+}
+
+static void map_add(node_map &m, const uint8_t *b, int len)
+{
+    hkey k;
+    h_keccak256(b, len, k.data());
+    m.emplace(k, std::make_pair(b, len));
+}
+
 extern "C" int sre_account_proof(sre_ctx *ctx, const uint8_t *targets,
                                  uint64_t n_targets, uint8_t *out_nodes,
                                  uint64_t cap_nodes, uint32_t *out_lens,
                                  uint64_t cap_lens, uint32_t *out_counts)
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
-    if (ctx->na == 0) {
-        set_err(ctx, "sre_account_proof: empty state (target cannot be present)");
-        return -1;
-    }
     if (n_targets == 0 || n_targets > 4096) {
         set_err(ctx, "sre_account_proof: 1..4096 targets");
         return -1;
+    }
+    if (ctx->na == 0) { // empty trie: every proof is the empty node list
+        for (uint64_t t = 0; t < n_targets; ++t)
+            out_counts[t] = 0;
+        return 0;
     }
     uint32_t n_t = (uint32_t)n_targets;
     DBuf err(ctx), acct_roots(ctx), root(ctx), dtgt(ctx), dti(ctx), dpres(ctx),
@@ -3014,13 +3255,19 @@ extern "C" int sre_account_proof(sre_ctx *ctx, const uint8_t *targets,
                              hipMemcpyDeviceToHost));
     HIP_CHECK(ctx, hipMemcpy(pres.data(), dpres.p, 4ull * n_t,
                              hipMemcpyDeviceToHost));
-    for (uint32_t t = 0; t < n_t; ++t)
-        if (!pres[t]) {
-            set_err(ctx, "sre_account_proof: target key not present "
-                         "(exclusion proofs are not supported in v1)");
-            return -1;
-        }
-    uint32_t cap_rows = n_t * 130 + 64;
+    // exclusion targets also need the LEFT neighbour's path: a target that
+    // sorts after every key of the divergent subtree has its lookup path
+    // under index ti-1, not ti
+    DBuf dti2(ctx);
+    HIP_CHECK(ctx, dti2.alloc(4ull * n_t));
+    {
+        std::vector<uint32_t> ti2(n_t);
+        for (uint32_t t = 0; t < n_t; ++t)
+            ti2[t] = (!pres[t] && ti[t] > 0) ? ti[t] - 1 : ti[t];
+        HIP_CHECK(ctx, hipMemcpyAsync(dti2.p, ti2.data(), 4ull * n_t,
+                                      hipMemcpyHostToDevice, ctx->stream));
+    }
+    uint32_t cap_rows = n_t * 260 + 64;
     HIP_CHECK(ctx, prows.alloc((uint64_t)cap_rows * sizeof(proof_row)));
     HIP_CHECK(ctx, prowc.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(prowc.p, 0, 4, ctx->stream));
@@ -3031,7 +3278,8 @@ extern "C" int sre_account_proof(sre_ctx *ctx, const uint8_t *targets,
     if (run_account_pass(ctx, acct_roots.as<uint8_t>(), 0, root.as<uint8_t>(),
                          nullptr, nullptr, &po, err.as<uint32_t>(), 0, nullptr,
                          nullptr, 0, dti.as<uint32_t>(), n_t,
-                         prows.as<proof_row>(), prowc.as<uint32_t>(), cap_rows))
+                         prows.as<proof_row>(), prowc.as<uint32_t>(), cap_rows,
+                         dti2.as<uint32_t>()))
         return -1;
     if (check_err(ctx, err.as<uint32_t>()))
         return -1;
@@ -3058,6 +3306,46 @@ extern "C" int sre_account_proof(sre_ctx *ctx, const uint8_t *targets,
     uint64_t nb = 0, nl = 0;
     for (uint32_t t = 0; t < n_t; ++t) {
         const uint8_t *key = targets + 32ull * t;
+        // node universe for this target: captured branches, their ext
+        // wrappers, and the candidate leaves at ti / ti-1
+        node_map byhash;
+        std::vector<std::vector<uint8_t>> owned;
+        owned.reserve(2 * per[t].size() + 2);
+        for (const proof_row *r : per[t]) {
+            map_add(byhash, r->rlp, (int)r->br_len);
+            if (r->d > r->P + 1) {
+                owned.emplace_back();
+                h_row_ext(r, owned.back());
+                map_add(byhash, owned.back().data(), (int)owned.back().size());
+            }
+        }
+        uint32_t cands[2];
+        int ncand = 0;
+        cands[ncand++] = ti[t] < ctx->na ? ti[t] : (uint32_t)(ctx->na - 1);
+        if (!pres[t] && ti[t] > 0 && ti[t] - 1 != cands[0])
+            cands[ncand++] = ti[t] - 1;
+        for (int c = 0; c < ncand; ++c) {
+            uint32_t idx = cands[c];
+            int dmax = -1;
+            for (const proof_row *r : per[t])
+                if (r->s <= idx && idx < r->e && r->d > dmax)
+                    dmax = r->d;
+            sre_account_entry ae;
+            uint8_t sroot[32];
+            HIP_CHECK(ctx,
+                      hipMemcpy(&ae,
+                                (const uint8_t *)ctx->d_acct +
+                                    (uint64_t)idx * sizeof(sre_account_entry),
+                                sizeof(ae), hipMemcpyDeviceToHost));
+            HIP_CHECK(ctx, hipMemcpy(sroot,
+                                     acct_roots.as<uint8_t>() + 32ull * idx,
+                                     32, hipMemcpyDeviceToHost));
+            owned.emplace_back(200);
+            int ll = h_leaf_node(owned.back().data(), ae.key, dmax + 1, &ae,
+                                 sroot);
+            owned.back().resize(ll);
+            map_add(byhash, owned.back().data(), ll);
+        }
         uint32_t cnt = 0;
         auto emit = [&](const uint8_t *node, int len) -> int {
             if (nl >= cap_lens || nb + (uint64_t)len > cap_nodes) {
@@ -3070,65 +3358,14 @@ extern "C" int sre_account_proof(sre_ctx *ctx, const uint8_t *targets,
             cnt++;
             return 0;
         };
-        bool first = true;
-        for (const proof_row *r : per[t]) {
-            if (r->d > r->P + 1) { // extension above this branch
-                uint8_t ext[80];
-                uint8_t cref[33];
-                int crl;
-                if (r->br_len >= 32) {
-                    cref[0] = 0xa0;
-                    h_keccak256(r->rlp, r->br_len, cref + 1);
-                    crl = 33;
-                } else { // inline child (never in account tries; defensive)
-                    memcpy(cref, r->rlp, r->br_len);
-                    crl = (int)r->br_len;
-                }
-                uint8_t hp[40];
-                int hl = h_hp_item(hp, key, r->P + 1, r->d, 0);
-                int pay = hl + crl;
-                int w = h_rlp_list_hdr(ext, pay);
-                memcpy(ext + w, hp, hl);
-                memcpy(ext + w + hl, cref, crl);
-                // inline (<32 B, non-root) nodes are embedded in their
-                // parent and are not separate proof entries (never hit in
-                // account tries; defensive exactness)
-                if ((w + pay >= 32 || r->P < 0) && emit(ext, w + pay))
-                    return -1;
-            }
-            if ((r->br_len >= 32 || (r->P < 0 && r->d == r->P + 1)) &&
-                emit(r->rlp, (int)r->br_len))
-                return -1;
-            first = false;
-        }
-        (void)first;
-        // the account leaf itself
-        int dmax = per[t].empty() ? -1 : per[t].back()->d;
-        sre_account_entry ae;
-        uint8_t sroot[32];
-        HIP_CHECK(ctx, hipMemcpy(&ae,
-                                 (const uint8_t *)ctx->d_acct +
-                                     (uint64_t)ti[t] * sizeof(sre_account_entry),
-                                 sizeof(ae), hipMemcpyDeviceToHost));
-        HIP_CHECK(ctx, hipMemcpy(sroot,
-                                 acct_roots.as<uint8_t>() + 32ull * ti[t], 32,
-                                 hipMemcpyDeviceToHost));
-        uint8_t leaf[200];
-        int ll = h_leaf_node(leaf, key, dmax + 1, &ae, sroot);
-        if (emit(leaf, ll))
+        int rc = proof_walk_emit(ctx, engine_root, byhash, key, emit);
+        if (rc < 0)
             return -1;
+        if ((rc == 0) != (pres[t] != 0)) {
+            set_err(ctx, "sre_account_proof: internal presence mismatch");
+            return -1;
+        }
         out_counts[t] = cnt;
-        // sanity: the first node must hash to the engine's own root
-        uint8_t h[32];
-        uint64_t firstoff = nb;
-        uint32_t firstlen = out_lens[nl - cnt];
-        for (uint32_t k2 = nl - cnt; k2 < nl; ++k2)
-            firstoff -= out_lens[k2];
-        h_keccak256(out_nodes + firstoff, firstlen, h);
-        if (memcmp(h, engine_root, 32) != 0) {
-            set_err(ctx, "sre_account_proof: internal root mismatch");
-            return -1;
-        }
     }
     return 0;
 }
@@ -3213,12 +3450,20 @@ extern "C" int sre_storage_proof(sre_ctx *ctx, const uint8_t *acct_keys,
     HIP_CHECK(ctx, hipMemcpy(presa.data(), dpra.p, 4ull * n_t,
                              hipMemcpyDeviceToHost));
     for (uint32_t t = 0; t < n_t; ++t)
-        if (!pres[t] || !presa[t]) {
-            set_err(ctx, "sre_storage_proof: (account, slot) not present "
-                         "(exclusion proofs are not supported in v1)");
+        if (!presa[t]) {
+            set_err(ctx, "sre_storage_proof: account not present");
             return -1;
         }
-    uint32_t cap_rows = n_t * 130 + 64;
+    DBuf dti2(ctx);
+    HIP_CHECK(ctx, dti2.alloc(4ull * n_t));
+    {
+        std::vector<uint32_t> ti2(n_t);
+        for (uint32_t t = 0; t < n_t; ++t)
+            ti2[t] = (!pres[t] && ti[t] > 0) ? ti[t] - 1 : ti[t];
+        HIP_CHECK(ctx, hipMemcpyAsync(dti2.p, ti2.data(), 4ull * n_t,
+                                      hipMemcpyHostToDevice, ctx->stream));
+    }
+    uint32_t cap_rows = n_t * 260 + 64;
     HIP_CHECK(ctx, prows.alloc((uint64_t)cap_rows * sizeof(proof_row)));
     HIP_CHECK(ctx, prowc.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(prowc.p, 0, 4, ctx->stream));
@@ -3226,7 +3471,7 @@ extern "C" int sre_storage_proof(sre_ctx *ctx, const uint8_t *acct_keys,
     pass_out po;
     if (run_storage_pass(ctx, acct_roots.as<uint8_t>(), &po, err.as<uint32_t>(),
                          dti.as<uint32_t>(), n_t, prows.as<proof_row>(),
-                         prowc.as<uint32_t>(), cap_rows))
+                         prowc.as<uint32_t>(), cap_rows, dti2.as<uint32_t>()))
         return -1;
     if (check_err(ctx, err.as<uint32_t>()))
         return -1;
@@ -3247,6 +3492,10 @@ extern "C" int sre_storage_proof(sre_ctx *ctx, const uint8_t *acct_keys,
                   });
 
     uint64_t nb = 0, nl = 0;
+    static const uint8_t EMPTY_ROOT_H2[32] = {
+        0x56, 0xe8, 0x1f, 0x17, 0x1b, 0xcc, 0x55, 0xa6, 0xff, 0x83, 0x45,
+        0xe6, 0x92, 0xc0, 0xf8, 0x6e, 0x5b, 0x48, 0xe0, 0x1b, 0x99, 0x6c,
+        0xad, 0xc0, 0x01, 0x62, 0x2f, 0xb5, 0xe3, 0x63, 0xb4, 0x21};
     for (uint32_t t = 0; t < n_t; ++t) {
         const uint8_t *key = slot_keys + 32ull * t;
         uint8_t sroot[32];
@@ -3254,6 +3503,54 @@ extern "C" int sre_storage_proof(sre_ctx *ctx, const uint8_t *acct_keys,
                                  acct_roots.as<uint8_t>() + 32ull * tia[t], 32,
                                  hipMemcpyDeviceToHost));
         memcpy(out_roots + 32ull * t, sroot, 32);
+        if (memcmp(sroot, EMPTY_ROOT_H2, 32) == 0) {
+            // account has no storage: the empty node list proves absence
+            if (pres[t]) {
+                set_err(ctx, "sre_storage_proof: internal presence mismatch");
+                return -1;
+            }
+            out_counts[t] = 0;
+            continue;
+        }
+        node_map byhash;
+        std::vector<std::vector<uint8_t>> owned;
+        owned.reserve(2 * per[t].size() + 2);
+        for (const proof_row *r : per[t]) {
+            map_add(byhash, r->rlp, (int)r->br_len);
+            if (r->d > r->P + 1) {
+                owned.emplace_back();
+                h_row_ext(r, owned.back());
+                map_add(byhash, owned.back().data(), (int)owned.back().size());
+            }
+        }
+        // candidate slot leaves at ti / ti-1, guarded to THIS account's
+        // segment (a neighbour index may belong to another storage trie)
+        uint32_t cands[2];
+        int ncand = 0;
+        if (ti[t] < ctx->ns)
+            cands[ncand++] = ti[t];
+        if (!pres[t] && ti[t] > 0)
+            cands[ncand++] = ti[t] - 1;
+        for (int c = 0; c < ncand; ++c) {
+            uint32_t idx = cands[c];
+            sre_storage_entry se;
+            HIP_CHECK(ctx,
+                      hipMemcpy(&se,
+                                (const uint8_t *)ctx->d_st +
+                                    (uint64_t)idx * sizeof(sre_storage_entry),
+                                sizeof(se), hipMemcpyDeviceToHost));
+            if (memcmp(se.acct_key, acct_keys + 32ull * t, 32) != 0)
+                continue; // other account's segment
+            int dmax = -1;
+            for (const proof_row *r : per[t])
+                if (r->s <= idx && idx < r->e && r->d > dmax)
+                    dmax = r->d;
+            owned.emplace_back(96);
+            int ll = h_storage_leaf(owned.back().data(), se.slot_key,
+                                    dmax + 1, se.value);
+            owned.back().resize(ll);
+            map_add(byhash, owned.back().data(), ll);
+        }
         uint32_t cnt = 0;
         auto emit = [&](const uint8_t *node, int len) -> int {
             if (nl >= cap_lens || nb + (uint64_t)len > cap_nodes) {
@@ -3266,52 +3563,14 @@ extern "C" int sre_storage_proof(sre_ctx *ctx, const uint8_t *acct_keys,
             cnt++;
             return 0;
         };
-        for (const proof_row *r : per[t]) {
-            if (r->d > r->P + 1) {
-                uint8_t ext[80];
-                uint8_t cref[33];
-                int crl;
-                if (r->br_len >= 32) {
-                    cref[0] = 0xa0;
-                    h_keccak256(r->rlp, r->br_len, cref + 1);
-                    crl = 33;
-                } else {
-                    memcpy(cref, r->rlp, r->br_len);
-                    crl = (int)r->br_len;
-                }
-                uint8_t hp[40];
-                int hl = h_hp_item(hp, key, r->P + 1, r->d, 0);
-                int pay = hl + crl;
-                int w = h_rlp_list_hdr(ext, pay);
-                memcpy(ext + w, hp, hl);
-                memcpy(ext + w + hl, cref, crl);
-                if ((w + pay >= 32 || r->P < 0) && emit(ext, w + pay))
-                    return -1;
-            }
-            if ((r->br_len >= 32 || (r->P < 0 && r->d == r->P + 1)) &&
-                emit(r->rlp, (int)r->br_len))
-                return -1;
-        }
-        int dmax = per[t].empty() ? -1 : per[t].back()->d;
-        sre_storage_entry se;
-        HIP_CHECK(ctx, hipMemcpy(&se,
-                                 (const uint8_t *)ctx->d_st +
-                                     (uint64_t)ti[t] * sizeof(sre_storage_entry),
-                                 sizeof(se), hipMemcpyDeviceToHost));
-        uint8_t leaf[80];
-        int ll = h_storage_leaf(leaf, key, dmax + 1, se.value);
-        if ((ll >= 32 || per[t].empty()) && emit(leaf, ll))
+        int rc = proof_walk_emit(ctx, sroot, byhash, key, emit);
+        if (rc < 0)
             return -1;
+        if ((rc == 0) != (pres[t] != 0)) {
+            set_err(ctx, "sre_storage_proof: internal presence mismatch");
+            return -1;
+        }
         out_counts[t] = cnt;
-        uint64_t firstoff = nb;
-        for (uint32_t k2 = nl - cnt; k2 < nl; ++k2)
-            firstoff -= out_lens[k2];
-        uint8_t h[32];
-        h_keccak256(out_nodes + firstoff, out_lens[nl - cnt], h);
-        if (memcmp(h, sroot, 32) != 0) {
-            set_err(ctx, "sre_storage_proof: internal root mismatch");
-            return -1;
-        }
     }
     return 0;
 }

@@ -214,10 +214,11 @@ class StateRootEngine:
         return bytes(out), rows
 
     def account_proof(self, targets) -> list:
-        """Account multiproof for PRESENT hashed keys: per target, the list
-        of proof node RLPs root-first (Proof::multiproof account surface,
-        include/sre.h sre_account_proof). targets: list of 32-byte keys.
-        Returns list-of-list-of-bytes in target order."""
+        """Account multiproof: per target, the proof node RLPs root-first
+        (Proof::multiproof account surface, include/sre.h
+        sre_account_proof). Present keys get the path to their leaf;
+        absent keys get the exclusion proof ending at the divergence.
+        targets: list of 32-byte hashed keys."""
         n = len(targets)
         tarr = np.frombuffer(b"".join(targets), dtype=np.uint8).reshape(n, 32)
         tarr = np.ascontiguousarray(tarr)
@@ -240,9 +241,10 @@ class StateRootEngine:
         return out
 
     def storage_proof(self, acct_keys, slot_keys):
-        """Storage multiproof for PRESENT (account, slot) pairs: returns
-        (roots, proofs) — per target the account's storage root (bytes) and
-        the root-first node-RLP list of its storage trie."""
+        """Storage multiproof for (account, slot) pairs (account must be
+        present; the slot may be absent -> exclusion proof): returns
+        (roots, proofs) — per target the account's storage root and the
+        root-first node-RLP list of its storage trie."""
         n = len(acct_keys)
         assert len(slot_keys) == n
         aarr = np.ascontiguousarray(

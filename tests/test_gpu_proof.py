@@ -75,7 +75,8 @@ def _replay(nodes, key, root):
         items = rlp_items(node)
         if len(items) == 17:
             ref = items[nib[pos]]
-            assert ref, "target child absent in branch"
+            if not ref:
+                return None  # empty child slot: absence proven
             pos += 1
             node = by_hash[ref] if len(ref) == 32 else ref
         else:
@@ -85,7 +86,8 @@ def _replay(nodes, key, root):
             path = ([hp[0] & 0xF] if odd else [])
             for byte in hp[1:]:
                 path += [byte >> 4, byte & 0xF]
-            assert nib[pos:pos + len(path)] == path, "path mismatch"
+            if nib[pos:pos + len(path)] != path:
+                return None  # path divergence: absence proven
             pos += len(path)
             if flag & 2:  # leaf
                 assert pos == 64
@@ -146,11 +148,59 @@ def test_proof_small_tries(eng):
     _check_state(eng, acc, sorted(acc))
 
 
-def test_proof_absent_key_rejected(eng):
-    acct, st = gen.gen_state_numpy(100, 0, bind.keccak256_batch)
+def test_proof_absent_keys_exclusion(eng):
+    acct, st = gen.gen_state_numpy(2000, 0, bind.keccak256_batch)
+    accounts = _dict_of(acct, st)
     eng.upload(acct, st)
-    with pytest.raises(RuntimeError):
-        eng.account_proof([bind.keccak256(b"definitely-absent")])
+    root = bind.state_root(acct, st)
+    assert eng.root() == root
+    absent = [bind.keccak256(b"definitely-absent" + bytes([i]))
+              for i in range(6)]
+    # mixed present/absent multiproof
+    keys = sorted(accounts)
+    targets = [keys[3]] + absent + [keys[-1]]
+    proofs = eng.account_proof(targets)
+    for k, nodes in zip(targets, proofs):
+        assert nodes == pyref.account_proof(accounts, k), k.hex()
+        got = _replay(nodes, k, root)
+        if k in accounts:
+            assert got is not None
+        else:
+            assert got is None  # replay must END at a proven divergence
+
+
+def test_storage_proof_absent_exclusion(eng):
+    acct, st = gen.gen_state_numpy(200, 8, bind.keccak256_batch)
+    accounts = _dict_of(acct, st)
+    eng.upload(acct, st)
+    assert eng.root() == bind.state_root(acct, st)
+    keys = sorted(accounts)
+    ak = keys[5]
+    absents = [bind.keccak256(b"no-such-slot" + bytes([i])) for i in range(3)]
+    pres = sorted(accounts[ak][3])[0]
+    aks = [ak] * 4
+    sks = absents + [pres]
+    roots, proofs = eng.storage_proof(aks, sks)
+    for sk, sr, nodes in zip(sks, roots, proofs):
+        want_root, want_nodes = pyref.storage_proof(accounts, ak, sk)
+        assert sr == want_root
+        assert nodes == want_nodes, sk.hex()
+        got = _replay(nodes, sk, sr)
+        assert (got is None) == (sk not in accounts[ak][3])
+
+
+def test_storage_proof_slotless_account(eng):
+    ke = bind.keccak256(b"")
+    a1, a2 = sorted([bind.keccak256(b"p"), bind.keccak256(b"q")])
+    sk = bind.keccak256(b"s1")
+    accounts = {a1: (1, 2, ke, {}), a2: (3, 4, ke, {sk: 9})}
+    acct, st = to_arrays(accounts)
+    eng.upload(acct, st)
+    # a1 has no storage: exclusion proof of any slot is the empty list
+    roots, proofs = eng.storage_proof([a1], [sk])
+    assert roots[0] == bytes.fromhex(
+        "56e81f171bcc55a6ff8345e692c0f86e5b48e01b996cadc001622fb5e363b421")
+    assert proofs[0] == []
 
 
 def test_storage_proof_parity(eng):

@@ -2,8 +2,6 @@
 storage_proof): every proof verifies by replay against the trie root, with
 inline-node skipping per eth_getProof semantics. (Engine parity is in
 tests/test_gpu_proof.py.)"""
-import pytest
-
 from oracle import bind, pyref
 from reth_amd import gen
 from tests.test_gpu_proof import _dict_of, _replay
@@ -33,8 +31,13 @@ def test_storage_proofs_replay_incl_inline():
         assert _replay(nodes, sk, root) == pyref.rlp_int(slots[sk])
 
 
-def test_absent_key_asserts():
+def test_absent_key_exclusion_replay():
     acct, st = gen.gen_state_numpy(50, 0, bind.keccak256_batch)
     accounts = _dict_of(acct, st)
-    with pytest.raises(AssertionError):
-        pyref.account_proof(accounts, bind.keccak256(b"nope"))
+    root = bind.state_root(acct, st)
+    for i in range(5):
+        k = bind.keccak256(b"nope" + bytes([i]))
+        nodes = pyref.account_proof(accounts, k)
+        assert nodes and bind.keccak256(nodes[0]) == root
+        assert _replay(nodes, k, root) is None  # ends at proven divergence
+    assert pyref.account_proof({}, bind.keccak256(b"x")) == []
