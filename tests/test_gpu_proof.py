@@ -246,11 +246,3 @@ def test_storage_proof_small_values_inline_nodes(eng):
         want_root, want_nodes = pyref.storage_proof(accounts, ak, sk)
         assert sr == want_root
         assert nodes == want_nodes
-
-
-def test_storage_proof_absent_rejected(eng):
-    acct, st = gen.gen_state_numpy(50, 2, bind.keccak256_batch)
-    eng.upload(acct, st)
-    ak = bytes(acct[0]["key"])
-    with pytest.raises(RuntimeError):
-        eng.storage_proof([ak], [bind.keccak256(b"missing-slot")])
