@@ -61,10 +61,12 @@ def test_fuzz_roots_updates_proofs(eng):
         keys = sorted(accounts)
         tsel = [keys[int(i)] for i in
                 rng.choice(len(keys), min(3, len(keys)), replace=False)]
+        tsel += [bind.keccak256(b"absent" + bytes([it, j])) for j in range(2)]
         proofs = eng.account_proof(tsel)
         for k, nodes in zip(tsel, proofs):
             assert nodes == pyref.account_proof(accounts, k), f"iter {it}"
-            _replay(nodes, k, root)
+            got = _replay(nodes, k, root)
+            assert (got is None) == (k not in accounts), f"iter {it}"
 
 
 def test_fuzz_incremental_chains(eng):
