@@ -659,7 +659,6 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
 // level machinery: selection / partition / merge / grouping
 // ---------------------------------------------------------------------------
 
-#define SEL_ITEMS 16
 
 // parallel exclusive scan of 256 per-thread counts in LDS; returns (excl,
 // total) — Hillis-Steele, 8 steps.
@@ -2214,11 +2213,6 @@ static inline uint32_t grid_for(uint64_t n)
     return (uint32_t)((n + BLOCK - 1) / BLOCK);
 }
 
-static inline uint32_t sel_grid_for(uint64_t n)
-{
-    uint64_t per = (uint64_t)BLOCK * SEL_ITEMS;
-    return (uint32_t)((n + per - 1) / per);
-}
 
 // exclusive scan of u32 in[n] -> out[n]; total returned synchronously.
 static int scan_u32(sre_ctx *ctx, const uint32_t *d_in, uint32_t *d_out, uint64_t n,
