@@ -213,27 +213,43 @@ def main():
     hash_blocks = stats["leaf_blocks"] + stats["branch_blocks"]
     hash_s = (stats["leaf_hash_ms"] + stats["branch_hash_ms"]) / 1000.0
 
-    # ---- CPU baseline: oracle on a bounded sample, scaled by leaf count ----
+    # ---- CPU baseline: oracle on a bounded sample, scaled by leaf count.
+    # Reported figure = one thread per host core (okc_state_root_par, the
+    # stronger and more honest comparison); the single-thread time is kept
+    # in cpu_baseline_1core. Both are the same algorithm (kind "port").
     cpu_baseline = None
+    cpu_baseline_1core = None
     if world == 1 and not args.no_cpu_baseline:
+        import os as _os
         from oracle import bind
         sa = min(args.cpu_sample_accounts, args.accounts)
         acct_s, st_s = gen.gen_state_numpy(sa, args.slots, bind.keccak256_batch)
         c0 = time.perf_counter()
-        bind.state_root(acct_s, st_s)
+        r1 = bind.state_root(acct_s, st_s)
         c1 = time.perf_counter()
+        rp = bind.state_root_par(acct_s, st_s)
+        c2 = time.perf_counter()
+        assert r1 == rp
+        ncores = _os.cpu_count() or 1
         sample_leaves = len(acct_s) + len(st_s)
         full_leaves = total_accounts + total_storage_leaves
-        cpu_ms = (c1 - c0) * 1000.0 * full_leaves / sample_leaves
+        scale = full_leaves / sample_leaves
         cpu_baseline = {
-            "value": round(cpu_ms, 1),
+            "value": round((c2 - c1) * 1000.0 * scale, 1),
+            "unit": "ms",
+            "cores": ncores,
+            "kind": "port",
+            "sample": f"oracle (C, OpenMP, {ncores} threads) on {sa} accounts"
+                      f" x {args.slots} slots = {sample_leaves} leaves, "
+                      f"{(c2 - c1):.1f}s measured, scaled linearly in leaves "
+                      f"to the full {full_leaves}-leaf job",
+        }
+        cpu_baseline_1core = {
+            "value": round((c1 - c0) * 1000.0 * scale, 1),
             "unit": "ms",
             "cores": 1,
             "kind": "port",
-            "sample": f"oracle (C, single thread) on {sa} accounts x "
-                      f"{args.slots} slots = {sample_leaves} leaves, "
-                      f"{(c1 - c0):.1f}s measured, scaled linearly in leaves "
-                      f"to the full {full_leaves}-leaf job",
+            "sample": f"same sample, single thread ({(c1 - c0):.1f}s)",
         }
 
     out = {
@@ -288,6 +304,7 @@ def main():
                     "full-rebuild bench line for the kernel roofline).",
         },
         "cpu_baseline": cpu_baseline,
+        "cpu_baseline_1core": cpu_baseline_1core,
         "keccak_ghs_leaf_kernel": round(keccak_ghs, 2),
         "keccak_ghs_all_kernels": round(hash_blocks / hash_s / 1e9, 2)
         if hash_s > 0 else 0.0,

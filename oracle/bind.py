@@ -98,6 +98,18 @@ def state_root(accounts: np.ndarray, storage: np.ndarray) -> bytes:
     return bytes(out)
 
 
+def state_root_par(accounts: np.ndarray, storage: np.ndarray,
+                   nthreads: int = 0) -> bytes:
+    """Multi-threaded (OpenMP) state root — the one-thread-per-core CPU
+    baseline leg of bench.py. nthreads 0 = all cores."""
+    assert accounts.dtype == ACCOUNT_DTYPE and storage.dtype == STORAGE_DTYPE
+    out = (ctypes.c_uint8 * 32)()
+    _check(lib().okc_state_root_par(_ptr(accounts), len(accounts),
+                                    _ptr(storage), len(storage),
+                                    ctypes.c_int(nthreads), out))
+    return bytes(out)
+
+
 def storage_roots(accounts: np.ndarray, storage: np.ndarray) -> np.ndarray:
     out = np.empty((len(accounts), 32), dtype=np.uint8)
     _check(lib().okc_storage_roots(_ptr(accounts), len(accounts),

@@ -612,3 +612,130 @@ int okc_finish_top(const uint8_t child_refs[16][33], const uint8_t child_lens[16
     okc_keccak256(rlp, h + w, out);
     return 0;
 }
+
+/* ---- multi-threaded state root (OpenMP) ------------------------------
+ * The "one thread per core" CPU baseline BASELINE.md's plan names: the
+ * same algorithm, storage-trie roots parallel over accounts and the
+ * account trie split by top nibble (the okc_subtree_roots decomposition)
+ * then finished with okc_finish_top. Used ONLY by bench.py's cpu_baseline
+ * leg — test infrastructure, never the product path. */
+#ifdef _OPENMP
+#include <omp.h>
+#endif
+
+int okc_state_root_par(const sre_account_entry *acct, uint64_t na,
+                       const sre_storage_entry *st, uint64_t ns,
+                       int nthreads, uint8_t out[32])
+{
+    if (na == 0) {
+        if (ns != 0)
+            return 3;
+        memcpy(out, EMPTY_ROOT, 32);
+        return 0;
+    }
+    for (uint64_t i = 1; i < na; i++)
+        if (memcmp(acct[i - 1].key, acct[i].key, 32) >= 0)
+            return 1;
+    for (uint64_t i = 1; i < ns; i++) {
+        int c = memcmp(st[i - 1].acct_key, st[i].acct_key, 32);
+        if (c > 0 ||
+            (c == 0 && memcmp(st[i - 1].slot_key, st[i].slot_key, 32) >= 0))
+            return 2;
+    }
+    for (uint64_t i = 0; i < ns; i++)
+        if (is_zero32(st[i].value))
+            return 4;
+    uint8_t *roots = (uint8_t *)malloc(na * 32);
+    uint64_t *seg_lo = (uint64_t *)malloc(na * sizeof(uint64_t));
+    uint64_t *seg_hi = (uint64_t *)malloc(na * sizeof(uint64_t));
+    if (!roots || !seg_lo || !seg_hi) {
+        free(roots);
+        free(seg_lo);
+        free(seg_hi);
+        return 5;
+    }
+    /* serial lockstep pass: per-account storage segment bounds */
+    uint64_t i = 0;
+    int rc = 0;
+    for (uint64_t j = 0; j < na; j++) {
+        if (i < ns && memcmp(st[i].acct_key, acct[j].key, 32) < 0) {
+            rc = 3;
+            break;
+        }
+        seg_lo[j] = i;
+        while (i < ns && memcmp(st[i].acct_key, acct[j].key, 32) == 0)
+            i++;
+        seg_hi[j] = i;
+    }
+    if (rc == 0 && i != ns)
+        rc = 3;
+    if (rc) {
+        free(roots);
+        free(seg_lo);
+        free(seg_hi);
+        return rc;
+    }
+#ifdef _OPENMP
+    if (nthreads > 0)
+        omp_set_num_threads(nthreads);
+#else
+    (void)nthreads;
+#endif
+#pragma omp parallel for schedule(dynamic, 64)
+    for (uint64_t j = 0; j < na; j++) {
+        if (seg_lo[j] == seg_hi[j]) {
+            memcpy(roots + 32 * j, EMPTY_ROOT, 32);
+        } else {
+            trie_src src;
+            src.upd = NULL;
+            src.key_stride = sizeof(sre_storage_entry);
+            src.value = storage_value;
+            src.keys = st[seg_lo[j]].slot_key;
+            src.vctx = st + seg_lo[j];
+            uint8_t rlp[NODE_MAX];
+            ref_t ref;
+            size_t len = build(&src, 0, seg_hi[j] - seg_lo[j], 0, rlp, &ref,
+                               NULL);
+            okc_keccak256(rlp, len, roots + 32 * j);
+        }
+    }
+    /* account trie: 16-way top-nibble split, then the shard finisher */
+    uint8_t child_refs[16][33], child_lens[16], root_hash[16][32];
+    uint64_t counts[16], starts[17];
+    memset(child_lens, 0, sizeof(child_lens));
+    memset(counts, 0, sizeof(counts));
+    {
+        uint64_t p = 0;
+        for (int b = 0; b < 16; b++) {
+            starts[b] = p;
+            while (p < na && nib(acct[p].key, 0) == (uint8_t)b)
+                p++;
+        }
+        starts[16] = na;
+    }
+#pragma omp parallel for schedule(dynamic, 1)
+    for (int b = 0; b < 16; b++) {
+        uint64_t lo = starts[b], hi = starts[b + 1];
+        if (lo == hi)
+            continue;
+        acct_vctx vc = { acct + lo, roots + 32 * lo };
+        trie_src src;
+        src.upd = NULL;
+        src.keys = acct[lo].key;
+        src.key_stride = sizeof(sre_account_entry);
+        src.value = account_value;
+        src.vctx = &vc;
+        uint8_t rlp[NODE_MAX];
+        ref_t ref;
+        build(&src, 0, hi - lo, 1, rlp, &ref, NULL);
+        memcpy(child_refs[b], ref.b, ref.len);
+        child_lens[b] = ref.len;
+        size_t len = build(&src, 0, hi - lo, 0, rlp, &ref, NULL);
+        okc_keccak256(rlp, len, root_hash[b]);
+        counts[b] = hi - lo;
+    }
+    free(seg_lo);
+    free(seg_hi);
+    free(roots);
+    return okc_finish_top(child_refs, child_lens, root_hash, counts, out);
+}

@@ -124,3 +124,15 @@ def test_random_cross_check(trial):
     assert bind.state_root(acct, st) == want
     refs, lens, roots, counts = bind.subtree_roots(acct, st)
     assert bind.finish_top(refs, lens, roots, counts) == want
+
+
+def test_state_root_par_matches_serial():
+    """okc_state_root_par (the per-core CPU-baseline leg) is the same
+    algorithm: identical roots on mixed shapes incl. single-nibble and
+    empty-storage edges."""
+    from reth_amd import gen
+    acct, st = gen.gen_state_numpy(5000, 4, bind.keccak256_batch)
+    assert bind.state_root_par(acct, st) == bind.state_root(acct, st)
+    acct2, st2 = gen.gen_state_numpy(300, 0, bind.keccak256_batch)
+    assert bind.state_root_par(acct2, st2) == bind.state_root(acct2, st2)
+    assert bind.state_root_par(acct2[:1], st2) == bind.state_root(acct2[:1], st2)
