@@ -172,23 +172,29 @@ int sre_apply_delta(sre_ctx *ctx,
                     const sre_account_delta *acct_delta, uint64_t n_acct,
                     const sre_storage_entry *st_delta, uint64_t n_st);
 
-/* Dirty-path incremental root (accounts-only states, v1).
+/* Dirty-path incremental root.
  *
- * sre_root_retaining computes the full root like sre_root AND captures the
- * engine's cell-top records (the unique trie nodes covering fixed 5-nibble
- * key prefixes). sre_incremental_root then applies an overlay delta
- * (semantics of sre_apply_delta) and recomputes ONLY the cells the delta
- * touches, seeding every clean cell's retained top record into the level
- * machinery — the walker-skip semantics of StateRoot::root with a prefix
- * set (crates/trie/trie/src/trie.rs:228-253 `walker.advance` over
- * `changed_prefixes`), expressed as cell-granular reuse. The resident
- * state is replaced by the merged result and retention is refreshed, so
- * deltas chain. Falls back internally to a full recompute (still correct)
- * if retained coverage is incomplete. */
+ * sre_root_retaining computes the full root like sre_root AND retains (a)
+ * the engine's cell-top records (the unique trie nodes covering fixed
+ * 5-nibble account-key prefix cells) and (b) every account's storage
+ * root. sre_incremental_root then applies an overlay delta (semantics of
+ * sre_apply_delta: account upserts/deletes + storage upserts/zero-deletes)
+ * and recomputes ONLY what the delta touches: the storage tries of
+ * accounts with storage-delta rows (other accounts' roots are carried
+ * across the merge; destroyed accounts drop theirs), and the account-trie
+ * cells containing any delta key — seeding every clean cell's retained
+ * top record into the level machinery. This is the walker-skip semantics
+ * of StateRoot::root with a prefix set (crates/trie/trie/src/trie.rs:228-253
+ * `walker.advance` over `changed_prefixes`), expressed as cell-granular
+ * reuse. The resident state is replaced by the merged result and retention
+ * is refreshed, so deltas chain. Uncovered positions fall back to leaf
+ * recomputation (still correct). */
 int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32]);
 int sre_incremental_root(sre_ctx *ctx,
                          const sre_account_delta *acct_delta,
-                         uint64_t n_delta, uint8_t out_root[32]);
+                         uint64_t n_acct,
+                         const sre_storage_entry *st_delta,
+                         uint64_t n_st, uint8_t out_root[32]);
 
 /* Account multiproof — the surface of Proof::account_proof /
  * Proof::multiproof restricted to account targets

@@ -1559,6 +1559,111 @@ __global__ void k_ovl_posmap(const sre_account_entry *__restrict__ base,
     map[i] = bexcl[i] + dexcl[p];
 }
 
+// carry surviving accounts' retained storage roots into the merged
+// positions (accounts replaced by a non-deleting delta row KEEP their
+// storage — hashed_state.rs:425-440 wipes only on destruction)
+__global__ void k_ovl_carry_roots(const sre_account_entry *__restrict__ base,
+                                  uint64_t nb,
+                                  const uint32_t *__restrict__ bexcl,
+                                  const sre_account_delta *__restrict__ dl,
+                                  uint64_t nd,
+                                  const uint32_t *__restrict__ dexcl,
+                                  const uint8_t *__restrict__ old_roots,
+                                  uint8_t *__restrict__ new_roots)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= nb)
+        return;
+    uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_account_delta), nd,
+                         base[i].key, 32);
+    if (p < nd && cmp_key32(dl[p].key, base[i].key) == 0 && dl[p].deleted)
+        return; // destroyed: no root to carry
+    uint64_t j = bexcl[i] + dexcl[p]; // new index (posmap formula)
+    const uint64_t *s = (const uint64_t *)(old_roots + 32 * i);
+    uint64_t *d = (uint64_t *)(new_roots + 32 * j);
+#pragma unroll
+    for (int k = 0; k < 4; ++k)
+        d[k] = s[k];
+}
+
+
+// per touched account: segment bounds in the merged storage array, the
+// account's index in the merged account array, and an EMPTY_ROOT reset
+// (recomputed segments are scattered over it afterwards)
+__global__ void k_touched_bounds(const sre_storage_entry *__restrict__ st,
+                                 uint64_t ns,
+                                 const sre_account_entry *__restrict__ acct,
+                                 uint64_t na,
+                                 const uint8_t *__restrict__ tkeys, uint32_t nt,
+                                 uint32_t *__restrict__ lo,
+                                 uint32_t *__restrict__ hi,
+                                 uint32_t *__restrict__ aidx,
+                                 uint8_t *__restrict__ roots,
+                                 uint32_t *__restrict__ err)
+{
+    uint32_t t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= nt)
+        return;
+    const uint8_t *key = tkeys + 32ull * t;
+    // storage range [lo, hi) of this account
+    uint64_t l = 0, h = ns;
+    while (l < h) {
+        uint64_t m = (l + h) / 2;
+        if (cmp_key32(st[m].acct_key, key) < 0)
+            l = m + 1;
+        else
+            h = m;
+    }
+    lo[t] = (uint32_t)l;
+    h = ns;
+    uint64_t l2 = l;
+    while (l2 < h) {
+        uint64_t m = (l2 + h) / 2;
+        if (cmp_key32(st[m].acct_key, key) <= 0)
+            l2 = m + 1;
+        else
+            h = m;
+    }
+    hi[t] = (uint32_t)l2;
+    uint64_t p = lb_keys((const uint8_t *)acct, sizeof(sre_account_entry), na,
+                         key, 32);
+    if (p >= na || cmp_key32(acct[p].key, key) != 0) {
+        if (lo[t] != hi[t])
+            atomicOr(err, 1u << E_INTERNAL); // slots for an absent account
+        aidx[t] = 0xFFFFFFFFu;
+        return;
+    }
+    aidx[t] = (uint32_t)p;
+    for (int k = 0; k < 32; ++k)
+        roots[32ull * p + k] = D_EMPTY_ROOT[k];
+}
+
+// gather the touched accounts' storage entries into a compact array
+__global__ void k_gather_touched(const sre_storage_entry *__restrict__ st,
+                                 const uint32_t *__restrict__ lo,
+                                 const uint32_t *__restrict__ hi,
+                                 const uint32_t *__restrict__ offs, uint32_t nt,
+                                 uint64_t total,
+                                 sre_storage_entry *__restrict__ out)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= total)
+        return;
+    // locate the touched account owning compact position j
+    uint32_t l = 0, h = nt;
+    while (l < h) {
+        uint32_t m = (l + h) / 2;
+        if ((uint64_t)offs[m + 1] <= j)
+            l = m + 1;
+        else
+            h = m;
+    }
+    const uint8_t *s = (const uint8_t *)&st[lo[l] + (j - offs[l])];
+    uint8_t *d = (uint8_t *)&out[j];
+    for (int k = 0; k < (int)sizeof(sre_storage_entry); k += 8)
+        *(uint64_t *)(d + k) = *(const uint64_t *)(s + k);
+}
+
 // base account survives iff its key is absent from the delta
 __global__ void k_ovl_base_acct_flags(const sre_account_entry *__restrict__ base,
                                       uint64_t nb,
@@ -1807,6 +1912,17 @@ __global__ void k_revalidate_rows(const cap_row *__restrict__ rows, uint64_t n,
         atomicAdd(&hist[threadIdx.x], hist_l[threadIdx.x]);
 }
 
+// dirty-cell marking from STORAGE delta rows (the account's leaf changes)
+__global__ void k_mark_delta_cells_st(const sre_storage_entry *__restrict__ dl,
+                                      uint64_t nd,
+                                      uint8_t *__restrict__ cell_dirty)
+{
+    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (j >= nd)
+        return;
+    cell_dirty[cell_of_key(dl[j].acct_key)] = 1;
+}
+
 // mark cells touched by delta keys
 __global__ void k_mark_delta_cells(const sre_account_delta *__restrict__ dl,
                                    uint64_t nd, uint8_t *__restrict__ cell_dirty)
@@ -1975,6 +2091,10 @@ struct sre_ctx {
     bool cells_valid = false;    // retained rows match the resident state
     void *d_cap_rows = nullptr;  // cap_row[cap_count]
     uint64_t cap_count = 0, cap_capacity = 0;
+    void *d_roots_ret = nullptr; // retained per-account storage roots (na x 32)
+    uint64_t roots_ret_capacity = 0; // in accounts
+    void *d_roots_ret2 = nullptr; // ping-pong partner (incremental updates)
+    uint64_t roots_ret2_capacity = 0;
     // size-class buffer pool: the level machinery allocates/frees dozens of
     // transient arrays per level; hipMalloc latency would dominate small
     // jobs. Freed buffers are cached by power-of-2 class and reused (also
@@ -2084,6 +2204,10 @@ extern "C" void sre_destroy(sre_ctx *ctx)
     release_st(ctx);
     if (ctx->d_cap_rows)
         (void)hipFree(ctx->d_cap_rows);
+    if (ctx->d_roots_ret)
+        (void)hipFree(ctx->d_roots_ret);
+    if (ctx->d_roots_ret2)
+        (void)hipFree(ctx->d_roots_ret2);
     for (auto &e : ctx->pool)
         (void)hipFree(e.second);
     ctx->pool.clear();
@@ -2636,12 +2760,20 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
                             proof_row *d_prows = nullptr,
                             uint32_t *d_prow_cnt = nullptr,
                             uint32_t prow_cap = 0,
-                            const uint32_t *d_pti2 = nullptr)
+                            const uint32_t *d_pti2 = nullptr,
+                            // subset mode (incremental): run over an
+                            // explicit entry array and scatter roots into
+                            // d_acct_roots WITHOUT resetting it first
+                            const sre_storage_entry *d_st_in = nullptr,
+                            uint64_t ns_in = ~0ull)
 {
-    uint64_t ns = ctx->ns, na = ctx->na;
-    hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(na)), dim3(BLOCK), 0,
-                       ctx->stream, d_acct_roots, na);
-    HIP_CHECK(ctx, hipGetLastError());
+    uint64_t ns = d_st_in ? ns_in : ctx->ns, na = ctx->na;
+    const sre_storage_entry *d_st = d_st_in ? d_st_in : d_st;
+    if (!d_st_in) {
+        hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(na)), dim3(BLOCK),
+                           0, ctx->stream, d_acct_roots, na);
+        HIP_CHECK(ctx, hipGetLastError());
+    }
     if (ns == 0)
         return 0;
 
@@ -2655,7 +2787,7 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
     HIP_CHECK(ctx, hipMemsetAsync(hist.p, 0, 66 * 4, ctx->stream));
 
     hipLaunchKernelGGL(k_seg_flags_lcp, dim3(grid_for(ns + 1)), dim3(BLOCK), 0,
-                       ctx->stream, ctx->d_st, ns, flags.as<uint32_t>(),
+                       ctx->stream, d_st, ns, flags.as<uint32_t>(),
                        lcp.as<int8_t>(), d_err);
     HIP_CHECK(ctx, hipGetLastError());
     // seg_id[i] = inclusive_scan(flags)[i] - 1 (segment index of entry i)
@@ -2674,7 +2806,7 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
                        flags.as<uint32_t>(), seg_id.as<uint32_t>(), ns,
                        seg_start.as<uint32_t>());
     hipLaunchKernelGGL(k_seg_acct, dim3(grid_for(n_seg)), dim3(BLOCK), 0, ctx->stream,
-                       ctx->d_st, seg_start.as<uint32_t>(), n_seg, ctx->d_acct, na,
+                       d_st, seg_start.as<uint32_t>(), n_seg, ctx->d_acct, na,
                        seg_acct.as<uint32_t>(), d_err);
     // input-contract violations (unsorted/orphan entries) are flagged by the
     // kernels above; bail BEFORE the trie machinery runs on garbage lcps.
@@ -2687,7 +2819,7 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
     hipEventCreate(&ev1);
     hipEventRecord(ev0, ctx->stream);
     hipLaunchKernelGGL(k_leaf_storage, dim3(grid_for(ns)), dim3(BLOCK), 0, ctx->stream,
-                       ctx->d_st, ns, lcp.as<int8_t>(), seg_id.as<uint32_t>(),
+                       d_st, ns, lcp.as<int8_t>(), seg_id.as<uint32_t>(),
                        recs.as<node_rec>(), depths.as<uint8_t>(), hist.as<uint32_t>(),
                        seg_roots.as<uint8_t>(), d_err);
     HIP_CHECK(ctx, hipGetLastError());
@@ -2709,7 +2841,7 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
         HIP_CHECK(ctx, bhash.alloc(ns * 32));
     size_t upd_start = ctx->updates.size();
     const uint8_t *keys =
-        (const uint8_t *)ctx->d_st + offsetof(sre_storage_entry, slot_key);
+        (const uint8_t *)d_st + offsetof(sre_storage_entry, slot_key);
     if (run_levels(ctx, ns, recs.as<node_rec>(), depths.as<uint8_t>(),
                    lcp.as<int8_t>(), keys, sizeof(sre_storage_entry), hist_host, 0,
                    seg_roots.as<uint8_t>(), nullptr, nullptr, d_err, po,
@@ -3765,7 +3897,11 @@ static bool row_less(const sre_update_row &a, const sre_update_row &b)
 static int apply_delta_impl(sre_ctx *ctx, const sre_account_delta *acct_delta,
                             uint64_t n_acct, const sre_storage_entry *st_delta,
                             uint64_t n_st, DBuf *map_out /* optional: old->new
-                            positions, (old na)+1 u32 */)
+                            positions, (old na)+1 u32 */,
+                            const uint8_t *d_old_roots = nullptr,
+                            uint8_t *d_new_roots = nullptr /* carry retained
+                            storage roots across the merge (new-na x 32,
+                            pre-filled EMPTY by the caller) */)
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     uint64_t nb = ctx->na, ns = ctx->ns;
@@ -3829,6 +3965,13 @@ static int apply_delta_impl(sre_ctx *ctx, const sre_account_delta *acct_delta,
                            ctx->stream, ctx->d_acct, nb, ea.as<uint32_t>(),
                            dl_a.as<sre_account_delta>(), n_acct,
                            ed.as<uint32_t>(), map_out->as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+    }
+    if (d_old_roots && d_new_roots && nb) {
+        hipLaunchKernelGGL(k_ovl_carry_roots, dim3(grid_for(nb)), dim3(BLOCK),
+                           0, ctx->stream, ctx->d_acct, nb, ea.as<uint32_t>(),
+                           dl_a.as<sre_account_delta>(), n_acct,
+                           ed.as<uint32_t>(), d_old_roots, d_new_roots);
         HIP_CHECK(ctx, hipGetLastError());
     }
     HIP_CHECK(ctx, sdel.alloc((Ndel ? Ndel : 1) * 32));
@@ -3913,13 +4056,22 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
 
 // Full root over an accounts-only state, retaining the cell-top records
 // for subsequent sre_incremental_root calls (dirty-path incremental).
+static int ensure_roots_ret(sre_ctx *ctx, uint64_t want_accounts)
+{
+    if (ctx->roots_ret_capacity >= want_accounts)
+        return 0;
+    if (ctx->d_roots_ret)
+        (void)hipFree(ctx->d_roots_ret);
+    ctx->d_roots_ret = nullptr;
+    ctx->roots_ret_capacity = 0;
+    HIP_CHECK(ctx, hipMalloc(&ctx->d_roots_ret, want_accounts * 32));
+    ctx->roots_ret_capacity = want_accounts;
+    return 0;
+}
+
 extern "C" int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32])
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
-    if (ctx->ns != 0) {
-        set_err(ctx, "sre_root_retaining: accounts-only states (v1)");
-        return -1;
-    }
     memset(&ctx->stats, 0, sizeof(ctx->stats));
     ctx->cells_valid = false;
     uint64_t want_cap = 2 * (ctx->na < (uint64_t)N_CELLS ? ctx->na
@@ -3933,11 +4085,17 @@ extern "C" int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32])
         ctx->cap_capacity = want_cap;
     }
     if (ctx->na == 0) {
+        if (ctx->ns != 0) {
+            set_err(ctx, "storage entries without accounts");
+            return -1;
+        }
         ctx->cap_count = 0;
         ctx->cells_valid = true;
         memcpy(out_root, EMPTY_ROOT_H, 32);
         return 0;
     }
+    if (ensure_roots_ret(ctx, ctx->na))
+        return -1;
     DBuf err(ctx), roots(ctx), capcnt(ctx);
     HIP_CHECK(ctx, err.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
@@ -3945,7 +4103,13 @@ extern "C" int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32])
     HIP_CHECK(ctx, capcnt.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(capcnt.p, 0, 4, ctx->stream));
     pass_out po;
-    if (run_account_pass(ctx, nullptr, 0, roots.as<uint8_t>(), nullptr, nullptr,
+    // full storage pass into the RETAINED roots buffer (chained deltas
+    // carry and patch it instead of recomputing every segment)
+    if (run_storage_pass(ctx, (uint8_t *)ctx->d_roots_ret, &po,
+                         err.as<uint32_t>()))
+        return -1;
+    if (run_account_pass(ctx, (const uint8_t *)ctx->d_roots_ret, 0,
+                         roots.as<uint8_t>(), nullptr, nullptr,
                          &po, err.as<uint32_t>(), CELL_NIBBLES,
                          (cap_row *)ctx->d_cap_rows, capcnt.as<uint32_t>(),
                          ctx->cap_capacity))
@@ -3960,18 +4124,20 @@ extern "C" int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32])
     return 0;
 }
 
-// Dirty-path incremental root: apply an accounts-only overlay delta and
-// recompute only the 5-nibble cells it touches, reusing every clean
-// cell-top record (the §3b walker-skip semantics expressed in this
+// Dirty-path incremental root: apply a HashedPostState overlay delta
+// (accounts + storage) and recompute only the 5-nibble cells it touches —
+// reusing every clean cell-top record and every untouched account's
+// retained storage root (the §3b walker-skip semantics expressed in this
 // engine's level machinery). Requires a prior sre_root_retaining.
 extern "C" int sre_incremental_root(sre_ctx *ctx,
                                     const sre_account_delta *acct_delta,
-                                    uint64_t n_delta, uint8_t out_root[32])
+                                    uint64_t n_acct,
+                                    const sre_storage_entry *st_delta,
+                                    uint64_t n_st, uint8_t out_root[32])
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
-    if (!ctx->cells_valid || ctx->ns != 0) {
-        set_err(ctx, "sre_incremental_root: needs sre_root_retaining on an "
-                     "accounts-only state first");
+    if (!ctx->cells_valid) {
+        set_err(ctx, "sre_incremental_root: needs sre_root_retaining first");
         return -1;
     }
     memset(&ctx->stats, 0, sizeof(ctx->stats));
@@ -3980,8 +4146,29 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     hipEventCreate(&t1);
     hipEventRecord(t0, ctx->stream);
 
+    // new roots buffer: the ping-pong partner of the retained one (no
+    // per-step hipMalloc at steady state), EMPTY-filled upper bound,
+    // carried across the merge
+    uint64_t max_na = ctx->na + n_acct;
+    if (ctx->roots_ret2_capacity < max_na) {
+        if (ctx->d_roots_ret2)
+            (void)hipFree(ctx->d_roots_ret2);
+        ctx->d_roots_ret2 = nullptr;
+        ctx->roots_ret2_capacity = 0;
+        HIP_CHECK(ctx, hipMalloc(&ctx->d_roots_ret2,
+                                 (max_na ? max_na : 1) * 32));
+        ctx->roots_ret2_capacity = max_na ? max_na : 1;
+    }
+    void *new_roots = ctx->d_roots_ret2;
+    hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(max_na ? max_na : 1)),
+                       dim3(BLOCK), 0, ctx->stream, (uint8_t *)new_roots,
+                       max_na ? max_na : 1);
+    HIP_CHECK(ctx, hipGetLastError());
+
     DBuf map(ctx);
-    if (apply_delta_impl(ctx, acct_delta, n_delta, nullptr, 0, &map))
+    if (apply_delta_impl(ctx, acct_delta, n_acct, st_delta, n_st, &map,
+                         (const uint8_t *)ctx->d_roots_ret,
+                         (uint8_t *)new_roots))
         return -1;
     uint64_t na = ctx->na;
     if (na == 0) {
@@ -3989,22 +4176,91 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
         memcpy(out_root, EMPTY_ROOT_H, 32);
         return 0;
     }
-    DBuf err(ctx), bitmap(ctx), dl(ctx), lcp(ctx), recs(ctx), depths(ctx),
-        hist(ctx), roots(ctx), capcnt(ctx);
+
+    DBuf err(ctx), bitmap(ctx), dl(ctx), dls(ctx), lcp(ctx), recs(ctx),
+        depths(ctx), hist(ctx), roots(ctx), capcnt(ctx);
     HIP_CHECK(ctx, err.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
     HIP_CHECK(ctx, bitmap.alloc(N_CELLS));
     HIP_CHECK(ctx, hipMemsetAsync(bitmap.p, 0, N_CELLS, ctx->stream));
-    HIP_CHECK(ctx, dl.alloc((n_delta ? n_delta : 1) * sizeof(sre_account_delta)));
-    if (n_delta) {
+    HIP_CHECK(ctx, dl.alloc((n_acct ? n_acct : 1) * sizeof(sre_account_delta)));
+    if (n_acct) {
         HIP_CHECK(ctx, hipMemcpyAsync(dl.p, acct_delta,
-                                      n_delta * sizeof(sre_account_delta),
+                                      n_acct * sizeof(sre_account_delta),
                                       hipMemcpyHostToDevice, ctx->stream));
-        hipLaunchKernelGGL(k_mark_delta_cells, dim3(grid_for(n_delta)),
+        hipLaunchKernelGGL(k_mark_delta_cells, dim3(grid_for(n_acct)),
                            dim3(BLOCK), 0, ctx->stream,
-                           dl.as<sre_account_delta>(), n_delta,
+                           dl.as<sre_account_delta>(), n_acct,
                            bitmap.as<uint8_t>());
         HIP_CHECK(ctx, hipGetLastError());
+    }
+    if (n_st) {
+        HIP_CHECK(ctx, dls.alloc(n_st * sizeof(sre_storage_entry)));
+        HIP_CHECK(ctx, hipMemcpyAsync(dls.p, st_delta,
+                                      n_st * sizeof(sre_storage_entry),
+                                      hipMemcpyHostToDevice, ctx->stream));
+        hipLaunchKernelGGL(k_mark_delta_cells_st, dim3(grid_for(n_st)),
+                           dim3(BLOCK), 0, ctx->stream,
+                           dls.as<sre_storage_entry>(), n_st,
+                           bitmap.as<uint8_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+    }
+    pass_out po;
+    // recompute ONLY the touched accounts' storage tries: gather their
+    // merged segments into a compact array and run the storage machinery
+    // over it, scattering fresh roots over the carried ones
+    if (n_st) {
+        std::vector<uint8_t> tkeys;
+        tkeys.reserve(32 * n_st);
+        for (uint64_t i = 0; i < n_st; ++i)
+            if (i == 0 || memcmp(st_delta[i].acct_key,
+                                 st_delta[i - 1].acct_key, 32) != 0)
+                tkeys.insert(tkeys.end(), st_delta[i].acct_key,
+                             st_delta[i].acct_key + 32);
+        uint32_t nt = (uint32_t)(tkeys.size() / 32);
+        DBuf dtk(ctx), tlo(ctx), thi(ctx), taidx(ctx);
+        HIP_CHECK(ctx, dtk.alloc(tkeys.size()));
+        HIP_CHECK(ctx, hipMemcpyAsync(dtk.p, tkeys.data(), tkeys.size(),
+                                      hipMemcpyHostToDevice, ctx->stream));
+        HIP_CHECK(ctx, tlo.alloc((uint64_t)nt * 4));
+        HIP_CHECK(ctx, thi.alloc((uint64_t)nt * 4));
+        HIP_CHECK(ctx, taidx.alloc((uint64_t)nt * 4));
+        hipLaunchKernelGGL(k_touched_bounds, dim3(grid_for(nt)), dim3(BLOCK),
+                           0, ctx->stream, ctx->d_st, ctx->ns, ctx->d_acct, na,
+                           dtk.as<uint8_t>(), nt, tlo.as<uint32_t>(),
+                           thi.as<uint32_t>(), taidx.as<uint32_t>(),
+                           (uint8_t *)new_roots, err.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        std::vector<uint32_t> lo(nt), hi(nt), offs(nt + 1);
+        HIP_CHECK(ctx, hipMemcpy(lo.data(), tlo.p, 4ull * nt,
+                                 hipMemcpyDeviceToHost));
+        HIP_CHECK(ctx, hipMemcpy(hi.data(), thi.p, 4ull * nt,
+                                 hipMemcpyDeviceToHost));
+        offs[0] = 0;
+        for (uint32_t t = 0; t < nt; ++t)
+            offs[t + 1] = offs[t] + (hi[t] - lo[t]);
+        uint64_t total = offs[nt];
+        if (total) {
+            DBuf doffs(ctx), compact(ctx);
+            HIP_CHECK(ctx, doffs.alloc(4ull * (nt + 1)));
+            HIP_CHECK(ctx, hipMemcpyAsync(doffs.p, offs.data(),
+                                          4ull * (nt + 1),
+                                          hipMemcpyHostToDevice, ctx->stream));
+            HIP_CHECK(ctx, compact.alloc(total * sizeof(sre_storage_entry)));
+            hipLaunchKernelGGL(k_gather_touched, dim3(grid_for(total)),
+                               dim3(BLOCK), 0, ctx->stream,
+                               ctx->d_st, tlo.as<uint32_t>(),
+                               thi.as<uint32_t>(), doffs.as<uint32_t>(), nt,
+                               total, compact.as<sre_storage_entry>());
+            HIP_CHECK(ctx, hipGetLastError());
+            if (run_storage_pass(ctx, (uint8_t *)new_roots, &po,
+                                 err.as<uint32_t>(), nullptr, 0, nullptr,
+                                 nullptr, 0, nullptr,
+                                 compact.as<sre_storage_entry>(), total))
+                return -1;
+        }
+        if (check_err(ctx, err.as<uint32_t>()))
+            return -1;
     }
     HIP_CHECK(ctx, lcp.alloc(na + 1));
     hipLaunchKernelGGL(k_lcp_account, dim3(grid_for(na + 1)), dim3(BLOCK), 0,
@@ -4030,14 +4286,15 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
                            depths.as<uint8_t>(), covered.as<uint8_t>(),
                            hist.as<uint32_t>());
     HIP_CHECK(ctx, hipGetLastError());
-    // rehash leaves of dirty cells and of positions no seed covers
-    pass_out po;
+    // rehash leaves of dirty cells and of positions no seed covers, with
+    // the carried+patched storage roots
     hipEvent_t ev0, ev1;
     hipEventCreate(&ev0);
     hipEventCreate(&ev1);
     hipEventRecord(ev0, ctx->stream);
     hipLaunchKernelGGL(k_leaf_account, dim3(grid_for(na)), dim3(BLOCK), 0,
-                       ctx->stream, ctx->d_acct, na, nullptr, lcp.as<int8_t>(),
+                       ctx->stream, ctx->d_acct, na,
+                       (const uint8_t *)new_roots, lcp.as<int8_t>(),
                        0, recs.as<node_rec>(), depths.as<uint8_t>(),
                        hist.as<uint32_t>(), roots.as<uint8_t>(), nullptr,
                        nullptr, bitmap.as<uint8_t>(), covered.as<uint8_t>());
@@ -4081,15 +4338,18 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     uint32_t cnt = 0;
     HIP_CHECK(ctx, hipMemcpy(&cnt, capcnt.p, 4, hipMemcpyDeviceToHost));
     ctx->cap_count = cnt;
+    // swap the ping-pong retained roots (chained deltas)
+    std::swap(ctx->d_roots_ret, ctx->d_roots_ret2);
+    std::swap(ctx->roots_ret_capacity, ctx->roots_ret2_capacity);
     ctx->cells_valid = true;
 
     hipEventRecord(t1, ctx->stream);
     HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
-    float total = 0;
-    hipEventElapsedTime(&total, t0, t1);
+    float total_ms = 0;
+    hipEventElapsedTime(&total_ms, t0, t1);
     hipEventDestroy(t0);
     hipEventDestroy(t1);
-    ctx->stats.total_ms = total;
+    ctx->stats.total_ms = total_ms;
     ctx->stats.leaf_hash_ms = po.leaf_ms;
     ctx->stats.branch_hash_ms = po.branch_ms;
     ctx->stats.branch_count = po.branch_count;

@@ -174,24 +174,32 @@ class StateRootEngine:
 
     # ---- compute ----
     def root_retaining(self) -> bytes:
-        """Full root that also retains cell-top trie records so following
-        incremental_root() calls recompute only dirty 5-nibble cells
-        (accounts-only states). See include/sre.h sre_root_retaining."""
+        """Full root that also retains cell-top trie records and every
+        account's storage root, so following incremental_root() calls
+        recompute only what a delta touches. See include/sre.h
+        sre_root_retaining."""
         out = (ctypes.c_uint8 * 32)()
         self._check(self._lib.sre_root_retaining(
             ctypes.c_void_p(self._ctx), out))
         return bytes(out)
 
-    def incremental_root(self, acct_delta: np.ndarray) -> bytes:
-        """Apply an accounts-only overlay delta and recompute the root along
-        dirty paths only (requires a prior root_retaining). The resident
-        state becomes the merged result and retention is refreshed, so
-        deltas chain."""
+    def incremental_root(self, acct_delta: np.ndarray,
+                         st_delta: np.ndarray = None) -> bytes:
+        """Apply a HashedPostState overlay delta (accounts + storage) and
+        recompute the root along dirty paths only (requires a prior
+        root_retaining): untouched accounts keep their retained storage
+        roots, touched storage tries are rebuilt from their merged
+        segments, and only delta-touched account-trie cells rehash. The
+        resident state becomes the merged result and retention is
+        refreshed, so deltas chain."""
         assert acct_delta.dtype == DELTA_DTYPE
+        if st_delta is None:
+            st_delta = np.zeros(0, dtype=STORAGE_DTYPE)
+        assert st_delta.dtype == STORAGE_DTYPE
         out = (ctypes.c_uint8 * 32)()
         self._check(self._lib.sre_incremental_root(
             ctypes.c_void_p(self._ctx), _np_ptr(acct_delta),
-            len(acct_delta), out))
+            len(acct_delta), _np_ptr(st_delta), len(st_delta), out))
         return bytes(out)
 
     def root(self) -> bytes:

@@ -273,11 +273,70 @@ def test_incremental_requires_retention(eng):
     # no root_retaining yet -> error (upload invalidates retention)
     with pytest.raises(RuntimeError):
         eng.incremental_root(np.zeros(0, DELTA_DTYPE))
-    # storage present -> root_retaining refuses (v1 accounts-only)
-    acct2, st2 = gen.gen_state_numpy(100, 2, bind.keccak256_batch)
-    eng.upload(acct2, st2)
-    with pytest.raises(RuntimeError):
-        eng.root_retaining()
+
+
+def test_incremental_with_storage_chained(eng):
+    rng = np.random.default_rng(99)
+    acct, st = gen.gen_state_numpy(2000, 4, bind.keccak256_batch)
+    accounts = _dict_of(acct, st)
+    eng.upload(acct, st)
+    assert eng.root_retaining() == bind.state_root(*_arrays_of(accounts))
+    ke = bind.keccak256(b"")
+    for step in range(4):
+        keys = sorted(accounts)
+        rows, strows = [], []
+        # slot upserts + deletions on random surviving accounts
+        for i in rng.choice(len(keys), 12, replace=False):
+            k = keys[int(i)]
+            slots = accounts[k][3]
+            if slots and rng.random() < 0.4:
+                dead = sorted(slots)[0]
+                strows.append((k, dead, 0))
+                del slots[dead]
+            nk = bind.keccak256(b"slot" + bytes([step]) + k[:4])
+            strows.append((k, nk, 1000 + step))
+            slots[nk] = 1000 + step
+        # delete one account WITH storage (wipe), insert one with storage
+        victim = keys[int(rng.integers(len(keys)))]
+        if not any(r[0] == victim for r in strows):
+            rows.append((victim, 0, 0, ke, 1))
+            del accounts[victim]
+        nk = bind.keccak256(b"newacct" + bytes([step]))
+        ns1 = bind.keccak256(b"ns" + bytes([step]))
+        rows.append((nk, 1, 5, ke, 0))
+        strows.append((nk, ns1, 77))
+        accounts[nk] = [1, 5, ke, {ns1: 77}]
+        # plain balance modification (storage must be KEPT)
+        k2 = sorted(accounts)[3]
+        if all(r[0] != k2 for r in rows):
+            accounts[k2][1] += 9
+            rows.append((k2, accounts[k2][0], accounts[k2][1],
+                         accounts[k2][2], 0))
+        rows = sorted(set(rows))
+        strows = sorted(set(strows))
+        d, s = _mk_delta(rows, strows)
+        want = bind.state_root(*_arrays_of(accounts))
+        assert eng.incremental_root(d, s) == want, f"step {step}"
+
+
+def test_incremental_storage_wipe_all_slots(eng):
+    ke = bind.keccak256(b"")
+    ak = bind.keccak256(b"wipeme")
+    slots = {bind.keccak256(b"s" + bytes([i])): i + 1 for i in range(5)}
+    accounts = {ak: [3, 9, ke, dict(slots)],
+                bind.keccak256(b"other"): [1, 1, ke, {}]}
+    eng.upload(*_arrays_of(accounts))
+    assert eng.root_retaining() == bind.state_root(*_arrays_of(accounts))
+    # delete every slot -> storage root must become EMPTY_ROOT
+    strows = [(ak, sk, 0) for sk in slots]
+    accounts[ak][3].clear()
+    d, s = _mk_delta([], strows)
+    assert eng.incremental_root(d, s) == bind.state_root(*_arrays_of(accounts))
+    # then re-add one
+    sk = bind.keccak256(b"back")
+    accounts[ak][3][sk] = 42
+    d, s = _mk_delta([], [(ak, sk, 42)])
+    assert eng.incremental_root(d, s) == bind.state_root(*_arrays_of(accounts))
 
 
 def test_incremental_matches_apply_delta_path(eng):
