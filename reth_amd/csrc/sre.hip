@@ -2768,7 +2768,7 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
                             uint64_t ns_in = ~0ull)
 {
     uint64_t ns = d_st_in ? ns_in : ctx->ns, na = ctx->na;
-    const sre_storage_entry *d_st = d_st_in ? d_st_in : d_st;
+    const sre_storage_entry *d_st = d_st_in ? d_st_in : ctx->d_st;
     if (!d_st_in) {
         hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(na)), dim3(BLOCK),
                            0, ctx->stream, d_acct_roots, na);
