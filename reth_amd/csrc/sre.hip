@@ -4146,11 +4146,15 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     hipEventCreate(&t1);
     hipEventRecord(t0, ctx->stream);
 
+    // accounts-only states with accounts-only deltas skip the roots
+    // machinery entirely (every storage root is EMPTY_ROOT; the leaf
+    // kernel's null-roots path substitutes it) — the configs[4] shape
+    bool no_storage = (ctx->ns == 0 && n_st == 0);
     // new roots buffer: the ping-pong partner of the retained one (no
     // per-step hipMalloc at steady state), EMPTY-filled upper bound,
     // carried across the merge
     uint64_t max_na = ctx->na + n_acct;
-    if (ctx->roots_ret2_capacity < max_na) {
+    if (!no_storage && ctx->roots_ret2_capacity < max_na) {
         if (ctx->d_roots_ret2)
             (void)hipFree(ctx->d_roots_ret2);
         ctx->d_roots_ret2 = nullptr;
@@ -4159,15 +4163,24 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
                                  (max_na ? max_na : 1) * 32));
         ctx->roots_ret2_capacity = max_na ? max_na : 1;
     }
-    void *new_roots = ctx->d_roots_ret2;
-    hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(max_na ? max_na : 1)),
-                       dim3(BLOCK), 0, ctx->stream, (uint8_t *)new_roots,
-                       max_na ? max_na : 1);
-    HIP_CHECK(ctx, hipGetLastError());
+    void *new_roots = no_storage ? nullptr : ctx->d_roots_ret2;
+    if (new_roots) {
+        hipLaunchKernelGGL(k_fill_empty_roots,
+                           dim3(grid_for(max_na ? max_na : 1)), dim3(BLOCK),
+                           0, ctx->stream, (uint8_t *)new_roots,
+                           max_na ? max_na : 1);
+        HIP_CHECK(ctx, hipGetLastError());
+    }
 
     DBuf map(ctx);
+    // carry only when the pre-delta state HAS storage: while ns stays 0
+    // every retained root is EMPTY_ROOT (and accounts-only fast-path
+    // deltas may have drifted d_roots_ret's indexing — harmless, since it
+    // is then never read)
     if (apply_delta_impl(ctx, acct_delta, n_acct, st_delta, n_st, &map,
-                         (const uint8_t *)ctx->d_roots_ret,
+                         (new_roots && ctx->ns > 0)
+                             ? (const uint8_t *)ctx->d_roots_ret
+                             : nullptr,
                          (uint8_t *)new_roots))
         return -1;
     uint64_t na = ctx->na;
@@ -4294,7 +4307,8 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     hipEventRecord(ev0, ctx->stream);
     hipLaunchKernelGGL(k_leaf_account, dim3(grid_for(na)), dim3(BLOCK), 0,
                        ctx->stream, ctx->d_acct, na,
-                       (const uint8_t *)new_roots, lcp.as<int8_t>(),
+                       (const uint8_t *)new_roots /* null => EMPTY_ROOT */,
+                       lcp.as<int8_t>(),
                        0, recs.as<node_rec>(), depths.as<uint8_t>(),
                        hist.as<uint32_t>(), roots.as<uint8_t>(), nullptr,
                        nullptr, bitmap.as<uint8_t>(), covered.as<uint8_t>());
@@ -4339,8 +4353,10 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     HIP_CHECK(ctx, hipMemcpy(&cnt, capcnt.p, 4, hipMemcpyDeviceToHost));
     ctx->cap_count = cnt;
     // swap the ping-pong retained roots (chained deltas)
-    std::swap(ctx->d_roots_ret, ctx->d_roots_ret2);
-    std::swap(ctx->roots_ret_capacity, ctx->roots_ret2_capacity);
+    if (!no_storage) {
+        std::swap(ctx->d_roots_ret, ctx->d_roots_ret2);
+        std::swap(ctx->roots_ret_capacity, ctx->roots_ret2_capacity);
+    }
     ctx->cells_valid = true;
 
     hipEventRecord(t1, ctx->stream);

@@ -360,3 +360,43 @@ def test_incremental_matches_apply_delta_path(eng):
     eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
     eng.root_retaining()
     assert eng.incremental_root(d) == want
+
+
+def test_incremental_storage_appears_after_drift(eng):
+    # accounts-only base, several accounts-only deltas (fast path), THEN a
+    # delta that introduces storage: retained roots indexing has drifted
+    # but must not matter (all EMPTY while ns == 0)
+    ke = bind.keccak256(b"")
+    acct, _ = gen.gen_state_numpy(500, 0, bind.keccak256_batch)
+    accounts = {bytes(a["key"]): [int(a["nonce"]),
+                int.from_bytes(bytes(a["balance"]), "big"),
+                bytes(a["code_hash"]), {}] for a in acct}
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    assert eng.root_retaining() == bind.state_root(
+        *_arrays_of({k: tuple(v) for k, v in accounts.items()}))
+    for step in range(3):  # drift: inserts + deletes shift positions
+        keys = sorted(accounts)
+        rows = [(keys[step * 7], 0, 0, ke, 1)]
+        del accounts[keys[step * 7]]
+        nk = bind.keccak256(b"drift" + bytes([step]))
+        rows.append((nk, 1, step, ke, 0))
+        accounts[nk] = [1, step, ke, {}]
+        d, _ = _mk_delta(sorted(rows), [])
+        want = bind.state_root(
+            *_arrays_of({k: tuple(v) for k, v in accounts.items()}))
+        assert eng.incremental_root(d) == want
+    # now storage appears
+    tgt = sorted(accounts)[10]
+    sk = bind.keccak256(b"first-slot")
+    accounts[tgt][3][sk] = 123
+    d, s = _mk_delta([], [(tgt, sk, 123)])
+    want = bind.state_root(
+        *_arrays_of({k: tuple(v) for k, v in accounts.items()}))
+    assert eng.incremental_root(d, s) == want
+    # and chains further with more storage
+    sk2 = bind.keccak256(b"second-slot")
+    accounts[tgt][3][sk2] = 9
+    d, s = _mk_delta([], [(tgt, sk2, 9)])
+    want = bind.state_root(
+        *_arrays_of({k: tuple(v) for k, v in accounts.items()}))
+    assert eng.incremental_root(d, s) == want
