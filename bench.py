@@ -57,9 +57,13 @@ def main():
     ap.add_argument("--dirty", action="store_true",
                     help="with --incremental: dirty-path recompute "
                          "(sre_incremental_root) instead of merge+full root")
+    ap.add_argument("--incremental-slots", type=int, default=0,
+                    help="with --incremental: slots/account in the resident "
+                         "base, plus per-step storage delta rows "
+                         "(0 = the configs[4] accounts-only shape)")
     args = ap.parse_args()
     if args.incremental:
-        args.slots = 0  # accounts-only base (see config note in the output)
+        args.slots = args.incremental_slots  # configs[4] default: accounts-only
 
     import torch
 
@@ -143,11 +147,42 @@ def main():
             d[i]["balance"] = np.frombuffer(b.to_bytes(32, "big"), np.uint8)
             d[i]["code_hash"] = np.frombuffer(ch, np.uint8)
             d[i]["deleted"] = dead
-        delta = (d, np.zeros(0, dtype=STORAGE_DTYPE))
+        st_rows = np.zeros(0, dtype=STORAGE_DTYPE)
+        if args.incremental_slots > 0:
+            # idempotent storage delta: absolute-valued upserts of 4 slots
+            # per modified account (slots this delta itself introduces, so
+            # the post-state is a fixed point after the first step)
+            smsgs = torch.zeros((n_mod * 4, 12), dtype=torch.uint8,
+                                device=f"cuda:{local_rank}")
+            for b in range(8):
+                smsgs[:, b] = torch.tensor(
+                    [(c >> (8 * b)) & 0xFF for c in counters[:n_mod]
+                     for _ in range(4)], dtype=torch.uint8)
+            smsgs[:, 8] = torch.tensor([q for _ in range(n_mod)
+                                        for q in range(4)], dtype=torch.uint8)
+            smsgs[:, 9] = 0xD5  # domain tag: delta slots
+            skeys = torch.empty((n_mod * 4, 32), dtype=torch.uint8,
+                                device=f"cuda:{local_rank}")
+            eng.keccak_batch_device(smsgs, 12, skeys)
+            skeys = skeys.cpu().numpy()
+            srows = []
+            for i in range(n_mod):
+                ak = dk[i].tobytes()
+                for q in range(4):
+                    srows.append((ak, skeys[4 * i + q].tobytes(),
+                                  10**9 + 7 * i + q))
+            srows.sort()
+            st_rows = np.zeros(len(srows), dtype=STORAGE_DTYPE)
+            for i, (ak, sk, v) in enumerate(srows):
+                st_rows[i]["acct_key"] = np.frombuffer(ak, np.uint8)
+                st_rows[i]["slot_key"] = np.frombuffer(sk, np.uint8)
+                st_rows[i]["value"] = np.frombuffer(v.to_bytes(32, "big"),
+                                                    np.uint8)
+        delta = (d, st_rows)
 
     def step():
         if args.incremental and args.dirty:
-            return eng.incremental_root(delta[0])
+            return eng.incremental_root(delta[0], delta[1])
         if args.incremental:
             eng.apply_delta(*delta)
             return eng.root()
@@ -272,7 +307,10 @@ def main():
                          "base; " +
                          ("dirty-path recompute via retained cell tops"
                           if args.dirty else
-                          "apply_delta + full device recompute") + ")")
+                          "apply_delta + full device recompute") +
+                         (f"; +{args.delta_accounts * 4 * 4 // 5} storage "
+                          "delta rows" if args.incremental_slots else "") +
+                         ")")
             if args.incremental else
             (f"{args.accounts} accounts x {args.slots} slots "
              "(BASELINE configs[3] shape; full job on every N)"),
