@@ -2122,8 +2122,17 @@ static void *pool_get(sre_ctx *ctx, size_t bytes)
         }
     }
     void *p = nullptr;
-    if (hipMalloc(&p, cls) != hipSuccess)
-        return nullptr;
+    if (hipMalloc(&p, cls) != hipSuccess) {
+        // pool pressure: drop every cached buffer and retry once (a big
+        // storage merge after a full-root pass can need the HBM the pool
+        // is hoarding)
+        (void)hipDeviceSynchronize();
+        for (auto &e : ctx->pool)
+            (void)hipFree(e.second);
+        ctx->pool.clear();
+        if (hipMalloc(&p, cls) != hipSuccess)
+            return nullptr;
+    }
     return p;
 }
 
