@@ -2132,6 +2132,10 @@ static void *pool_get(sre_ctx *ctx, size_t bytes)
         ctx->pool.clear();
         if (hipMalloc(&p, cls) != hipSuccess)
             return nullptr;
+        // the failed first attempt left a sticky hipErrorOutOfMemory in
+        // the per-thread last-error slot; consume it so the next launch's
+        // hipGetLastError() check doesn't report a phantom OOM
+        (void)hipGetLastError();
     }
     return p;
 }
