@@ -891,7 +891,9 @@ __global__ void k_group_starts(const uint32_t *__restrict__ flags,
 // per-lane LDS slot capped occupancy at 4 waves/CU and left the kernel 99%
 // latency-stalled (profiles/r01_2Mx64_sq_pmc.txt); the appender keeps the
 // kernel at full occupancy and every scratch store coalesced.
-#define BLOCK_A 256u
+#ifndef BLOCK_A
+#define BLOCK_A 256
+#endif
 
 struct byte_appender {
     uint64_t cur;
