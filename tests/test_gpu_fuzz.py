@@ -34,9 +34,12 @@ def _rand_state(rng, na, slots):
     return accounts
 
 
+FUZZ_ITERS = int(__import__("os").environ.get("SRE_FUZZ_ITERS", "12"))
+
+
 def test_fuzz_roots_updates_proofs(eng):
     rng = np.random.default_rng(0xF00D)
-    for it in range(12):
+    for it in range(FUZZ_ITERS):
         na = int(rng.integers(1, 400))
         slots = int(rng.integers(0, 6))
         accounts = _rand_state(rng, na, slots)
@@ -72,7 +75,7 @@ def test_fuzz_roots_updates_proofs(eng):
 def test_fuzz_incremental_chains(eng):
     rng = np.random.default_rng(0xBEEF)
     ke = bind.keccak256(b"")
-    for it in range(5):
+    for it in range(max(5, FUZZ_ITERS // 2)):
         na = int(rng.integers(2, 300))
         accounts = {k: list(v[:3]) + [{}]
                     for k, v in _rand_state(rng, na, 0).items()}
