@@ -111,16 +111,19 @@ __device__ __forceinline__ uint64_t rotl64c(uint64_t x, int n)
 // scratch — cdna_hip_programming.md §5.4 rule 20).
 __device__ void keccak_f(uint64_t s[25])
 {
-    constexpr int ROT[25] = {
-        0, 1, 62, 28, 27, 36, 44, 6, 55, 20, 3, 10, 43, 25, 39,
-        41, 45, 15, 21, 8, 18, 2, 61, 56, 14,
-    };
-#ifndef KECCAK_UNROLL
+    #ifndef KECCAK_UNROLL
 #define KECCAK_UNROLL 2
 #endif
+    // rho+pi as the XKCP in-place cycle walk (one temp, no b[25] copy —
+    // the copy cost ~15 v_mov_b64 per round) and chi row-wise in place
+    // with two saved lanes. Validated bit-exact against FIPS-202 vectors.
+    constexpr int PILN[24] = {10, 7,  11, 17, 18, 3, 5,  16, 8,  21, 24, 4,
+                              15, 23, 19, 13, 12, 2, 20, 14, 22, 9,  6,  1};
+    constexpr int ROTC[24] = {1,  3,  6,  10, 15, 21, 28, 36, 45, 55, 2,  14,
+                              27, 41, 56, 8,  25, 43, 62, 18, 39, 61, 20, 44};
 #pragma unroll KECCAK_UNROLL
     for (int r = 0; r < 24; ++r) {
-        uint64_t c[5], d[5], b[25];
+        uint64_t c[5], d[5];
 #pragma unroll
         for (int x = 0; x < 5; ++x)
             c[x] = s[x] ^ s[x + 5] ^ s[x + 10] ^ s[x + 15] ^ s[x + 20];
@@ -130,21 +133,22 @@ __device__ void keccak_f(uint64_t s[25])
 #pragma unroll
         for (int i = 0; i < 25; ++i)
             s[i] ^= d[i % 5];
+        uint64_t t = s[1], u;
 #pragma unroll
-        for (int x = 0; x < 5; ++x) {
-#pragma unroll
-            for (int y = 0; y < 5; ++y) {
-                const int src = x + 5 * y;
-                const int dst = y + 5 * ((2 * x + 3 * y) % 5);
-                b[dst] = ROT[src] ? rotl64c(s[src], ROT[src]) : s[src];
-            }
+        for (int k = 0; k < 24; ++k) {
+            const int j = PILN[k];
+            u = s[j];
+            s[j] = rotl64c(t, ROTC[k]);
+            t = u;
         }
 #pragma unroll
-        for (int y = 0; y < 5; ++y) {
-#pragma unroll
-            for (int x = 0; x < 5; ++x)
-                s[x + 5 * y] =
-                    b[x + 5 * y] ^ ((~b[(x + 1) % 5 + 5 * y]) & b[(x + 2) % 5 + 5 * y]);
+        for (int y = 0; y < 25; y += 5) {
+            uint64_t a0 = s[y], a1 = s[y + 1];
+            s[y] ^= (~a1) & s[y + 2];
+            s[y + 1] ^= (~s[y + 2]) & s[y + 3];
+            s[y + 2] ^= (~s[y + 3]) & s[y + 4];
+            s[y + 3] ^= (~s[y + 4]) & a0;
+            s[y + 4] ^= (~a0) & a1;
         }
         s[0] ^= KRC[r];
     }
