@@ -208,7 +208,7 @@ def main():
             # the first delta replaced the resident state with the merged
             # (engine-owned) arrays; the borrowed generation tensors are
             # dead weight (62 GB at 10M x 64) — drop them
-            eng._keep = []
+            eng.release_borrowed()
             acct_t = st_t = None  # noqa: F841
             torch.cuda.empty_cache()
 

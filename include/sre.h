@@ -29,6 +29,11 @@
  *     duplicates, values nonzero (zero-valued slots are absent, not stored:
  *     crates/trie/trie/src/trie.rs:819-825), and every acct_key present in
  *     the accounts upload.
+ *   - capacity: at most 2^32 - 1 account entries and 2^32 - 1 storage
+ *     entries per context (leaf intervals are tracked as u32 indices).
+ *     That is ~4.29 billion leaves per GPU — 6.6x the headline 650M-leaf
+ *     config; shard across GPUs (sre_subtree_roots) beyond it. Exceeding
+ *     the limit is rejected at upload with an error, never truncated.
  *
  * Thread model: all calls on one host thread per ctx; the engine is
  * internally multi-stream. The engine COPIES input buffers at upload; the

@@ -2108,8 +2108,18 @@ struct sre_ctx {
     std::vector<std::pair<size_t, void *>> pool;
 };
 
+// Size classes: power-of-two up to 256 MiB (transient level-machinery
+// buffers — reuse across wildly varying level sizes), then 256 MiB-step
+// quantization. Pure pow2 would round the two giant state arrays (e.g.
+// ~61 GB of storage entries) up to the next power of two and, with the
+// apply_delta ping-pong partner, hold ~2x their footprint in dead
+// rounding at near-capacity configs; the step classes bound the waste at
+// 256 MiB per huge buffer while keeping exact-class reuse.
 static size_t pool_class(size_t bytes)
 {
+    const size_t STEP = 256ull << 20;
+    if (bytes > STEP)
+        return (bytes + STEP - 1) / STEP * STEP;
     size_t c = 256;
     while (c < bytes)
         c <<= 1;
@@ -2237,9 +2247,23 @@ extern "C" void sre_destroy(sre_ctx *ctx)
     delete ctx;
 }
 
+// leaf intervals are u32 (node_rec.s/e) — reject, never truncate (sre.h
+// capacity contract)
+static int check_cap(sre_ctx *ctx, uint64_t n)
+{
+    if (n >= 0xFFFFFFFFull) {
+        set_err(ctx, "entry count exceeds the 2^32-1 per-GPU limit "
+                     "(u32 leaf intervals); shard across GPUs");
+        return -1;
+    }
+    return 0;
+}
+
 extern "C" int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entries,
                                    uint64_t n)
 {
+    if (check_cap(ctx, n))
+        return -1;
     ctx->cells_valid = false;
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     release_acct(ctx);
@@ -2258,6 +2282,8 @@ extern "C" int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entrie
 extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries,
                                   uint64_t n)
 {
+    if (check_cap(ctx, n))
+        return -1;
     ctx->cells_valid = false;
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     release_st(ctx);
@@ -2276,6 +2302,8 @@ extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries
 // Borrow device-resident inputs (zero-copy; caller keeps them alive).
 extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint64_t n)
 {
+    if (check_cap(ctx, n))
+        return -1;
     ctx->cells_valid = false;
     release_acct(ctx);
     ctx->d_acct = (const sre_account_entry *)d_entries;
@@ -2286,6 +2314,8 @@ extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint
 
 extern "C" int sre_set_storage_device(sre_ctx *ctx, const void *d_entries, uint64_t n)
 {
+    if (check_cap(ctx, n))
+        return -1;
     ctx->cells_valid = false;
     release_st(ctx);
     ctx->d_st = (const sre_storage_entry *)d_entries;

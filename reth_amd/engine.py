@@ -161,6 +161,15 @@ class StateRootEngine:
             ctypes.c_void_p(self._ctx), ctypes.c_void_p(st_u8.data_ptr()),
             st_u8.shape[0]))
 
+    def release_borrowed(self):
+        """Drop the references keeping borrowed device tensors alive
+        (set_device_tensors). Precondition: the engine no longer reads
+        them — i.e. the resident state has been replaced by engine-owned
+        arrays (apply_delta / incremental_root adopt merged copies first),
+        or new tensors were set. After this the caller may free the
+        tensors (torch.cuda.empty_cache())."""
+        self._keep = []
+
     def apply_delta(self, acct_delta: np.ndarray, st_delta: np.ndarray):
         """Apply a HashedPostState overlay delta to the resident state
         (post-state wins, zero value deletes a slot, deleted accounts wipe
