@@ -259,6 +259,19 @@ __device__ __forceinline__ uint64_t be64_at(const uint8_t *p)
     return __builtin_bswap64(v);
 }
 
+// Dynamic byte-granular funnels over little-endian-packed byte streams
+// (byte k of a word = stream byte 8w+k). v_lshlrev_b64/v_lshrrev_b64 are
+// single VOP3 ops on CDNA, so each funnel is ~4 VALU; the double-shift
+// avoids the undefined 64-bit shift at tb == 0.
+__device__ __forceinline__ uint64_t dn8(uint64_t lo, uint64_t hi, int tb)
+{   // stream advanced by tb bytes: out byte k = concat(lo,hi) byte tb+k
+    return (lo >> (8 * tb)) | ((hi << (63 - 8 * tb)) << 1);
+}
+__device__ __forceinline__ uint64_t up8(uint64_t prev, uint64_t cur, int tb)
+{   // stream delayed by tb bytes: out byte k = (k>=tb) ? cur byte k-tb : prev tail
+    return (cur << (8 * tb)) | ((prev >> (63 - 8 * tb)) >> 1);
+}
+
 // lcp in nibbles of two 32-byte keys; sets *gt if a > b.
 __device__ __forceinline__ int key_lcp(const uint8_t *a, const uint8_t *b, bool *gt)
 {
@@ -465,6 +478,16 @@ __global__ void k_lcp_account(const sre_account_entry *__restrict__ acct, uint64
 // ---------------------------------------------------------------------------
 
 // Storage leaf: rlp <= 2 + 34 + 34 = 70 B -> always one keccak block.
+//
+// Fast path (D <= 13, i.e. the short key starts within the first 8 key
+// bytes — always true for real states; 16^14 colliding keys otherwise):
+// the hex-prefix packed short key is a byte-aligned SUFFIX of the slot
+// key (nibbles from `from` pack into key bytes [(from+1)/2, 32) whether
+// `from` is odd or even), so the whole message is three byte-streams —
+// head, key suffix, value item — combined into the 17 keccak state words
+// by register byte-funnels. No LDS traffic, no per-nibble loads, and the
+// keccak pad byte rides the value funnel as a 5th input word. The slow
+// path keeps the original byte-wise LDS assembly.
 #define SLOT_STO 136
 __global__ void __launch_bounds__(BLOCK) k_leaf_storage(
     const sre_storage_entry *__restrict__ st, uint64_t ns,
@@ -481,22 +504,181 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_storage(
 
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i < ns) {
-        uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_STO;
-        uint64_t *slot64 = (uint64_t *)slot;
-#pragma unroll
-        for (int k = 0; k < SLOT_STO / 8; ++k)
-            slot64[k] = 0;
-
         int8_t l0 = lcp[i], l1 = lcp[i + 1];
         int D = l0 > l1 ? l0 : l1;
         int from = D + 1;
 
         const uint8_t *key = st[i].slot_key;
         const uint8_t *val = st[i].value;
-        int vlen = min_be_len(val);
+
+        // entries are 96 B at a 16-B-aligned base -> slot_key/value are
+        // 16-B aligned; pull both as dwordx4 pairs once.
+        const uint4 *e4 = (const uint4 *)&st[i];
+        uint4 ka = e4[2], kc = e4[3], va = e4[4], vc = e4[5];
+        uint64_t K[6], V[6];
+        K[0] = (uint64_t)ka.x | ((uint64_t)ka.y << 32);
+        K[1] = (uint64_t)ka.z | ((uint64_t)ka.w << 32);
+        K[2] = (uint64_t)kc.x | ((uint64_t)kc.y << 32);
+        K[3] = (uint64_t)kc.z | ((uint64_t)kc.w << 32);
+        K[4] = 0; K[5] = 0;
+        V[0] = (uint64_t)va.x | ((uint64_t)va.y << 32);
+        V[1] = (uint64_t)va.z | ((uint64_t)va.w << 32);
+        V[2] = (uint64_t)vc.x | ((uint64_t)vc.y << 32);
+        V[3] = (uint64_t)vc.z | ((uint64_t)vc.w << 32);
+        V[4] = 0x01; // keccak pad start byte, rides the value funnel
+        V[5] = 0;
+
+        // vlen: value is big-endian bytes; first nonzero byte in memory
+        // order (= lowest bytes of the LE-loaded words).
+        int j0;
+        if (V[0])      j0 = __builtin_ctzll(V[0]) >> 3;
+        else if (V[1]) j0 = 8 + (__builtin_ctzll(V[1]) >> 3);
+        else if (V[2]) j0 = 16 + (__builtin_ctzll(V[2]) >> 3);
+        else if (V[3]) j0 = 24 + (__builtin_ctzll(V[3]) >> 3);
+        else           j0 = 32;
+        int vlen = 32 - j0;
         if (vlen == 0) {
             atomicOr(err, 1u << E_ZERO_VALUE);
+        } else if (D <= 13) {
+            // ---- fast register path ----
+            int kb = (from + 1) >> 1; // suffix start byte, 0..7
+            int rj = j0 & 7, q0 = j0 >> 3;
+            // value suffix stream (+pad): VS[j] = F[q0+j]
+            uint64_t F0 = dn8(V[0], V[1], rj), F1 = dn8(V[1], V[2], rj),
+                     F2 = dn8(V[2], V[3], rj), F3 = dn8(V[3], V[4], rj),
+                     F4 = dn8(V[4], V[5], rj);
+            uint64_t VS0 = q0 == 0 ? F0 : q0 == 1 ? F1 : q0 == 2 ? F2 : F3;
+            uint64_t VS1 = q0 == 0 ? F1 : q0 == 1 ? F2 : q0 == 2 ? F3 : F4;
+            uint64_t VS2 = q0 == 0 ? F2 : q0 == 1 ? F3 : q0 == 2 ? F4 : 0;
+            uint64_t VS3 = q0 == 0 ? F3 : q0 == 1 ? F4 : 0;
+            uint64_t VS4 = q0 == 0 ? F4 : 0;
+            int top = (int)(VS0 & 0xFF);
+            // storage value = encode_fixed_size(U256) (trie.rs:824) wrapped
+            // once more as a string item; both prefix bytes appear together
+            int vrlp_len = (vlen == 1 && top < 0x80) ? 1 : 1 + vlen;
+            int vitem_len = vrlp_len == 1 ? 1 : 1 + vrlp_len;
+            uint64_t VI0, VI1, VI2, VI3, VI4;
+            if (vrlp_len == 1) {
+                VI0 = VS0; VI1 = VS1; VI2 = VS2; VI3 = VS3; VI4 = VS4;
+            } else {
+                uint64_t pfx = (uint64_t)(0x80 + vrlp_len) |
+                               ((uint64_t)(0x80 + vlen) << 8);
+                VI0 = pfx | (VS0 << 16);
+                VI1 = (VS1 << 16) | (VS0 >> 48);
+                VI2 = (VS2 << 16) | (VS1 >> 48);
+                VI3 = (VS3 << 16) | (VS2 >> 48);
+                VI4 = (VS4 << 16) | (VS3 >> 48);
+            }
+            int n = 63 - D;  // short-key nibbles (>= 50 here)
+            int odd = n & 1;
+            int hb = 1 + (n >> 1);
+            int payload = 1 + hb + vitem_len;
+            int h = payload < 56 ? 1 : 2;
+            int len = h + payload;
+            int hl = h + 2;
+            uint8_t first = (uint8_t)(0x20 | (odd ? 0x10 |
+                ((int)(K[0] >> ((8 * kb - 8) & 63)) & 0xF) : 0));
+            uint64_t head;
+            if (h == 1)
+                head = (uint64_t)(0xc0 + payload) |
+                       ((uint64_t)(0x80 + hb) << 8) | ((uint64_t)first << 16);
+            else
+                head = 0xf8ull | ((uint64_t)payload << 8) |
+                       ((uint64_t)(0x80 + hb) << 16) | ((uint64_t)first << 24);
+            // key suffix: message byte x (x >= hl) = key byte x - t, t = hl-kb
+            int t = hl - kb; // in [-4, 4]
+            uint64_t lowmask = (1ull << (8 * hl)) - 1;
+            uint64_t m0, m1, m2, m3, m4;
+            if (t >= 0) {
+                m0 = head | (up8(0, K[0], t) & ~lowmask);
+                m1 = up8(K[0], K[1], t);
+                m2 = up8(K[1], K[2], t);
+                m3 = up8(K[2], K[3], t);
+                m4 = up8(K[3], K[4], t);
+            } else {
+                int u = -t;
+                m0 = head | (dn8(K[0], K[1], u) & ~lowmask);
+                m1 = dn8(K[1], K[2], u);
+                m2 = dn8(K[2], K[3], u);
+                m3 = dn8(K[3], K[4], u);
+                m4 = 0;
+            }
+            uint64_t m5 = 0, m6 = 0, m7 = 0, m8 = 0;
+            // value item at vo = hl + (32 - kb) = 32 + t (28..36)
+            int vo = 32 + t;
+            int rv = vo & 7;
+            uint64_t G0 = up8(0, VI0, rv), G1 = up8(VI0, VI1, rv),
+                     G2 = up8(VI1, VI2, rv), G3 = up8(VI2, VI3, rv),
+                     G4 = up8(VI3, VI4, rv), G5 = up8(VI4, 0, rv);
+            if (vo < 32) {
+                m3 |= G0; m4 |= G1; m5 = G2; m6 = G3; m7 = G4; m8 = G5;
+            } else {
+                // vi stream <= 35 B + rv <= 4 -> G5 is provably zero here
+                m4 |= G0; m5 = G1; m6 = G2; m7 = G3; m8 = G4;
+            }
+            // single 136-B block; m9..m15 zero, end bit constant in s[16]
+            uint64_t s[25];
+            s[0] = m0; s[1] = m1; s[2] = m2; s[3] = m3; s[4] = m4;
+            s[5] = m5; s[6] = m6; s[7] = m7; s[8] = m8;
+#pragma unroll
+            for (int k = 9; k < 25; ++k)
+                s[k] = 0;
+            s[16] = 0x8000000000000000ull;
+            keccak_f(s);
+
+            uint32_t seg = seg_id[i];
+            if (len >= 32) {
+                // compose the whole 48-B record in registers, store as
+                // 3 x dwordx4 (byte-wise ref stores were address-bound)
+                uint8_t pb = D >= 0
+                    ? (uint8_t)((K[0] >> ((8 * (D >> 1) +
+                                           ((D & 1) ? 0 : 4)) & 63)) & 0xF)
+                    : 0;
+                uint32_t w[12];
+                w[0] = (uint32_t)i;
+                w[1] = (uint32_t)(i + 1);
+                w[2] = seg;
+                w[3] = (uint32_t)(uint8_t)(int8_t)D | (33u << 8) |
+                       (0xa0u << 16) | ((uint32_t)(s[0] & 0xFF) << 24);
+                w[4] = (uint32_t)(s[0] >> 8);
+                w[5] = (uint32_t)((s[0] >> 40) | (s[1] << 24));
+                w[6] = (uint32_t)(s[1] >> 8);
+                w[7] = (uint32_t)((s[1] >> 40) | (s[2] << 24));
+                w[8] = (uint32_t)(s[2] >> 8);
+                w[9] = (uint32_t)((s[2] >> 40) | (s[3] << 24));
+                w[10] = (uint32_t)(s[3] >> 8);
+                w[11] = (uint32_t)((s[3] >> 40) & 0xFFFFFF) |
+                        ((uint32_t)pb << 24);
+                uint4 *r4 = (uint4 *)&recs[i];
+                r4[0] = make_uint4(w[0], w[1], w[2], w[3]);
+                r4[1] = make_uint4(w[4], w[5], w[6], w[7]);
+                r4[2] = make_uint4(w[8], w[9], w[10], w[11]);
+            } else {
+                // inline (<32 B) leaf ref: rare; message bytes from m0..m3
+                node_rec *r = &recs[i];
+                r->s = (uint32_t)i;
+                r->e = (uint32_t)(i + 1);
+                r->seg = seg;
+                r->depth = (int8_t)D;
+                r->ref_len = (uint8_t)len;
+                uint64_t mm[4] = {m0, m1, m2, m3};
+                for (int k = 0; k < len; ++k)
+                    r->ref[k] = (uint8_t)(mm[k >> 3] >> (8 * (k & 7)));
+                r->pad_ = D >= 0 ? nib_of(key, D) : 0;
+            }
+            depths[i] = (uint8_t)(D + 1);
+            atomicAdd(&hist_l[D + 1], 1u);
+            if (D == -1) { // single-slot storage trie: root = keccak(leaf RLP)
+                uint64_t *sr = (uint64_t *)(seg_roots + 32ull * seg);
+                sr[0] = s[0]; sr[1] = s[1]; sr[2] = s[2]; sr[3] = s[3];
+            }
         } else {
+            // ---- slow LDS path (deep shared prefixes) ----
+            uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_STO;
+            uint64_t *slot64 = (uint64_t *)slot;
+#pragma unroll
+            for (int k = 0; k < SLOT_STO / 8; ++k)
+                slot64[k] = 0;
             // storage value = encode_fixed_size(U256) (trie.rs:824); the leaf
             // stores that RLP as a string item -> wrap once more.
             int vrlp_len = (vlen == 1 && val[31] < 0x80) ? 1 : 1 + vlen;
@@ -529,7 +711,7 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_storage(
             r->pad_ = D >= 0 ? nib_of(key, D) : 0;
             depths[i] = (uint8_t)(D + 1);
             atomicAdd(&hist_l[D + 1], 1u);
-            if (D == -1) // single-slot storage trie: root = keccak(leaf RLP)
+            if (D == -1)
                 memcpy(seg_roots + 32ull * seg, hash, 32);
         }
     }
@@ -934,6 +1116,31 @@ struct byte_appender {
             cur = 0;
         }
     }
+    // append nbytes (1..33) from the LE-packed word stream R[0..4]
+    // (stream byte k = R[k/8] byte k%8): word funnels + predicated flushes
+    // instead of per-byte flush checks. Garbage bytes of R beyond nbytes
+    // never reach the scratch: flushed words only cover stream bytes
+    // < pos+nbytes, and the carried partial word is masked.
+    __device__ __forceinline__ void put_bytes33(const uint64_t R[5], int nbytes)
+    {
+        int total = pos + nbytes;
+        uint64_t S0 = cur | (R[0] << (8 * pos));
+        uint64_t S1 = up8(R[0], R[1], pos);
+        uint64_t S2 = up8(R[1], R[2], pos);
+        uint64_t S3 = up8(R[2], R[3], pos);
+        uint64_t S4 = up8(R[3], R[4], pos);
+        int F = total >> 3;
+        if (F > 0) *slot64(widx) = S0;
+        if (F > 1) *slot64(widx + 1) = S1;
+        if (F > 2) *slot64(widx + 2) = S2;
+        if (F > 3) *slot64(widx + 3) = S3;
+        if (F > 4) *slot64(widx + 4) = S4;
+        widx += F;
+        cur = F == 0 ? S0 : F == 1 ? S1 : F == 2 ? S2 : F == 3 ? S3
+            : F == 4 ? S4 : 0;
+        pos = total & 7;
+        cur &= pos ? (1ull << (8 * pos)) - 1 : 0;
+    }
     // finish the message: flush, zero-fill to the last word of block nb,
     // and set the keccak pad end bit (0x80 in the final byte).
     __device__ __forceinline__ void finish(int nb)
@@ -1081,12 +1288,19 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
 #pragma unroll
                 for (int k = 0; k < 9; ++k)
                     w[k] = rec32[3 + k]; // bytes 12..48 of the record
-                // ref byte i lives at record byte 14+i
-#pragma unroll
-                for (int i = 0; i < 33; ++i)
-                    if (i < rl)
-                        ap.put((uint8_t)(w[((14 + i) >> 2) - 3]
-                                         >> (8 * ((14 + i) & 3))));
+                // ref byte i lives at record byte 14+i: repack into the
+                // LE word stream R and append word-wise
+                uint64_t R[5];
+                R[0] = ((uint64_t)w[0] >> 16) | ((uint64_t)w[1] << 16) |
+                       ((uint64_t)w[2] << 48);
+                R[1] = ((uint64_t)w[2] >> 16) | ((uint64_t)w[3] << 16) |
+                       ((uint64_t)w[4] << 48);
+                R[2] = ((uint64_t)w[4] >> 16) | ((uint64_t)w[5] << 16) |
+                       ((uint64_t)w[6] << 48);
+                R[3] = ((uint64_t)w[6] >> 16) | ((uint64_t)w[7] << 16) |
+                       ((uint64_t)w[8] << 48);
+                R[4] = ((uint64_t)w[8] >> 16) & 0xFF;
+                ap.put_bytes33(R, rl);
                 m++;
             } else {
                 ap.put(0x80);
@@ -1131,11 +1345,11 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     if (active) {
         mt = meta[g]; // meta/scratch by thread slot; record by group index
         r = &out[perm ? perm[g] : g];
-        r->s = mt.s;
-        r->e = mt.e;
-        r->seg = mt.seg;
-        r->pad_ = 0;
         if (mt.br_len == 0) { // error group: dead record (err already flagged)
+            r->s = mt.s;
+            r->e = mt.e;
+            r->seg = mt.seg;
+            r->pad_ = 0;
             r->depth = -1;
             r->ref_len = 0;
             active = false;
@@ -1168,39 +1382,46 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         for (int i = 0; i < 4; ++i)
             br_hash[i] = s[i];
     }
-    uint8_t br_ref[33], br_ref_len;
-    // make_ref needs the raw rlp only for the inline (<32 B) case
-    if (mt.br_len < 32) {
+    // branch ref as an LE word stream (bytes beyond br_ref_len are never
+    // consumed: the assemble appender masks them, record stores carry them
+    // deterministically)
+    uint64_t br_refw[5];
+    uint8_t br_ref_len;
+    if (mt.br_len < 32) { // inline: raw rlp words from the scratch slot
         br_ref_len = (uint8_t)mt.br_len;
 #pragma unroll
-        for (int k = 0; k < 31; ++k)
-            if (k < br_ref_len)
-    #if SRE_SCRATCH_ROWMAJOR
-            br_ref[k] = (uint8_t)(scr64[(uint64_t)g * (SLOT_BR_ROW / 8) + (k >> 3)]
-                                      >> (8 * (k & 7)));
+        for (int k = 0; k < 4; ++k)
+#if SRE_SCRATCH_ROWMAJOR
+            br_refw[k] = scr64[(uint64_t)g * (SLOT_BR_ROW / 8) + k];
 #else
-            br_ref[k] = (uint8_t)(scr64[(uint64_t)(k >> 3) * scratch_stride + g]
-                                      >> (8 * (k & 7)));
+            br_refw[k] = scr64[(uint64_t)k * scratch_stride + g];
 #endif
+        br_refw[4] = 0;
     } else {
         br_ref_len = 33;
-        br_ref[0] = 0xa0;
-        memcpy(br_ref + 1, br_hash, 32);
+        br_refw[0] = 0xa0ull | (br_hash[0] << 8);
+        br_refw[1] = (br_hash[0] >> 56) | (br_hash[1] << 8);
+        br_refw[2] = (br_hash[1] >> 56) | (br_hash[2] << 8);
+        br_refw[3] = (br_hash[2] >> 56) | (br_hash[3] << 8);
+        br_refw[4] = br_hash[3] >> 56;
     }
     int kblocks = nblocks;
     const uint8_t *key0 = keys + (uint64_t)mt.s * key_stride;
     uint8_t *ext = lds + (uint64_t)threadIdx.x * SLOT_EXT;
     uint64_t *ext64 = (uint64_t *)ext;
 
-    // extension wrap over key0[from..d) in the LDS slot (<= 72 B)
-    auto wrap = [&](int from, uint64_t hash[4], uint8_t *ref, uint8_t *ref_len) {
+    // extension wrap over key0[from..d): result in whash/wrefw/wrl
+    uint64_t whash[4], wrefw[5];
+    uint8_t wrl;
+    auto wrap = [&](int from) {
         if (d == from) {
 #pragma unroll
             for (int k = 0; k < 4; ++k)
-                hash[k] = br_hash[k];
-            for (int k = 0; k < br_ref_len; ++k)
-                ref[k] = br_ref[k];
-            *ref_len = br_ref_len;
+                whash[k] = br_hash[k];
+#pragma unroll
+            for (int k = 0; k < 5; ++k)
+                wrefw[k] = br_refw[k];
+            wrl = br_ref_len;
             return;
         }
 #pragma unroll
@@ -1209,8 +1430,11 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         int pay = hp_item_len(from, d) + br_ref_len;
         int hh = rlp_list_hdr_write(ext, pay);
         int p = hh + hp_item_write(ext + hh, key0, from, d, 0);
-        for (int k = 0; k < br_ref_len; ++k)
-            ext[p++] = br_ref[k];
+#pragma unroll
+        for (int k = 0; k < 33; ++k)
+            if (k < br_ref_len)
+                ext[p + k] = (uint8_t)(br_refw[k >> 3] >> (8 * (k & 7)));
+        p += br_ref_len;
         int len = hh + pay; // <= 69 < SLOT_EXT
         ext[len] = 0x01;    // pad start; end bit lands in lane 16 below
         // single 136-B keccak block: absorb SLOT_EXT bytes + implicit zeros
@@ -1225,15 +1449,20 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         keccak_f(s);
 #pragma unroll
         for (int k = 0; k < 4; ++k)
-            hash[k] = s[k];
+            whash[k] = s[k];
         if (len < 32) {
-            *ref_len = (uint8_t)len;
-            for (int k = 0; k < len; ++k)
-                ref[k] = ext[k];
+            wrl = (uint8_t)len;
+#pragma unroll
+            for (int k = 0; k < 4; ++k)
+                wrefw[k] = ext64[k];
+            wrefw[4] = 0;
         } else {
-            *ref_len = 33;
-            ref[0] = 0xa0;
-            memcpy(ref + 1, hash, 32);
+            wrl = 33;
+            wrefw[0] = 0xa0ull | (s[0] << 8);
+            wrefw[1] = (s[0] >> 56) | (s[1] << 8);
+            wrefw[2] = (s[1] >> 56) | (s[2] << 8);
+            wrefw[3] = (s[2] >> 56) | (s[3] << 8);
+            wrefw[4] = s[3] >> 56;
         }
         kblocks += 1;
     };
@@ -1250,27 +1479,42 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     uint8_t upd_bits = (uint8_t)(((mt.flags & 1) << 4) |
                                  ((mt.br_len >= 32 ? 1 : 0) << 5));
     if (mt.P >= 0) {
-        r->depth = mt.P;
-        uint64_t hash[4];
-        uint8_t rl;
-        wrap(mt.P + 1, hash, r->ref, &rl);
-        r->ref_len = rl;
-        r->pad_ = (uint8_t)(nib_of(key0, mt.P) | upd_bits);
+        wrap(mt.P + 1);
+        uint8_t pb = (uint8_t)(nib_of(key0, mt.P) | upd_bits);
+        // whole 48-B record composed in registers, 3 x dwordx4 stores
+        uint32_t w3 = (uint32_t)(uint8_t)(int8_t)mt.P | ((uint32_t)wrl << 8) |
+                      ((uint32_t)(wrefw[0] & 0xFFFF) << 16);
+        uint4 *r4 = (uint4 *)r;
+        r4[0] = make_uint4(mt.s, mt.e, mt.seg, w3);
+        r4[1] = make_uint4((uint32_t)(wrefw[0] >> 16),
+                           (uint32_t)((wrefw[0] >> 48) | (wrefw[1] << 16)),
+                           (uint32_t)(wrefw[1] >> 16),
+                           (uint32_t)((wrefw[1] >> 48) | (wrefw[2] << 16)));
+        r4[2] = make_uint4((uint32_t)(wrefw[2] >> 16),
+                           (uint32_t)((wrefw[2] >> 48) | (wrefw[3] << 16)),
+                           (uint32_t)(wrefw[3] >> 16),
+                           (uint32_t)((wrefw[3] >> 48) & 0xFFFF) |
+                               ((uint32_t)(wrefw[4] & 0xFF) << 16) |
+                               ((uint32_t)pb << 24));
         atomicAdd(&hist_l[mt.P + 1], 1u);
     } else {
+        r->s = mt.s;
+        r->e = mt.e;
+        r->seg = mt.seg;
+        r->pad_ = 0;
         r->depth = -1;
         r->ref_len = 0;
-        uint64_t hash[4];
-        uint8_t ref_len;
         if (subtree) {
-            wrap(1, hash, child_refs + 33ull * mt.seg, &ref_len);
-            child_lens[mt.seg] = ref_len;
+            wrap(1);
+            uint8_t *cr = child_refs + 33ull * mt.seg;
+#pragma unroll
+            for (int k = 0; k < 33; ++k)
+                if (k < wrl)
+                    cr[k] = (uint8_t)(wrefw[k >> 3] >> (8 * (k & 7)));
+            child_lens[mt.seg] = wrl;
         }
-        // standalone form: only the hash matters; write the ref into the
-        // (dead) record's ref bytes to keep everything in global memory
-        wrap(0, hash, r->ref, &ref_len);
-        memcpy(seg_roots + 32ull * mt.seg, hash, 32);
-        r->ref_len = 0;
+        wrap(0); // standalone form: only the hash matters
+        memcpy(seg_roots + 32ull * mt.seg, whash, 32);
     }
     atomicAdd(&hist_l[65], (uint32_t)kblocks);
     } // active
