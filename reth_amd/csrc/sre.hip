@@ -44,6 +44,7 @@
 #include <vector>
 #include <map>
 #include <array>
+#include <memory>
 
 #include "../../include/sre.h"
 
@@ -996,6 +997,51 @@ __global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
     copy_rec(&out[i + lo], &A[i]);
 }
 
+// k-way merge of sorted runs (distinct .s keys across all runs): each
+// element's output position = own index + sum of lower-bound ranks in the
+// other runs. Replaces the eager re-merge of per-depth carries: every
+// branch output is positioned exactly once, when its level is consumed.
+#define KWAY_MAX 12
+struct kway_desc {
+    const node_rec *run[KWAY_MAX];
+    uint64_t cnt[KWAY_MAX];
+    uint64_t acc[KWAY_MAX + 1]; // exclusive prefix of cnt
+    int nruns;
+};
+
+__global__ void k_merge_kway(kway_desc kd, uint64_t total,
+                             node_rec *__restrict__ out)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= total)
+        return;
+    int r = 0;
+#pragma unroll
+    for (int k = 1; k < KWAY_MAX; ++k)
+        if (k < kd.nruns && i >= kd.acc[k])
+            r = k;
+    uint64_t idx = i - kd.acc[r];
+    const node_rec *me = &kd.run[r][idx];
+    uint32_t key = me->s;
+    uint64_t pos = idx;
+#pragma unroll
+    for (int k = 0; k < KWAY_MAX; ++k) {
+        if (k >= kd.nruns || k == r)
+            continue;
+        const node_rec *B = kd.run[k];
+        uint64_t lo = 0, hi = kd.cnt[k];
+        while (lo < hi) {
+            uint64_t mid = (lo + hi) / 2;
+            if (B[mid].s < key)
+                lo = mid + 1;
+            else
+                hi = mid;
+        }
+        pos += lo;
+    }
+    copy_rec(&out[pos], me);
+}
+
 __global__ void k_merge_b(const node_rec *__restrict__ A, uint64_t nA,
                           const node_rec *__restrict__ B, uint64_t nB,
                           node_rec *__restrict__ out)
@@ -1100,6 +1146,8 @@ struct byte_appender {
     }
     __device__ __forceinline__ uint64_t *slot64(int w)
     {
+        if (stride == 0) // direct row (fused kernel's LDS slot)
+            return base + w;
 #if SRE_SCRATCH_ROWMAJOR
         return base + (uint64_t)g * (SLOT_BR_ROW / 8) + w;
 #else
@@ -1314,9 +1362,157 @@ __global__ void __launch_bounds__(BLOCK_A) k_branch_assemble(
     meta[g] = mt;
 }
 
+// Everything after a branch's RLP keccak: ref composition, extension/root
+// wraps, record/bhash/seg-root emission and the pending-histogram bumps.
+// Shared by the split hash kernel and the fused 1-block kernel.
+// `inline_src64` points at the branch RLP's word 0 for the <32-B inline
+// case, with `inline_stride` u64s between message words (column-major
+// scratch: stride = chunk; LDS row: stride = 1). `ext` is a per-lane LDS
+// scratch of >= SLOT_EXT bytes (may alias the message slot: the inline
+// words are consumed before the first wrap reuses it).
+#define SLOT_EXT 88
+__device__ __forceinline__ void branch_tail(
+    const br_meta &mt, int subtree, const uint8_t *keys, uint64_t key_stride,
+    const uint64_t br_hash[4], const uint64_t *inline_src64,
+    uint64_t inline_stride, uint8_t *ext, node_rec *r, uint8_t *seg_roots,
+    uint8_t *child_refs, uint8_t *child_lens, uint32_t *hist_l,
+    uint8_t *bhash_by_s, sre_update_row *urows, uint32_t urowidx_val)
+{
+    int d = mt.d;
+    int kblocks = mt.br_len / 136 + 1;
+    uint64_t *ext64 = (uint64_t *)ext;
+    // branch ref as an LE word stream (bytes beyond br_ref_len are never
+    // consumed: the assemble appender masks them, record stores carry them
+    // deterministically)
+    uint64_t br_refw[5];
+    uint8_t br_ref_len;
+    if (mt.br_len < 32) { // inline: raw rlp words
+        br_ref_len = (uint8_t)mt.br_len;
+#pragma unroll
+        for (int k = 0; k < 4; ++k)
+            br_refw[k] = inline_src64[(uint64_t)k * inline_stride];
+        br_refw[4] = 0;
+    } else {
+        br_ref_len = 33;
+        br_refw[0] = 0xa0ull | (br_hash[0] << 8);
+        br_refw[1] = (br_hash[0] >> 56) | (br_hash[1] << 8);
+        br_refw[2] = (br_hash[1] >> 56) | (br_hash[2] << 8);
+        br_refw[3] = (br_hash[2] >> 56) | (br_hash[3] << 8);
+        br_refw[4] = br_hash[3] >> 56;
+    }
+    const uint8_t *key0 = keys + (uint64_t)mt.s * key_stride;
+
+    // extension wrap over key0[from..d): result in whash/wrefw/wrl
+    uint64_t whash[4], wrefw[5];
+    uint8_t wrl;
+    auto wrap = [&](int from) {
+        if (d == from) {
+#pragma unroll
+            for (int k = 0; k < 4; ++k)
+                whash[k] = br_hash[k];
+#pragma unroll
+            for (int k = 0; k < 5; ++k)
+                wrefw[k] = br_refw[k];
+            wrl = br_ref_len;
+            return;
+        }
+#pragma unroll
+        for (int k = 0; k < SLOT_EXT / 8; ++k)
+            ext64[k] = 0;
+        int pay = hp_item_len(from, d) + br_ref_len;
+        int hh = rlp_list_hdr_write(ext, pay);
+        int p = hh + hp_item_write(ext + hh, key0, from, d, 0);
+#pragma unroll
+        for (int k = 0; k < 33; ++k)
+            if (k < br_ref_len)
+                ext[p + k] = (uint8_t)(br_refw[k >> 3] >> (8 * (k & 7)));
+        p += br_ref_len;
+        int len = hh + pay; // <= 69 < SLOT_EXT
+        ext[len] = 0x01;    // pad start; end bit lands in lane 16 below
+        // single 136-B keccak block: absorb SLOT_EXT bytes + implicit zeros
+        uint64_t s[25];
+#pragma unroll
+        for (int i = 0; i < 25; ++i)
+            s[i] = 0;
+#pragma unroll
+        for (int i = 0; i < SLOT_EXT / 8; ++i)
+            s[i] ^= ext64[i];
+        s[16] ^= 0x8000000000000000ULL; // pad end bit of the 136-B block
+        keccak_f(s);
+#pragma unroll
+        for (int k = 0; k < 4; ++k)
+            whash[k] = s[k];
+        if (len < 32) {
+            wrl = (uint8_t)len;
+#pragma unroll
+            for (int k = 0; k < 4; ++k)
+                wrefw[k] = ext64[k];
+            wrefw[4] = 0;
+        } else {
+            wrl = 33;
+            wrefw[0] = 0xa0ull | (s[0] << 8);
+            wrefw[1] = (s[0] >> 56) | (s[1] << 8);
+            wrefw[2] = (s[1] >> 56) | (s[2] << 8);
+            wrefw[3] = (s[2] >> 56) | (s[3] << 8);
+            wrefw[4] = s[3] >> 56;
+        }
+        kblocks += 1;
+    };
+
+    if (bhash_by_s) {
+        memcpy(bhash_by_s + 32ull * mt.s, br_hash, 32);
+        if (mt.d == 0 && urows && urowidx_val != 0xFFFFFFFFu) {
+            // path-[] row: root branch carries its own hash
+            sre_update_row *ur = &urows[urowidx_val];
+            ur->root_hash_set = 1;
+            memcpy(ur->root_hash, br_hash, 32);
+        }
+    }
+    uint8_t upd_bits = (uint8_t)(((mt.flags & 1) << 4) |
+                                 ((mt.br_len >= 32 ? 1 : 0) << 5));
+    if (mt.P >= 0) {
+        wrap(mt.P + 1);
+        uint8_t pb = (uint8_t)(nib_of(key0, mt.P) | upd_bits);
+        // whole 48-B record composed in registers, 3 x dwordx4 stores
+        uint32_t w3 = (uint32_t)(uint8_t)(int8_t)mt.P | ((uint32_t)wrl << 8) |
+                      ((uint32_t)(wrefw[0] & 0xFFFF) << 16);
+        uint4 *r4 = (uint4 *)r;
+        r4[0] = make_uint4(mt.s, mt.e, mt.seg, w3);
+        r4[1] = make_uint4((uint32_t)(wrefw[0] >> 16),
+                           (uint32_t)((wrefw[0] >> 48) | (wrefw[1] << 16)),
+                           (uint32_t)(wrefw[1] >> 16),
+                           (uint32_t)((wrefw[1] >> 48) | (wrefw[2] << 16)));
+        r4[2] = make_uint4((uint32_t)(wrefw[2] >> 16),
+                           (uint32_t)((wrefw[2] >> 48) | (wrefw[3] << 16)),
+                           (uint32_t)(wrefw[3] >> 16),
+                           (uint32_t)((wrefw[3] >> 48) & 0xFFFF) |
+                               ((uint32_t)(wrefw[4] & 0xFF) << 16) |
+                               ((uint32_t)pb << 24));
+        atomicAdd(&hist_l[mt.P + 1], 1u);
+    } else {
+        r->s = mt.s;
+        r->e = mt.e;
+        r->seg = mt.seg;
+        r->pad_ = 0;
+        r->depth = -1;
+        r->ref_len = 0;
+        if (subtree) {
+            wrap(1);
+            uint8_t *cr = child_refs + 33ull * mt.seg;
+#pragma unroll
+            for (int k = 0; k < 33; ++k)
+                if (k < wrl)
+                    cr[k] = (uint8_t)(wrefw[k >> 3] >> (8 * (k & 7)));
+            child_lens[mt.seg] = wrl;
+        }
+        wrap(0); // standalone form: only the hash matters
+        memcpy(seg_roots + 32ull * mt.seg, whash, 32);
+    }
+    atomicAdd(&hist_l[65], (uint32_t)kblocks);
+}
+
 // Hash one branch per lane: absorb the padded slot straight from global,
 // then do extension/root wraps in a small LDS slot (88 B -> high occupancy).
-#define SLOT_EXT 88
 __global__ void __launch_bounds__(BLOCK) k_branch_hash(
     const uint8_t *__restrict__ scratch, uint64_t scratch_stride,
     const br_meta *__restrict__ meta,
@@ -1382,147 +1578,149 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
         for (int i = 0; i < 4; ++i)
             br_hash[i] = s[i];
     }
-    // branch ref as an LE word stream (bytes beyond br_ref_len are never
-    // consumed: the assemble appender masks them, record stores carry them
-    // deterministically)
-    uint64_t br_refw[5];
-    uint8_t br_ref_len;
-    if (mt.br_len < 32) { // inline: raw rlp words from the scratch slot
-        br_ref_len = (uint8_t)mt.br_len;
-#pragma unroll
-        for (int k = 0; k < 4; ++k)
 #if SRE_SCRATCH_ROWMAJOR
-            br_refw[k] = scr64[(uint64_t)g * (SLOT_BR_ROW / 8) + k];
+    const uint64_t *inl = scr64 + (uint64_t)g * (SLOT_BR_ROW / 8);
+    const uint64_t inl_stride = 1;
 #else
-            br_refw[k] = scr64[(uint64_t)k * scratch_stride + g];
+    const uint64_t *inl = scr64 + g;
+    const uint64_t inl_stride = scratch_stride;
 #endif
-        br_refw[4] = 0;
-    } else {
-        br_ref_len = 33;
-        br_refw[0] = 0xa0ull | (br_hash[0] << 8);
-        br_refw[1] = (br_hash[0] >> 56) | (br_hash[1] << 8);
-        br_refw[2] = (br_hash[1] >> 56) | (br_hash[2] << 8);
-        br_refw[3] = (br_hash[2] >> 56) | (br_hash[3] << 8);
-        br_refw[4] = br_hash[3] >> 56;
-    }
-    int kblocks = nblocks;
-    const uint8_t *key0 = keys + (uint64_t)mt.s * key_stride;
-    uint8_t *ext = lds + (uint64_t)threadIdx.x * SLOT_EXT;
-    uint64_t *ext64 = (uint64_t *)ext;
-
-    // extension wrap over key0[from..d): result in whash/wrefw/wrl
-    uint64_t whash[4], wrefw[5];
-    uint8_t wrl;
-    auto wrap = [&](int from) {
-        if (d == from) {
-#pragma unroll
-            for (int k = 0; k < 4; ++k)
-                whash[k] = br_hash[k];
-#pragma unroll
-            for (int k = 0; k < 5; ++k)
-                wrefw[k] = br_refw[k];
-            wrl = br_ref_len;
-            return;
-        }
-#pragma unroll
-        for (int k = 0; k < SLOT_EXT / 8; ++k)
-            ext64[k] = 0;
-        int pay = hp_item_len(from, d) + br_ref_len;
-        int hh = rlp_list_hdr_write(ext, pay);
-        int p = hh + hp_item_write(ext + hh, key0, from, d, 0);
-#pragma unroll
-        for (int k = 0; k < 33; ++k)
-            if (k < br_ref_len)
-                ext[p + k] = (uint8_t)(br_refw[k >> 3] >> (8 * (k & 7)));
-        p += br_ref_len;
-        int len = hh + pay; // <= 69 < SLOT_EXT
-        ext[len] = 0x01;    // pad start; end bit lands in lane 16 below
-        // single 136-B keccak block: absorb SLOT_EXT bytes + implicit zeros
-        uint64_t s[25];
-#pragma unroll
-        for (int i = 0; i < 25; ++i)
-            s[i] = 0;
-#pragma unroll
-        for (int i = 0; i < SLOT_EXT / 8; ++i)
-            s[i] ^= ext64[i];
-        s[16] ^= 0x8000000000000000ULL; // pad end bit of the 136-B block
-        keccak_f(s);
-#pragma unroll
-        for (int k = 0; k < 4; ++k)
-            whash[k] = s[k];
-        if (len < 32) {
-            wrl = (uint8_t)len;
-#pragma unroll
-            for (int k = 0; k < 4; ++k)
-                wrefw[k] = ext64[k];
-            wrefw[4] = 0;
-        } else {
-            wrl = 33;
-            wrefw[0] = 0xa0ull | (s[0] << 8);
-            wrefw[1] = (s[0] >> 56) | (s[1] << 8);
-            wrefw[2] = (s[1] >> 56) | (s[2] << 8);
-            wrefw[3] = (s[2] >> 56) | (s[3] << 8);
-            wrefw[4] = s[3] >> 56;
-        }
-        kblocks += 1;
-    };
-
-    if (bhash_by_s) {
-        memcpy(bhash_by_s + 32ull * mt.s, br_hash, 32);
-        if (mt.d == 0 && urowidx && urowidx[g] != 0xFFFFFFFFu) {
-            // path-[] row: root branch carries its own hash
-            sre_update_row *ur = &urows[urowidx[g]];
-            ur->root_hash_set = 1;
-            memcpy(ur->root_hash, br_hash, 32);
-        }
-    }
-    uint8_t upd_bits = (uint8_t)(((mt.flags & 1) << 4) |
-                                 ((mt.br_len >= 32 ? 1 : 0) << 5));
-    if (mt.P >= 0) {
-        wrap(mt.P + 1);
-        uint8_t pb = (uint8_t)(nib_of(key0, mt.P) | upd_bits);
-        // whole 48-B record composed in registers, 3 x dwordx4 stores
-        uint32_t w3 = (uint32_t)(uint8_t)(int8_t)mt.P | ((uint32_t)wrl << 8) |
-                      ((uint32_t)(wrefw[0] & 0xFFFF) << 16);
-        uint4 *r4 = (uint4 *)r;
-        r4[0] = make_uint4(mt.s, mt.e, mt.seg, w3);
-        r4[1] = make_uint4((uint32_t)(wrefw[0] >> 16),
-                           (uint32_t)((wrefw[0] >> 48) | (wrefw[1] << 16)),
-                           (uint32_t)(wrefw[1] >> 16),
-                           (uint32_t)((wrefw[1] >> 48) | (wrefw[2] << 16)));
-        r4[2] = make_uint4((uint32_t)(wrefw[2] >> 16),
-                           (uint32_t)((wrefw[2] >> 48) | (wrefw[3] << 16)),
-                           (uint32_t)(wrefw[3] >> 16),
-                           (uint32_t)((wrefw[3] >> 48) & 0xFFFF) |
-                               ((uint32_t)(wrefw[4] & 0xFF) << 16) |
-                               ((uint32_t)pb << 24));
-        atomicAdd(&hist_l[mt.P + 1], 1u);
-    } else {
-        r->s = mt.s;
-        r->e = mt.e;
-        r->seg = mt.seg;
-        r->pad_ = 0;
-        r->depth = -1;
-        r->ref_len = 0;
-        if (subtree) {
-            wrap(1);
-            uint8_t *cr = child_refs + 33ull * mt.seg;
-#pragma unroll
-            for (int k = 0; k < 33; ++k)
-                if (k < wrl)
-                    cr[k] = (uint8_t)(wrefw[k >> 3] >> (8 * (k & 7)));
-            child_lens[mt.seg] = wrl;
-        }
-        wrap(0); // standalone form: only the hash matters
-        memcpy(seg_roots + 32ull * mt.seg, whash, 32);
-    }
-    atomicAdd(&hist_l[65], (uint32_t)kblocks);
+    branch_tail(mt, subtree, keys, key_stride, br_hash, inl, inl_stride,
+                lds + (uint64_t)threadIdx.x * SLOT_EXT, r, seg_roots,
+                child_refs, child_lens, hist_l, bhash_by_s, urows,
+                urowidx ? urowidx[g] : 0xFFFFFFFFu);
     } // active
     __syncthreads();
     if (threadIdx.x < 66 && hist_l[threadIdx.x])
         atomicAdd(&pending[threadIdx.x], hist_l[threadIdx.x]);
 }
 
+
+// Fused assemble+hash for the 1-block branch class (nmem <= 3, the
+// majority of groups at uniform keys): the branch RLP is built in a
+// per-lane 136-B LDS slot and hashed in the same kernel — no global
+// scratch round trip, no meta array, no second kernel pass. Runs on the
+// main stream concurrently with the split pipeline's assemble (stream2)
+// handling classes 1-3. Not used in updates/proof mode (those read meta/
+// scratch across kernels).
+#define SLOT_F1 136
+__global__ void __launch_bounds__(BLOCK) k_branch_fused1(
+    const node_rec *__restrict__ L, const uint32_t *__restrict__ gs,
+    uint32_t n_groups /* class-0 groups of this chunk */,
+    const int8_t *__restrict__ lcp, const uint8_t *__restrict__ keys,
+    uint64_t key_stride, int d, int subtree, node_rec *__restrict__ out,
+    const uint32_t *__restrict__ perm /* class-0 slice */,
+    uint8_t *__restrict__ seg_roots, uint8_t *__restrict__ child_refs,
+    uint8_t *__restrict__ child_lens, uint32_t *__restrict__ pending,
+    uint32_t *__restrict__ err)
+{
+    __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_F1 + 66 * 4];
+    uint32_t *hist_l = (uint32_t *)(lds + BLOCK * SLOT_F1);
+    if (threadIdx.x < 66)
+        hist_l[threadIdx.x] = 0;
+    __syncthreads();
+    uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
+    if (g < n_groups) {
+        uint32_t gg = perm[g];
+        uint64_t j = gs[gg], jend = gs[gg + 1];
+        br_meta mt;
+        mt.s = L[j].s;
+        mt.e = L[jend - 1].e;
+        mt.seg = L[j].seg;
+        mt.d = (uint8_t)d;
+        int8_t pl = lcp[mt.s], pr = lcp[mt.e];
+        mt.P = pl > pr ? pl : pr;
+        int nmem = (int)(jend - j);
+        uint64_t nibs = 0;
+        int payload = 1 + (16 - nmem);
+        bool order_ok = nmem >= 2 && nmem <= 3;
+        bool stored = false;
+        {
+            int prev = -1;
+            for (uint64_t mm = j; mm < jend && order_ok; ++mm) {
+                uint8_t pb = L[mm].pad_;
+                int nbm = pb & 0xF;
+                order_ok &= nbm > prev;
+                prev = nbm;
+                nibs |= (uint64_t)nbm << (4 * (mm - j));
+                payload += L[mm].ref_len;
+                stored |= (pb & 0x20) != 0;
+            }
+        }
+        mt.flags = stored ? 1 : 0;
+        node_rec *r = &out[gg];
+        if (!order_ok || payload > 116) {
+            atomicOr(err, 1u << E_INTERNAL);
+            r->s = mt.s;
+            r->e = mt.e;
+            r->seg = mt.seg;
+            r->pad_ = 0;
+            r->depth = -1;
+            r->ref_len = 0;
+        } else {
+            int h = rlp_list_hdr_len(payload);
+            mt.br_len = (uint16_t)(h + payload);
+            uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_F1;
+            uint64_t *slot64 = (uint64_t *)slot;
+            byte_appender ap;
+            ap.init(slot64, 0 /* direct row */, 0);
+            if (payload < 56) {
+                ap.put((uint8_t)(0xc0 + payload));
+            } else {
+                ap.put(0xf8);
+                ap.put((uint8_t)payload);
+            }
+            {
+                uint64_t m = j;
+                for (int b = 0; b < 16; ++b) {
+                    if (m < jend &&
+                        (int)((nibs >> (4 * (m - j))) & 0xf) == b) {
+                        int rl = L[m].ref_len;
+                        const uint32_t *rec32 = (const uint32_t *)&L[m];
+                        uint32_t w[9];
+#pragma unroll
+                        for (int k = 0; k < 9; ++k)
+                            w[k] = rec32[3 + k];
+                        uint64_t R[5];
+                        R[0] = ((uint64_t)w[0] >> 16) | ((uint64_t)w[1] << 16) |
+                               ((uint64_t)w[2] << 48);
+                        R[1] = ((uint64_t)w[2] >> 16) | ((uint64_t)w[3] << 16) |
+                               ((uint64_t)w[4] << 48);
+                        R[2] = ((uint64_t)w[4] >> 16) | ((uint64_t)w[5] << 16) |
+                               ((uint64_t)w[6] << 48);
+                        R[3] = ((uint64_t)w[6] >> 16) | ((uint64_t)w[7] << 16) |
+                               ((uint64_t)w[8] << 48);
+                        R[4] = ((uint64_t)w[8] >> 16) & 0xFF;
+                        ap.put_bytes33(R, rl);
+                        m++;
+                    } else {
+                        ap.put(0x80);
+                    }
+                }
+                ap.put(0x80); // empty value item
+            }
+            ap.put(0x01); // keccak pad start
+            ap.finish(1);
+            uint64_t s[25];
+#pragma unroll
+            for (int i = 0; i < 17; ++i)
+                s[i] = slot64[i];
+#pragma unroll
+            for (int i = 17; i < 25; ++i)
+                s[i] = 0;
+            keccak_f(s);
+            uint64_t br_hash[4] = {s[0], s[1], s[2], s[3]};
+            // ext scratch aliases the message slot: branch_tail reads the
+            // inline words before any wrap reuses it
+            branch_tail(mt, subtree, keys, key_stride, br_hash, slot64, 1,
+                        slot, r, seg_roots, child_refs, child_lens, hist_l,
+                        nullptr, nullptr, 0xFFFFFFFFu);
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x < 66 && hist_l[threadIdx.x])
+        atomicAdd(&pending[threadIdx.x], hist_l[threadIdx.x]);
+}
 
 // TrieUpdates emission (updates mode): one row per STORED branch
 // (hash_mask != 0 — semantics in sre.h, pinned by the reference tests).
@@ -2704,17 +2902,22 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
     if (maxd < 0)
         return 0; // every leaf was already a segment root
 
-    // Per-depth carry arrays: a branch node produced with parent depth p
-    // waits in carries[p] (sorted by interval start) until level p runs.
-    // This replaces partitioning/rewriting one big carry at every level —
-    // each node now enters exactly one carry once.
-    std::vector<DBuf> carries;
-    carries.reserve(64);
-    for (int i = 0; i < 64; ++i)
-        carries.emplace_back(ctx);
-    uint64_t carry_cnt[64] = {0};
+    // Per-depth carry RUN LISTS: a branch node produced with parent depth p
+    // waits, in place inside its level's depth-bucketed output buffer, as
+    // part of a sorted run; level p consumes all its runs with ONE k-way
+    // positioning merge. This replaces the eager re-merge of an
+    // accumulated carry (each arrival used to re-read/rewrite the whole
+    // carry); now every node is positioned exactly once. The output
+    // buffers stay alive (pool-backed) until the pass ends.
+    struct run_ref {
+        const node_rec *p;
+        uint64_t n;
+    };
+    std::vector<run_ref> druns[64];
+    uint64_t drun_total[64] = {0};
+    std::vector<std::unique_ptr<DBuf>> live;
 
-    DBuf Lbuf(ctx), newn(ctx), newp(ctx), cmerge(ctx);
+    DBuf Lbuf(ctx), newn(ctx);
     DBuf flags(ctx), gidx(ctx), pend(ctx);
     DBuf gs(ctx), scratch(ctx), meta(ctx), urows(ctx), urow_cnt(ctx),
         urowidx(ctx);
@@ -2755,7 +2958,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 
     for (int d = maxd; d >= 0; --d) {
         uint64_t nA = hist_host[d + 1];
-        uint64_t nB = carry_cnt[d];
+        uint64_t nB = drun_total[d];
         if (nA == 0 && nB == 0)
             continue;
         po->levels++;
@@ -2763,20 +2966,66 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 
         // 1. this level's fresh leaves: the depth-(d+1) slice of dsorted
         node_rec *Lslice = dsorted.as<node_rec>() + doff[d + 1];
-        // 2. merge leaves with this depth's carry into the level input L
+        // 2. one k-way positioning merge of the fresh slice + carried runs
         node_rec *L;
         if (nB == 0) {
             L = Lslice;
-        } else if (nA == 0) {
-            L = carries[d].as<node_rec>();
+        } else if (nA == 0 && druns[d].size() == 1) {
+            L = (node_rec *)druns[d][0].p;
         } else {
+            kway_desc kd{};
+            int nr = 0;
+            if (nA) {
+                kd.run[nr] = Lslice;
+                kd.cnt[nr] = nA;
+                nr++;
+            }
+            for (const run_ref &rr : druns[d]) {
+                if (nr < KWAY_MAX) {
+                    kd.run[nr] = rr.p;
+                    kd.cnt[nr] = rr.n;
+                    nr++;
+                } else {
+                    // overflow run (deep tries only): pairwise-merge the
+                    // two smallest resident runs to make room
+                    int a = 0, b = 1;
+                    if (kd.cnt[b] < kd.cnt[a])
+                        std::swap(a, b);
+                    for (int k = 2; k < nr; ++k) {
+                        if (kd.cnt[k] < kd.cnt[a]) {
+                            b = a;
+                            a = k;
+                        } else if (kd.cnt[k] < kd.cnt[b]) {
+                            b = k;
+                        }
+                    }
+                    uint64_t tot = kd.cnt[a] + kd.cnt[b];
+                    auto mb = std::make_unique<DBuf>(ctx);
+                    HIP_CHECK(ctx, mb->alloc(tot * sizeof(node_rec)));
+                    hipLaunchKernelGGL(k_merge_a, dim3(grid_for(kd.cnt[a])),
+                                       dim3(BLOCK), 0, ctx->stream, kd.run[a],
+                                       kd.cnt[a], kd.run[b], kd.cnt[b],
+                                       mb->as<node_rec>());
+                    hipLaunchKernelGGL(k_merge_b, dim3(grid_for(kd.cnt[b])),
+                                       dim3(BLOCK), 0, ctx->stream, kd.run[a],
+                                       kd.cnt[a], kd.run[b], kd.cnt[b],
+                                       mb->as<node_rec>());
+                    HIP_CHECK(ctx, hipGetLastError());
+                    kd.run[a] = mb->as<node_rec>();
+                    kd.cnt[a] = tot;
+                    live.push_back(std::move(mb));
+                    kd.run[b] = rr.p;
+                    kd.cnt[b] = rr.n;
+                }
+            }
+            kd.nruns = nr;
+            kd.acc[0] = 0;
+            for (int k = 0; k < nr; ++k)
+                kd.acc[k + 1] = kd.acc[k] + kd.cnt[k];
             HIP_CHECK(ctx, Lbuf.alloc(n_level * sizeof(node_rec)));
-            hipLaunchKernelGGL(k_merge_a, dim3(grid_for(nA)), dim3(BLOCK), 0,
-                               ctx->stream, Lslice, nA,
-                               carries[d].as<node_rec>(), nB, Lbuf.as<node_rec>());
-            hipLaunchKernelGGL(k_merge_b, dim3(grid_for(nB)), dim3(BLOCK), 0,
-                               ctx->stream, Lslice, nA,
-                               carries[d].as<node_rec>(), nB, Lbuf.as<node_rec>());
+            hipLaunchKernelGGL(k_merge_kway, dim3(grid_for(n_level)),
+                               dim3(BLOCK), 0, ctx->stream, kd, n_level,
+                               Lbuf.as<node_rec>());
             HIP_CHECK(ctx, hipGetLastError());
             L = Lbuf.as<node_rec>();
         }
@@ -2826,6 +3075,12 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         DBuf perm(ctx), ccnt(ctx), coff(ctx), scratch2(ctx), meta2(ctx);
         const uint32_t CLS_MIN = 1u << 14; // below this the win is noise
         bool use_cls = n_groups >= CLS_MIN;
+        // fused 1-block kernel handles the class-0 slice of each chunk on
+        // the main stream (classes 1-3 keep the split assemble/hash
+        // pipeline); meta/scratch are produced inside the fused kernel, so
+        // updates/proof modes (which read them across kernels) disable it
+        bool use_fused = use_cls && n_pt == 0 && updates_kind < 0;
+        std::vector<uint32_t> c0s;
         DBuf cinv(ctx);
         if (use_cls && n_pt)
             HIP_CHECK(ctx, cinv.alloc((uint64_t)n_groups * 4));
@@ -2852,6 +3107,12 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                    perm.as<uint32_t>() + g0,
                                    n_pt ? cinv.as<uint32_t>() + g0 : nullptr);
                 HIP_CHECK(ctx, hipGetLastError());
+                if (use_fused) { // class-0 count = start offset of class 1
+                    uint32_t c0 = 0;
+                    HIP_CHECK(ctx, hipMemcpy(&c0, coff.as<uint32_t>() + nblk,
+                                             4, hipMemcpyDeviceToHost));
+                    c0s.push_back(c0);
+                }
             }
         }
         bool pipe2 = n_groups > chunk; // >1 chunk: overlap pays for 2nd buf
@@ -2879,16 +3140,34 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                           : scratch.as<uint8_t>();
             br_meta *mt = (pipe2 && buf) ? meta2.as<br_meta>()
                                          : meta.as<br_meta>();
-            uint32_t *d_perm = use_cls ? perm.as<uint32_t>() + g0 : nullptr;
+            // class-0 slice -> fused kernel (main stream); classes 1-3 ->
+            // split assemble (stream2) / hash (main), overlapped
+            uint32_t c0 = use_fused ? c0s[chunk_i] : 0;
+            uint32_t gsplit = gc - c0;
+            uint32_t *d_perm = use_cls ? perm.as<uint32_t>() + g0 + c0
+                                       : nullptr;
             hipStream_t s_asm = pipe2 ? ctx->stream2 : ctx->stream;
             if (pipe2) // wait until the hash consuming this buffer finished
                 hipStreamWaitEvent(s_asm, ev_hash[buf], 0);
-            hipLaunchKernelGGL(k_branch_assemble,
-                               dim3((gc + BLOCK_A - 1) / BLOCK_A), dim3(BLOCK_A),
-                               0, s_asm, L, gs.as<uint32_t>() + g0, gc,
-                               d_lcp, d_keys, key_stride, d, scr,
-                               chunk, mt, d_perm, d_err);
+            if (gsplit)
+                hipLaunchKernelGGL(k_branch_assemble,
+                                   dim3((gsplit + BLOCK_A - 1) / BLOCK_A),
+                                   dim3(BLOCK_A),
+                                   0, s_asm, L, gs.as<uint32_t>() + g0, gsplit,
+                                   d_lcp, d_keys, key_stride, d, scr,
+                                   chunk, mt, d_perm, d_err);
             HIP_CHECK(ctx, hipGetLastError());
+            if (c0) {
+                hipLaunchKernelGGL(k_branch_fused1, dim3(grid_for(c0)),
+                                   dim3(BLOCK), 0, ctx->stream, L,
+                                   gs.as<uint32_t>() + g0, c0, d_lcp, d_keys,
+                                   key_stride, d, subtree,
+                                   newn.as<node_rec>() + g0,
+                                   perm.as<uint32_t>() + g0, d_seg_roots,
+                                   d_child_refs, d_child_lens,
+                                   pend.as<uint32_t>(), d_err);
+                HIP_CHECK(ctx, hipGetLastError());
+            }
             if (pipe2) {
                 hipEventRecord(ev_asm[buf], s_asm);
                 hipStreamWaitEvent(ctx->stream, ev_asm[buf], 0);
@@ -2916,17 +3195,22 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                    urowidx.as<uint32_t>());
                 HIP_CHECK(ctx, hipGetLastError());
             }
-            hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(gc)), dim3(BLOCK), 0,
-                               ctx->stream, scr, chunk, mt,
-                               gc, d_keys, key_stride, subtree,
-                               newn.as<node_rec>() + g0, d_perm,
-                               d_seg_roots, d_child_refs,
-                               d_child_lens, pend.as<uint32_t>(), d_err, d_bhash,
-                               updates_kind >= 0 ? urows.as<sre_update_row>()
-                                                 : nullptr,
-                               updates_kind >= 0 ? urowidx.as<uint32_t>()
-                                                 : nullptr);
-            HIP_CHECK(ctx, hipGetLastError());
+            if (gsplit) {
+                hipLaunchKernelGGL(k_branch_hash, dim3(grid_for(gsplit)),
+                                   dim3(BLOCK), 0,
+                                   ctx->stream, scr, chunk, mt,
+                                   gsplit, d_keys, key_stride, subtree,
+                                   newn.as<node_rec>() + g0, d_perm,
+                                   d_seg_roots, d_child_refs,
+                                   d_child_lens, pend.as<uint32_t>(), d_err,
+                                   d_bhash,
+                                   updates_kind >= 0
+                                       ? urows.as<sre_update_row>()
+                                       : nullptr,
+                                   updates_kind >= 0 ? urowidx.as<uint32_t>()
+                                                     : nullptr);
+                HIP_CHECK(ctx, hipGetLastError());
+            }
             if (pipe2)
                 hipEventRecord(ev_hash[buf], ctx->stream);
             if (updates_kind >= 0) {
@@ -2954,9 +3238,10 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         po->branch_ms += ms;
         po->branch_count += n_groups;
 
-        // this level's inputs are consumed
-        carries[d].release();
-        carry_cnt[d] = 0;
+        // this level's inputs are consumed (slices stay in their live
+        // buffers; freed when the pass ends)
+        druns[d].clear();
+        drun_total[d] = 0;
 
         // 5. distribute the new nodes into their target per-depth carries.
         // pend[p+1] - previous snapshot = nodes newly pending at depth p.
@@ -2986,7 +3271,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                                              : 0);
         }
         if (any_fresh) {
-            HIP_CHECK(ctx, newp.alloc((uint64_t)n_groups * sizeof(node_rec)));
+            auto nb = std::make_unique<DBuf>(ctx);
+            HIP_CHECK(ctx, nb->alloc((uint64_t)n_groups * sizeof(node_rec)));
             uint32_t nblk = (uint32_t)((n_groups + BLOCK - 1) / BLOCK);
             DBuf dc2(ctx), do2(ctx);
             HIP_CHECK(ctx, dc2.alloc((uint64_t)66 * nblk * 4));
@@ -3001,42 +3287,23 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                 return -1;
             hipLaunchKernelGGL(k_depth_scatter66_rec, dim3(nblk), dim3(BLOCK),
                                0, ctx->stream, newn.as<node_rec>(), n_groups,
-                               nblk, do2.as<uint32_t>(), newp.as<node_rec>());
+                               nblk, do2.as<uint32_t>(), nb->as<node_rec>());
             HIP_CHECK(ctx, hipGetLastError());
-        }
-        for (int p = d - 1; p >= 0; --p) {
-            uint64_t fresh = fresh_cnt[p];
-            if (fresh == 0)
-                continue;
-            node_rec *slice = newp.as<node_rec>() + slice_off[p + 1];
-            if (carry_cnt[p] == 0) {
-                HIP_CHECK(ctx, carries[p].alloc(fresh * sizeof(node_rec)));
-                HIP_CHECK(ctx, hipMemcpyAsync(carries[p].p, slice,
-                                              fresh * sizeof(node_rec),
-                                              hipMemcpyDeviceToDevice,
-                                              ctx->stream));
-                carry_cnt[p] = fresh;
-            } else {
-                uint64_t total = carry_cnt[p] + fresh;
-                HIP_CHECK(ctx, cmerge.alloc(total * sizeof(node_rec)));
-                hipLaunchKernelGGL(k_merge_a, dim3(grid_for(carry_cnt[p])),
-                                   dim3(BLOCK), 0, ctx->stream,
-                                   carries[p].as<node_rec>(), carry_cnt[p],
-                                   slice, fresh, cmerge.as<node_rec>());
-                hipLaunchKernelGGL(k_merge_b, dim3(grid_for(fresh)), dim3(BLOCK),
-                                   0, ctx->stream, carries[p].as<node_rec>(),
-                                   carry_cnt[p], slice, fresh,
-                                   cmerge.as<node_rec>());
-                HIP_CHECK(ctx, hipGetLastError());
-                swap_bufs(carries[p], cmerge);
-                carry_cnt[p] = total;
-            }
+            // record the per-depth slices as carried runs, in place
+            node_rec *basep = nb->as<node_rec>();
+            for (int p = 0; p < d; ++p)
+                if (fresh_cnt[p]) {
+                    druns[p].push_back({basep + slice_off[p + 1],
+                                        fresh_cnt[p]});
+                    drun_total[p] += fresh_cnt[p];
+                }
+            live.push_back(std::move(nb));
         }
     }
     hipEventDestroy(ev0);
     hipEventDestroy(ev1);
     for (int p = 0; p < 64; ++p)
-        if (carry_cnt[p] != 0) {
+        if (drun_total[p] != 0) {
             set_err(ctx, "internal: carry not empty after level 0");
             return -1;
         }
