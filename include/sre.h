@@ -85,7 +85,18 @@ typedef struct sre_ctx sre_ctx;
  *   - tree_mask bit c is set iff child c's subtree top branch is itself
  *     stored;
  *   - the path-[] row (root is a branch) carries root_hash.
- * kind 0 = account trie; kind 1 = storage trie of acct_key. 624 bytes. */
+ * kind 0 = account trie; kind 1 = storage trie of acct_key. 624 bytes.
+ *
+ * `removed` (incremental mode only; always 0 from full rebuilds):
+ *   0 = upsert (the row is stored at this path now);
+ *   1 = removal — the path was stored before the delta and no longer is
+ *       (reth TrieUpdates.removed_nodes / StorageTrieUpdates.removed_nodes,
+ *       recorded by the walker at crates/trie/trie/src/walker.rs:363-369;
+ *       masks/hashes fields are zero);
+ *   2 = whole-storage-trie deletion marker for a destroyed account
+ *       (kind 1, path_len 0) — reth's StorageTrieUpdates::set_deleted(true)
+ *       for prefix_sets.destroyed_accounts
+ *       (crates/trie/common/src/updates.rs:154-157). */
 typedef struct {
     uint8_t  acct_key[32];   /* kind 1 only */
     uint8_t  kind;
@@ -96,7 +107,8 @@ typedef struct {
     uint16_t state_mask, tree_mask, hash_mask;
     uint8_t  root_hash[32];
     uint8_t  hashes[16][32];
-    uint8_t  pad_[6];
+    uint8_t  removed;        /* see above */
+    uint8_t  pad_[5];
 } sre_update_row;
 
 /* Per-call timing/throughput stats (HIP-event measured, for bench reporting;
@@ -200,6 +212,36 @@ int sre_incremental_root(sre_ctx *ctx,
                          uint64_t n_acct,
                          const sre_storage_entry *st_delta,
                          uint64_t n_st, uint8_t out_root[32]);
+
+/* Incremental root WITH TrieUpdates — the engine's equivalent of
+ * StateRoot::root_with_updates driven by a prefix set, i.e. the Merkle
+ * stage's incremental regime (crates/stages/stages/src/stages/merkle.rs:
+ * 319-342 -> incremental_root_with_updates, crates/trie/db/src/state.rs:62),
+ * including removed_nodes (walker.rs:363-369) and destroyed-account
+ * storage-trie deletion (updates.rs:140,154-157).
+ *
+ * Arm with sre_root_retaining_with_updates: computes the full root,
+ * retains cell-tops AND the full stored-row set (readable via
+ * sre_updates_count/get, all rows removed=0). Each following
+ * sre_incremental_root_with_updates applies the delta, recomputes dirty
+ * paths, and exposes the NET row diff vs the pre-delta trie:
+ * upserts (removed=0, new or changed rows), removals (removed=1), and
+ * destroyed-account markers (removed=2). Applying the diff to the
+ * pre-delta row set yields exactly the full-rebuild row set of the
+ * post-delta state — the same database effect as reth's
+ * finalize + into_sorted stream (which may additionally re-write
+ * unchanged rows on dirty paths; the engine suppresses those).
+ * Rows are ordered like TrieUpdatesSorted (account rows path-sorted
+ * first, upserts and removals interleaved; then storage rows grouped by
+ * account). Deltas chain; a plain sre_incremental_root or any state
+ * replacement disarms the row retention (rearm with
+ * sre_root_retaining_with_updates). */
+int sre_root_retaining_with_updates(sre_ctx *ctx, uint8_t out_root[32]);
+int sre_incremental_root_with_updates(sre_ctx *ctx,
+                                      const sre_account_delta *acct_delta,
+                                      uint64_t n_acct,
+                                      const sre_storage_entry *st_delta,
+                                      uint64_t n_st, uint8_t out_root[32]);
 
 /* Account multiproof — the surface of Proof::account_proof /
  * Proof::multiproof restricted to account targets

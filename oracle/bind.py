@@ -43,7 +43,8 @@ UPDATE_DTYPE = np.dtype([
     ("hash_mask", "<u2"),
     ("root_hash", np.uint8, 32),
     ("hashes", np.uint8, (16, 32)),
-    ("pad", np.uint8, 6),
+    ("removed", np.uint8),  # oracle always emits 0 (full rebuilds)
+    ("pad", np.uint8, 5),
 ])  # 624 bytes, matches sre_update_row
 assert UPDATE_DTYPE.itemsize == 624
 

@@ -1794,7 +1794,7 @@ __global__ void __launch_bounds__(BLOCK) k_emit_updates(
         row->acct_key[q] = 0;
     memcpy(row->pad_, &mt.seg, 4);
     row->pad_[4] = 0;
-    row->pad_[5] = 0;
+    row->removed = 0;
 }
 
 // ---------------------------------------------------------------------------
@@ -2265,11 +2265,14 @@ __global__ void k_ovl_scatter_st(const sre_storage_entry *__restrict__ base,
 #define N_CELLS (1u << (4 * CELL_NIBBLES))
 
 struct cap_row {
-    node_rec rec;    // interval, parent depth, ref — as of capture
-    uint32_t cell;   // 5-nibble prefix of the cell's keys
+    node_rec rec;     // interval, parent depth, ref — as of capture
+    uint32_t cell;    // 5-nibble prefix of the cell's keys
     uint32_t pad_[3];
+    uint8_t bhash[32]; // updates mode: bhash_by_s[rec.s] at capture time
+                       // (the keccak of the subtree-top BRANCH below any
+                       // extension — what ancestor TrieUpdates rows embed)
 };
-static_assert(sizeof(cap_row) == 64, "cap_row must be 64 bytes");
+static_assert(sizeof(cap_row) == 96, "cap_row must be 96 bytes");
 
 __device__ __forceinline__ uint32_t cell_of_key(const uint8_t *key)
 {
@@ -2284,7 +2287,9 @@ __global__ void k_capture_L(const node_rec *__restrict__ L, uint64_t n,
                             const uint8_t *__restrict__ keys,
                             uint64_t key_stride, cap_row *__restrict__ out,
                             uint32_t *__restrict__ cnt, uint64_t capacity,
-                            uint32_t *__restrict__ err)
+                            uint32_t *__restrict__ err,
+                            const uint8_t *__restrict__ bhash_by_s /* null
+                            unless updates mode */)
 {
     __shared__ uint32_t lds[BLOCK];
     __shared__ uint32_t base;
@@ -2306,6 +2311,10 @@ __global__ void k_capture_L(const node_rec *__restrict__ L, uint64_t n,
     copy_rec(&r.rec, &L[j]);
     r.cell = cell_of_key(keys + (uint64_t)L[j].s * key_stride);
     r.pad_[0] = r.pad_[1] = r.pad_[2] = 0;
+    if (bhash_by_s)
+        memcpy(r.bhash, bhash_by_s + 32ull * L[j].s, 32);
+    else
+        memset(r.bhash, 0, 32);
     out[slot] = r;
 }
 
@@ -2326,7 +2335,9 @@ __global__ void k_revalidate_rows(const cap_row *__restrict__ rows, uint64_t n,
                                   node_rec *__restrict__ recs,
                                   uint8_t *__restrict__ depths,
                                   uint8_t *__restrict__ covered,
-                                  uint32_t *__restrict__ hist)
+                                  uint32_t *__restrict__ hist,
+                                  uint8_t *__restrict__ bhash_out /* updates
+                                  mode: seed bhash_by_s at the new pos */)
 {
     __shared__ uint32_t hist_l[66];
     if (threadIdx.x < 66)
@@ -2350,6 +2361,8 @@ __global__ void k_revalidate_rows(const cap_row *__restrict__ rows, uint64_t n,
                 atomicAdd(&hist_l[nd + 1], 1u);
                 for (uint32_t p = ns_; p < ne_; ++p)
                     covered[p] = 1;
+                if (bhash_out)
+                    memcpy(bhash_out + 32ull * ns_, r.bhash, 32);
             } else {
                 cell_dirty[r.cell] = 1; // junction moved: rebuild the cell
             }
@@ -2512,6 +2525,18 @@ __global__ void k_scan_add(uint32_t *__restrict__ out, uint64_t n,
         }                                                                        \
     } while (0)
 
+// One stored row of the CURRENT trie, as (sort key, 64-bit content hash):
+// the retained row-set snapshot that lets the incremental mode emit a NET
+// TrieUpdates diff (upserts/removals) per delta. Content equality by h64
+// (FNV-1a over masks+hashes+root fields; collision odds are ~1e-19 per
+// pair — far below the GPU's own soft-error rate).
+struct snap_row {
+    uint8_t kind, path_len;
+    uint8_t acct_key[32];
+    uint8_t path[32];
+    uint64_t h64;
+};
+
 struct sre_ctx {
     int device = 0;
     hipStream_t stream = nullptr;
@@ -2531,6 +2556,10 @@ struct sre_ctx {
     // TrieUpdates retention (sre_root_with_updates)
     bool retain_updates = false;
     std::vector<sre_update_row> updates;
+    // Incremental TrieUpdates: current stored-row set, sorted like
+    // TrieUpdatesSorted (armed by sre_root_retaining_with_updates)
+    std::vector<snap_row> snap;
+    bool snap_valid = false;
     // Dirty-path incremental retention (sre_root_retaining /
     // sre_incremental_root): "cell-top" node records — the unique active
     // node of each 5-nibble key-prefix cell, captured from the level inputs
@@ -2707,6 +2736,7 @@ extern "C" int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entrie
     if (check_cap(ctx, n))
         return -1;
     ctx->cells_valid = false;
+    ctx->snap_valid = false;
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     release_acct(ctx);
     void *p = nullptr;
@@ -2727,6 +2757,7 @@ extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries
     if (check_cap(ctx, n))
         return -1;
     ctx->cells_valid = false;
+    ctx->snap_valid = false;
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     release_st(ctx);
     void *p = nullptr;
@@ -2747,6 +2778,7 @@ extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint
     if (check_cap(ctx, n))
         return -1;
     ctx->cells_valid = false;
+    ctx->snap_valid = false;
     release_acct(ctx);
     ctx->d_acct = (const sre_account_entry *)d_entries;
     ctx->na = n;
@@ -2759,6 +2791,7 @@ extern "C" int sre_set_storage_device(sre_ctx *ctx, const void *d_entries, uint6
     if (check_cap(ctx, n))
         return -1;
     ctx->cells_valid = false;
+    ctx->snap_valid = false;
     release_st(ctx);
     ctx->d_st = (const sre_storage_entry *)d_entries;
     ctx->ns = n;
@@ -2917,7 +2950,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
     uint64_t drun_total[64] = {0};
     std::vector<std::unique_ptr<DBuf>> live;
 
-    DBuf Lbuf(ctx), newn(ctx);
+    DBuf Lbuf(ctx), Cbuf(ctx), newn(ctx);
     DBuf flags(ctx), gidx(ctx), pend(ctx);
     DBuf gs(ctx), scratch(ctx), meta(ctx), urows(ctx), urow_cnt(ctx),
         urowidx(ctx);
@@ -2966,7 +2999,9 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 
         // 1. this level's fresh leaves: the depth-(d+1) slice of dsorted
         node_rec *Lslice = dsorted.as<node_rec>() + doff[d + 1];
-        // 2. one k-way positioning merge of the fresh slice + carried runs
+        // 2. merge carried runs (lazily, ONCE, k-way — they are small
+        // relative to the fresh slice), then 2-way with the fresh slice so
+        // the bulk elements do exactly one positioning search
         node_rec *L;
         if (nB == 0) {
             L = Lslice;
@@ -2975,11 +3010,6 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         } else {
             kway_desc kd{};
             int nr = 0;
-            if (nA) {
-                kd.run[nr] = Lslice;
-                kd.cnt[nr] = nA;
-                nr++;
-            }
             for (const run_ref &rr : druns[d]) {
                 if (nr < KWAY_MAX) {
                     kd.run[nr] = rr.p;
@@ -3018,21 +3048,40 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                     kd.cnt[b] = rr.n;
                 }
             }
-            kd.nruns = nr;
-            kd.acc[0] = 0;
-            for (int k = 0; k < nr; ++k)
-                kd.acc[k + 1] = kd.acc[k] + kd.cnt[k];
-            HIP_CHECK(ctx, Lbuf.alloc(n_level * sizeof(node_rec)));
-            hipLaunchKernelGGL(k_merge_kway, dim3(grid_for(n_level)),
-                               dim3(BLOCK), 0, ctx->stream, kd, n_level,
-                               Lbuf.as<node_rec>());
-            HIP_CHECK(ctx, hipGetLastError());
-            L = Lbuf.as<node_rec>();
+            const node_rec *carry;
+            if (nr == 1) {
+                carry = kd.run[0];
+            } else {
+                kd.nruns = nr;
+                kd.acc[0] = 0;
+                for (int k = 0; k < nr; ++k)
+                    kd.acc[k + 1] = kd.acc[k] + kd.cnt[k];
+                HIP_CHECK(ctx, Cbuf.alloc(nB * sizeof(node_rec)));
+                hipLaunchKernelGGL(k_merge_kway, dim3(grid_for(nB)),
+                                   dim3(BLOCK), 0, ctx->stream, kd, nB,
+                                   Cbuf.as<node_rec>());
+                HIP_CHECK(ctx, hipGetLastError());
+                carry = Cbuf.as<node_rec>();
+            }
+            if (nA == 0) {
+                L = (node_rec *)carry;
+            } else {
+                HIP_CHECK(ctx, Lbuf.alloc(n_level * sizeof(node_rec)));
+                hipLaunchKernelGGL(k_merge_a, dim3(grid_for(nA)), dim3(BLOCK),
+                                   0, ctx->stream, Lslice, nA, carry, nB,
+                                   Lbuf.as<node_rec>());
+                hipLaunchKernelGGL(k_merge_b, dim3(grid_for(nB)), dim3(BLOCK),
+                                   0, ctx->stream, Lslice, nA, carry, nB,
+                                   Lbuf.as<node_rec>());
+                HIP_CHECK(ctx, hipGetLastError());
+                L = Lbuf.as<node_rec>();
+            }
         }
         if (capture_depth > 0 && d == capture_depth - 1) {
             hipLaunchKernelGGL(k_capture_L, dim3(grid_for(n_level)), dim3(BLOCK),
                                0, ctx->stream, L, n_level, d_keys, key_stride,
-                               d_cap, d_cap_cnt, cap_capacity, d_err);
+                               d_cap, d_cap_cnt, cap_capacity, d_err,
+                               updates_kind >= 0 ? d_bhash : nullptr);
             HIP_CHECK(ctx, hipGetLastError());
         }
         // 3. group flags + scan
@@ -3411,20 +3460,36 @@ static int run_storage_pass(sre_ctx *ctx, uint8_t *d_acct_roots, pass_out *po,
         return -1;
     if (ctx->retain_updates && ctx->updates.size() > upd_start) {
         // patch acct_key from the stashed seg ids: seg -> account index ->
-        // 32-byte hashed key (strided D2H of the key column)
+        // 32-byte hashed key
         std::vector<uint32_t> seg_acct_h(n_seg);
         HIP_CHECK(ctx, hipMemcpy(seg_acct_h.data(), seg_acct.p,
                                  (uint64_t)n_seg * 4, hipMemcpyDeviceToHost));
-        std::vector<uint8_t> keys_h((uint64_t)na * 32);
-        HIP_CHECK(ctx, hipMemcpy2D(keys_h.data(), 32, ctx->d_acct,
-                                   sizeof(sre_account_entry), 32, na,
-                                   hipMemcpyDeviceToHost));
+        std::vector<uint8_t> keys_h;
+        if (!d_st_in || (uint64_t)n_seg * 8 > na) {
+            // full pass (or dense subset): strided D2H of the key column
+            keys_h.resize((uint64_t)na * 32);
+            HIP_CHECK(ctx, hipMemcpy2D(keys_h.data(), 32, ctx->d_acct,
+                                       sizeof(sre_account_entry), 32, na,
+                                       hipMemcpyDeviceToHost));
+        } else {
+            // incremental subset: fetch only the touched accounts' keys
+            // instead of all na x 32 B
+            keys_h.resize((uint64_t)n_seg * 32);
+            for (uint32_t s2 = 0; s2 < n_seg; ++s2)
+                HIP_CHECK(ctx, hipMemcpy(keys_h.data() + 32ull * s2,
+                                         (const uint8_t *)&ctx->d_acct
+                                             [seg_acct_h[s2]],
+                                         32, hipMemcpyDeviceToHost));
+        }
+        bool subset = d_st_in && (uint64_t)n_seg * 8 <= na;
         for (size_t r = upd_start; r < ctx->updates.size(); ++r) {
             sre_update_row &row = ctx->updates[r];
             uint32_t seg;
             memcpy(&seg, row.pad_, 4);
             memset(row.pad_, 0, sizeof(row.pad_));
-            memcpy(row.acct_key, keys_h.data() + 32ull * seg_acct_h[seg], 32);
+            memcpy(row.acct_key,
+                   keys_h.data() + 32ull * (subset ? seg : seg_acct_h[seg]),
+                   32);
         }
     }
 
@@ -4454,6 +4519,57 @@ static bool row_less(const sre_update_row &a, const sre_update_row &b)
     return a.path_len < b.path_len;
 }
 
+// FNV-1a over a row's content (masks + hashes + root fields): the snapshot
+// compares rows by this to suppress unchanged re-emits from the net diff.
+static uint64_t row_h64(const sre_update_row &r)
+{
+    uint64_t h = 1469598103934665603ull;
+    auto mix = [&](const void *p, size_t n) {
+        const uint8_t *b = (const uint8_t *)p;
+        for (size_t i = 0; i < n; ++i) {
+            h ^= b[i];
+            h *= 1099511628211ull;
+        }
+    };
+    mix(&r.state_mask, 2);
+    mix(&r.tree_mask, 2);
+    mix(&r.hash_mask, 2);
+    mix(&r.num_hashes, 1);
+    mix(&r.root_hash_set, 1);
+    mix(r.root_hash, 32);
+    mix(&r.hashes[0][0], 32ull * r.num_hashes);
+    return h;
+}
+
+static snap_row snap_of(const sre_update_row &r)
+{
+    snap_row s{};
+    s.kind = r.kind;
+    s.path_len = r.path_len;
+    memcpy(s.acct_key, r.acct_key, 32);
+    memcpy(s.path, r.path, 32);
+    s.h64 = row_h64(r);
+    return s;
+}
+
+// 3-way key compare, the same total order as row_less: paths are packed
+// high-nibble-first and zero-padded, so a 32-byte memcmp + length
+// tiebreak IS the nibble-lexicographic prefix-first order.
+static int snap_cmp(const snap_row &a, const snap_row &b)
+{
+    if (a.kind != b.kind)
+        return a.kind < b.kind ? -1 : 1;
+    if (a.kind == 1) {
+        int c = memcmp(a.acct_key, b.acct_key, 32);
+        if (c)
+            return c;
+    }
+    int c = memcmp(a.path, b.path, 32);
+    if (c)
+        return c;
+    return (int)a.path_len - (int)b.path_len;
+}
+
 static int apply_delta_impl(sre_ctx *ctx, const sre_account_delta *acct_delta,
                             uint64_t n_acct, const sre_storage_entry *st_delta,
                             uint64_t n_st, DBuf *map_out /* optional: old->new
@@ -4610,7 +4726,8 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
                                uint64_t n_acct,
                                const sre_storage_entry *st_delta, uint64_t n_st)
 {
-    ctx->cells_valid = false; // a plain apply invalidates cell retention
+    ctx->cells_valid = false;
+    ctx->snap_valid = false; // a plain apply invalidates cell retention
     return apply_delta_impl(ctx, acct_delta, n_acct, st_delta, n_st, nullptr);
 }
 
@@ -4634,6 +4751,7 @@ extern "C" int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32])
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     memset(&ctx->stats, 0, sizeof(ctx->stats));
     ctx->cells_valid = false;
+    ctx->snap_valid = false;
     uint64_t want_cap = 2 * (ctx->na < (uint64_t)N_CELLS ? ctx->na
                                                          : (uint64_t)N_CELLS) +
                         8192;
@@ -4689,11 +4807,16 @@ extern "C" int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32])
 // reusing every clean cell-top record and every untouched account's
 // retained storage root (the §3b walker-skip semantics expressed in this
 // engine's level machinery). Requires a prior sre_root_retaining.
-extern "C" int sre_incremental_root(sre_ctx *ctx,
-                                    const sre_account_delta *acct_delta,
-                                    uint64_t n_acct,
-                                    const sre_storage_entry *st_delta,
-                                    uint64_t n_st, uint8_t out_root[32])
+// with_updates: ctx->retain_updates is armed by the caller; additionally
+// seeds bhash from the captured rows and copies the final dirty-cell
+// bitmap out for the host-side net diff.
+static int incremental_root_impl(sre_ctx *ctx,
+                                 const sre_account_delta *acct_delta,
+                                 uint64_t n_acct,
+                                 const sre_storage_entry *st_delta,
+                                 uint64_t n_st, uint8_t out_root[32],
+                                 bool with_updates,
+                                 std::vector<uint8_t> *bitmap_out)
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     if (!ctx->cells_valid) {
@@ -4746,6 +4869,7 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     uint64_t na = ctx->na;
     if (na == 0) {
         ctx->cells_valid = false;
+    ctx->snap_valid = false;
         memcpy(out_root, EMPTY_ROOT_H, 32);
         return 0;
     }
@@ -4849,6 +4973,9 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     HIP_CHECK(ctx, hipMemsetAsync(depths.p, 0xFF, na, ctx->stream));
     HIP_CHECK(ctx, hipMemsetAsync(covered.p, 0, na, ctx->stream));
     HIP_CHECK(ctx, hipMemsetAsync(hist.p, 0, 66 * 4, ctx->stream));
+    DBuf abhash(ctx);
+    if (with_updates)
+        HIP_CHECK(ctx, abhash.alloc(na * 32));
     // seed clean cell-tops; anything invalid dirties its cell
     if (ctx->cap_count)
         hipLaunchKernelGGL(k_revalidate_rows, dim3(grid_for(ctx->cap_count)),
@@ -4857,7 +4984,8 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
                            map.as<uint32_t>(), lcp.as<int8_t>(),
                            bitmap.as<uint8_t>(), recs.as<node_rec>(),
                            depths.as<uint8_t>(), covered.as<uint8_t>(),
-                           hist.as<uint32_t>());
+                           hist.as<uint32_t>(),
+                           with_updates ? abhash.as<uint8_t>() : nullptr);
     HIP_CHECK(ctx, hipGetLastError());
     // rehash leaves of dirty cells and of positions no seed covers, with
     // the carried+patched storage roots
@@ -4882,6 +5010,11 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     hipEventDestroy(ev1);
     if (check_err(ctx, err.as<uint32_t>()))
         return -1;
+    if (bitmap_out) { // final dirty-cell set (incl. revalidation misses)
+        bitmap_out->resize(N_CELLS);
+        HIP_CHECK(ctx, hipMemcpy(bitmap_out->data(), bitmap.p, N_CELLS,
+                                 hipMemcpyDeviceToHost));
+    }
     uint32_t hist_host[66];
     HIP_CHECK(ctx, hipMemcpy(hist_host, hist.p, 66 * 4, hipMemcpyDeviceToHost));
     // fresh capture for the next delta; the retained rows were consumed by
@@ -4901,12 +5034,18 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     HIP_CHECK(ctx, hipMemsetAsync(capcnt.p, 0, 4, ctx->stream));
     const uint8_t *keys =
         (const uint8_t *)ctx->d_acct + offsetof(sre_account_entry, key);
+    size_t upd_start = ctx->updates.size();
     if (run_levels(ctx, na, recs.as<node_rec>(), depths.as<uint8_t>(),
                    lcp.as<int8_t>(), keys, sizeof(sre_account_entry), hist_host,
                    0, roots.as<uint8_t>(), nullptr, nullptr, err.as<uint32_t>(),
-                   &po, -1, nullptr, CELL_NIBBLES, (cap_row *)ctx->d_cap_rows,
+                   &po, with_updates ? 0 : -1,
+                   with_updates ? abhash.as<uint8_t>() : nullptr,
+                   CELL_NIBBLES, (cap_row *)ctx->d_cap_rows,
                    capcnt.as<uint32_t>(), ctx->cap_capacity))
         return -1;
+    if (with_updates) // account rows carry no acct_key; clear the seg stash
+        for (size_t r = upd_start; r < ctx->updates.size(); ++r)
+            memset(ctx->updates[r].pad_, 0, sizeof(ctx->updates[r].pad_));
     if (check_err(ctx, err.as<uint32_t>()))
         return -1;
     uint32_t cnt = 0;
@@ -4932,6 +5071,167 @@ extern "C" int sre_incremental_root(sre_ctx *ctx,
     ctx->stats.levels = po.levels;
 
     HIP_CHECK(ctx, hipMemcpy(out_root, roots.p, 32, hipMemcpyDeviceToHost));
+    return 0;
+}
+
+extern "C" int sre_incremental_root(sre_ctx *ctx,
+                                    const sre_account_delta *acct_delta,
+                                    uint64_t n_acct,
+                                    const sre_storage_entry *st_delta,
+                                    uint64_t n_st, uint8_t out_root[32])
+{
+    // the capture refresh runs without branch hashes -> the row snapshot
+    // can no longer seed future update emission
+    ctx->snap_valid = false;
+    return incremental_root_impl(ctx, acct_delta, n_acct, st_delta, n_st,
+                                 out_root, false, nullptr);
+}
+
+extern "C" int sre_root_retaining_with_updates(sre_ctx *ctx,
+                                               uint8_t out_root[32])
+{
+    ctx->retain_updates = true;
+    ctx->updates.clear();
+    int rc = sre_root_retaining(ctx, out_root);
+    ctx->retain_updates = false;
+    if (rc)
+        return rc;
+    std::sort(ctx->updates.begin(), ctx->updates.end(), row_less);
+    ctx->snap.clear();
+    ctx->snap.reserve(ctx->updates.size());
+    for (auto &r : ctx->updates) {
+        r.removed = 0;
+        ctx->snap.push_back(snap_of(r));
+    }
+    ctx->snap_valid = true;
+    return 0;
+}
+
+extern "C" int sre_incremental_root_with_updates(
+    sre_ctx *ctx, const sre_account_delta *acct_delta, uint64_t n_acct,
+    const sre_storage_entry *st_delta, uint64_t n_st, uint8_t out_root[32])
+{
+    if (!ctx->snap_valid) {
+        set_err(ctx, "sre_incremental_root_with_updates: needs "
+                     "sre_root_retaining_with_updates first");
+        return -1;
+    }
+    // region keys from the (sorted) delta, host side
+    std::vector<std::array<uint8_t, 32>> touched, destroyed;
+    for (uint64_t i = 0; i < n_st; ++i)
+        if (i == 0 ||
+            memcmp(st_delta[i].acct_key, st_delta[i - 1].acct_key, 32)) {
+            std::array<uint8_t, 32> k;
+            memcpy(k.data(), st_delta[i].acct_key, 32);
+            touched.push_back(k);
+        }
+    for (uint64_t i = 0; i < n_acct; ++i)
+        if (acct_delta[i].deleted) {
+            std::array<uint8_t, 32> k;
+            memcpy(k.data(), acct_delta[i].key, 32);
+            destroyed.push_back(k);
+        }
+
+    ctx->retain_updates = true;
+    ctx->updates.clear();
+    std::vector<uint8_t> bitmap;
+    int rc = incremental_root_impl(ctx, acct_delta, n_acct, st_delta, n_st,
+                                   out_root, true, &bitmap);
+    ctx->retain_updates = false;
+    if (rc) {
+        ctx->snap_valid = false;
+        return rc;
+    }
+
+    // NET diff: emitted rows vs the retained snapshot. The engine
+    // re-emits every row in the rebuilt region (depth <= 4 account rows,
+    // dirty/uncovered cells, rebuilt storage tries); unchanged re-emits
+    // are suppressed by content hash, region rows missing from the
+    // emission become removals (walker.rs:363-369 semantics), destroyed
+    // accounts get whole-trie markers (updates.rs:154-157).
+    std::vector<sre_update_row> E;
+    E.swap(ctx->updates);
+    std::sort(E.begin(), E.end(), row_less);
+    std::vector<snap_row> Es;
+    Es.reserve(E.size());
+    for (auto &r : E) {
+        r.removed = 0;
+        Es.push_back(snap_of(r));
+    }
+
+    auto key_in = [](const std::vector<std::array<uint8_t, 32>> &v,
+                     const uint8_t *k) {
+        auto it = std::lower_bound(
+            v.begin(), v.end(), k,
+            [](const std::array<uint8_t, 32> &a, const uint8_t *b) {
+                return memcmp(a.data(), b, 32) < 0;
+            });
+        return it != v.end() && memcmp(it->data(), k, 32) == 0;
+    };
+    auto in_region = [&](const snap_row &s) {
+        if (s.kind == 0) {
+            if (s.path_len <= 4)
+                return true;
+            uint32_t cell = ((uint32_t)s.path[0] << 12) |
+                            ((uint32_t)s.path[1] << 4) |
+                            ((uint32_t)s.path[2] >> 4);
+            return bitmap[cell] != 0;
+        }
+        return key_in(touched, s.acct_key);
+    };
+
+    std::vector<sre_update_row> diff;
+    std::vector<snap_row> nsnap;
+    nsnap.reserve(ctx->snap.size() + Es.size());
+    size_t i = 0, j = 0;
+    while (i < ctx->snap.size() || j < Es.size()) {
+        int c;
+        if (i >= ctx->snap.size())
+            c = 1;
+        else if (j >= Es.size())
+            c = -1;
+        else
+            c = snap_cmp(ctx->snap[i], Es[j]);
+        if (c == 0) {
+            if (ctx->snap[i].h64 != Es[j].h64)
+                diff.push_back(E[j]); // changed row
+            nsnap.push_back(Es[j]);
+            i++;
+            j++;
+        } else if (c > 0) { // new row
+            diff.push_back(E[j]);
+            nsnap.push_back(Es[j]);
+            j++;
+        } else { // snapshot row not re-emitted this pass
+            const snap_row &s = ctx->snap[i];
+            if (s.kind == 1 && key_in(destroyed, s.acct_key)) {
+                // dropped wholesale; the removed=2 marker covers the trie
+            } else if (in_region(s)) {
+                sre_update_row rr{};
+                rr.kind = s.kind;
+                rr.path_len = s.path_len;
+                memcpy(rr.path, s.path, 32);
+                if (s.kind == 1)
+                    memcpy(rr.acct_key, s.acct_key, 32);
+                rr.removed = 1;
+                diff.push_back(rr);
+            } else {
+                nsnap.push_back(s); // untouched region: row kept
+            }
+            i++;
+        }
+    }
+    for (const auto &k : destroyed) { // StorageTrieUpdates::set_deleted(true)
+        sre_update_row rr{};
+        rr.kind = 1;
+        memcpy(rr.acct_key, k.data(), 32);
+        rr.removed = 2;
+        diff.push_back(rr);
+    }
+    std::sort(diff.begin(), diff.end(), row_less);
+    ctx->snap.swap(nsnap);
+    ctx->updates.swap(diff);
+    ctx->snap_valid = true;
     return 0;
 }
 

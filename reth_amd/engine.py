@@ -58,7 +58,10 @@ UPDATE_DTYPE = np.dtype([
     ("hash_mask", "<u2"),
     ("root_hash", np.uint8, 32),
     ("hashes", np.uint8, (16, 32)),
-    ("pad", np.uint8, 6),
+    # incremental net-diff marker: 0 upsert, 1 removed path,
+    # 2 whole-storage-trie deletion (destroyed account) — include/sre.h
+    ("removed", np.uint8),
+    ("pad", np.uint8, 5),
 ])
 assert UPDATE_DTYPE.itemsize == 624
 
@@ -216,6 +219,43 @@ class StateRootEngine:
         self._check(self._lib.sre_root(ctypes.c_void_p(self._ctx), out))
         return bytes(out)
 
+    def _fetch_updates(self):
+        n = self._lib.sre_updates_count(ctypes.c_void_p(self._ctx))
+        rows = np.zeros(n, dtype=UPDATE_DTYPE)
+        if n:
+            self._check(self._lib.sre_updates_get(
+                ctypes.c_void_p(self._ctx), _np_ptr(rows), n))
+        return rows
+
+    def root_retaining_with_updates(self):
+        """Full root + full TrieUpdates row set, AND arms both the cell-top
+        retention and the stored-row snapshot so following
+        incremental_root_with_updates calls emit net row diffs
+        (include/sre.h sre_root_retaining_with_updates)."""
+        out = (ctypes.c_uint8 * 32)()
+        self._check(self._lib.sre_root_retaining_with_updates(
+            ctypes.c_void_p(self._ctx), out))
+        return bytes(out), self._fetch_updates()
+
+    def incremental_root_with_updates(self, acct_delta: np.ndarray,
+                                      st_delta: np.ndarray = None):
+        """Dirty-path incremental root + the NET TrieUpdates diff vs the
+        pre-delta trie: upserts (removed=0), removals (removed=1,
+        walker.rs:363-369 removed_nodes semantics) and destroyed-account
+        whole-storage-trie markers (removed=2, updates.rs:154-157).
+        Applying the diff to the pre-delta row set reproduces the full
+        row set of the post-delta state. Requires a prior
+        root_retaining_with_updates; deltas chain."""
+        assert acct_delta.dtype == DELTA_DTYPE
+        if st_delta is None:
+            st_delta = np.zeros(0, dtype=STORAGE_DTYPE)
+        assert st_delta.dtype == STORAGE_DTYPE
+        out = (ctypes.c_uint8 * 32)()
+        self._check(self._lib.sre_incremental_root_with_updates(
+            ctypes.c_void_p(self._ctx), _np_ptr(acct_delta),
+            len(acct_delta), _np_ptr(st_delta), len(st_delta), out))
+        return bytes(out), self._fetch_updates()
+
     def root_with_updates(self):
         """State root + TrieUpdates rows (UPDATE_DTYPE), sorted like reth's
         TrieUpdates::into_sorted. Surface of
@@ -223,12 +263,7 @@ class StateRootEngine:
         out = (ctypes.c_uint8 * 32)()
         self._check(self._lib.sre_root_with_updates(
             ctypes.c_void_p(self._ctx), out))
-        n = self._lib.sre_updates_count(ctypes.c_void_p(self._ctx))
-        rows = np.zeros(n, dtype=UPDATE_DTYPE)
-        if n:
-            self._check(self._lib.sre_updates_get(
-                ctypes.c_void_p(self._ctx), _np_ptr(rows), n))
-        return bytes(out), rows
+        return bytes(out), self._fetch_updates()
 
     def account_proof(self, targets) -> list:
         """Account multiproof: per target, the proof node RLPs root-first
