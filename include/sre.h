@@ -243,6 +243,29 @@ int sre_incremental_root_with_updates(sre_ctx *ctx,
                                       const sre_storage_entry *st_delta,
                                       uint64_t n_st, uint8_t out_root[32]);
 
+/* state_root_from_nodes equivalent — the remaining StateRootProvider
+ * methods (state_root_from_nodes{,_with_updates} over TrieInput,
+ * crates/storage/storage-api/src/trie.rs:26-40; TrieInput
+ * crates/trie/common/src/input.rs:10): compute the root of
+ * (resident state + delta) using `rows` — a stored-node overlay in the
+ * engine's own sre_update_row format, e.g. the output of
+ * sre_root_with_updates — in place of recomputing every storage trie.
+ * kind-1 path-[] rows' root_hash short-circuits untouched accounts'
+ * storage roots exactly like reth's walker skip over stored roots
+ * (walker.rs:195-230); tries without a usable row or touched by the
+ * delta are rebuilt from their entries; the account trie is rebuilt
+ * on-device (it is ~2% of the leaf work — the account-node skip that
+ * kind-0 rows enable on a CPU walk saves nothing here, so those rows
+ * are accepted and ignored). Removal rows (removed != 0) are invalid
+ * input. The resident state is REPLACED by the merged result. For the
+ * _with_updates form, call sre_root_with_updates afterwards or use the
+ * incremental surface above. */
+int sre_root_from_nodes(sre_ctx *ctx,
+                        const sre_update_row *rows, uint64_t n_rows,
+                        const sre_account_delta *acct_delta, uint64_t n_acct,
+                        const sre_storage_entry *st_delta, uint64_t n_st,
+                        uint8_t out_root[32]);
+
 /* Account multiproof — the surface of Proof::account_proof /
  * Proof::multiproof restricted to account targets
  * (crates/trie/trie/src/proof/mod.rs:59-137 `multiproof`, collecting the

@@ -1235,23 +1235,42 @@ __global__ void k_class_hist(const uint32_t *__restrict__ gs, uint32_t n_groups,
         cnts[threadIdx.x * nblk + blockIdx.x] = c_l[threadIdx.x];
 }
 
-// scatter: perm[offs[c*nblk + b] + rank] = g (rank via LDS bump; stability
-// is unnecessary — records are written back at their original group index)
+// scatter: perm[offs[c*nblk + b] + rank] = g. STABLE (ballot ranks +
+// per-wave LDS counts): group order within each class is preserved, so
+// the assemble/fused kernels' member gathers stay coalesced — the old
+// atomic-bump scatter randomized the order and left the fused kernel
+// gather-latency bound.
 __global__ void k_class_scatter(const uint32_t *__restrict__ gs,
                                 uint32_t n_groups, uint32_t nblk,
                                 const uint32_t *__restrict__ offs,
                                 uint32_t *__restrict__ perm,
                                 uint32_t *__restrict__ inv /* nullable */)
 {
-    __shared__ uint32_t base_l[4];
+    __shared__ uint32_t wh[4][CLS_BLOCK / 64];
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
     if (threadIdx.x < 4)
-        base_l[threadIdx.x] = offs[threadIdx.x * nblk + blockIdx.x];
+#pragma unroll
+        for (int w = 0; w < CLS_BLOCK / 64; ++w)
+            wh[threadIdx.x][w] = 0;
     __syncthreads();
     uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
-    if (g >= n_groups)
+    int c = (g < n_groups) ? nmem_class(gs[g + 1] - gs[g]) : 255;
+    uint32_t rank_in_wave = 0;
+    for (int k = 0; k < 4; ++k) {
+        uint64_t m = __ballot(c == k);
+        if (c == k)
+            rank_in_wave = __popcll(m & ((1ull << lane) - 1));
+        if (lane == 0 && m)
+            wh[k][wid] = (uint32_t)__popcll(m);
+    }
+    __syncthreads();
+    if (c > 3)
         return;
-    int c = nmem_class(gs[g + 1] - gs[g]);
-    uint32_t slot = atomicAdd(&base_l[c], 1u);
+    uint32_t before = 0;
+    for (int w = 0; w < wid; ++w)
+        before += wh[c][w];
+    uint32_t slot = offs[(uint32_t)c * nblk + blockIdx.x] + before +
+                    rank_in_wave;
     perm[slot] = g;
     if (inv)
         inv[g] = slot;
@@ -2371,6 +2390,38 @@ __global__ void k_revalidate_rows(const cap_row *__restrict__ rows, uint64_t n,
     __syncthreads();
     if (threadIdx.x < 66 && hist_l[threadIdx.x])
         atomicAdd(&hist[threadIdx.x], hist_l[threadIdx.x]);
+}
+
+// sre_root_from_nodes: per storage segment, either seed the account's
+// storage root from a supplied stored path-[] row (untouched trie) or
+// flag the segment for rebuild (no row, or touched by the delta).
+__global__ void k_seg_need(const sre_storage_entry *__restrict__ st,
+                           const uint32_t *__restrict__ seg_start,
+                           uint32_t n_seg,
+                           const uint8_t *__restrict__ root_kv, /* nr x 64:
+                           acct_key || root, key-sorted */
+                           uint64_t nr,
+                           const uint8_t *__restrict__ touched, /* nt x 32 */
+                           uint64_t nt,
+                           const uint32_t *__restrict__ seg_acct,
+                           uint8_t *__restrict__ acct_roots,
+                           uint32_t *__restrict__ need)
+{
+    uint32_t s = blockIdx.x * blockDim.x + threadIdx.x;
+    if (s >= n_seg)
+        return;
+    const uint8_t *key = st[seg_start[s]].acct_key;
+    uint64_t t = lb_keys(touched, 32, nt, key, 32);
+    if (!(t < nt && cmp_key32(touched + 32 * t, key) == 0)) {
+        uint64_t rp = lb_keys(root_kv, 64, nr, key, 32);
+        if (rp < nr && cmp_key32(root_kv + 64 * rp, key) == 0) {
+            memcpy(acct_roots + 32ull * seg_acct[s], root_kv + 64 * rp + 32,
+                   32);
+            need[s] = 0;
+            return;
+        }
+    }
+    need[s] = 1;
 }
 
 // dirty-cell marking from STORAGE delta rows (the account's leaf changes)
@@ -5232,6 +5283,201 @@ extern "C" int sre_incremental_root_with_updates(
     ctx->snap.swap(nsnap);
     ctx->updates.swap(diff);
     ctx->snap_valid = true;
+    return 0;
+}
+
+// state_root_from_nodes equivalent (StateRootProvider, crates/storage/
+// storage-api/src/trie.rs:26-40; TrieInput, crates/trie/common/src/
+// input.rs:10): compute the root of (resident state + HashedPostState
+// delta) using a supplied stored-node overlay instead of recomputing
+// every storage trie. The engine consumes exactly what reth's walker
+// would: kind-1 path-[] rows' root_hash seeds untouched accounts'
+// storage roots (the stored-root skip of walker.rs:195-230); tries
+// without a usable row (small tries whose root branch is unstored, or
+// tries touched by the delta) are rebuilt from their entries; the
+// account trie is rebuilt in full on-device (cheap: it is ~2% of the
+// leaf work). kind-0 rows are accepted and ignored — the account-trie
+// skip they enable on CPU saves nothing here. Removal rows are invalid
+// input. The resident state is REPLACED by the merged result.
+extern "C" int sre_root_from_nodes(sre_ctx *ctx,
+                                   const sre_update_row *rows,
+                                   uint64_t n_rows,
+                                   const sre_account_delta *acct_delta,
+                                   uint64_t n_acct,
+                                   const sre_storage_entry *st_delta,
+                                   uint64_t n_st, uint8_t out_root[32])
+{
+    HIP_CHECK(ctx, hipSetDevice(ctx->device));
+    memset(&ctx->stats, 0, sizeof(ctx->stats));
+    ctx->cells_valid = false;
+    ctx->snap_valid = false;
+    hipEvent_t t0, t1;
+    hipEventCreate(&t0);
+    hipEventCreate(&t1);
+    hipEventRecord(t0, ctx->stream);
+
+    // supplied storage roots: kind-1 path-[] rows with root_hash
+    std::vector<std::array<uint8_t, 64>> kv; // key || root
+    for (uint64_t i = 0; i < n_rows; ++i) {
+        if (rows[i].removed) {
+            set_err(ctx, "sre_root_from_nodes: removal rows are not a "
+                         "valid node overlay");
+            return -1;
+        }
+        if (rows[i].kind == 1 && rows[i].path_len == 0 &&
+            rows[i].root_hash_set) {
+            std::array<uint8_t, 64> e;
+            memcpy(e.data(), rows[i].acct_key, 32);
+            memcpy(e.data() + 32, rows[i].root_hash, 32);
+            kv.push_back(e);
+        }
+    }
+    std::sort(kv.begin(), kv.end(),
+              [](const std::array<uint8_t, 64> &a,
+                 const std::array<uint8_t, 64> &b) {
+                  return memcmp(a.data(), b.data(), 32) < 0;
+              });
+    std::vector<uint8_t> tkeys; // touched storage accounts (delta, sorted)
+    for (uint64_t i = 0; i < n_st; ++i)
+        if (i == 0 ||
+            memcmp(st_delta[i].acct_key, st_delta[i - 1].acct_key, 32))
+            tkeys.insert(tkeys.end(), st_delta[i].acct_key,
+                         st_delta[i].acct_key + 32);
+
+    if (apply_delta_impl(ctx, acct_delta, n_acct, st_delta, n_st, nullptr))
+        return -1;
+    uint64_t na = ctx->na, ns = ctx->ns;
+    if (na == 0) {
+        if (ns != 0) {
+            set_err(ctx, "storage entries without accounts");
+            return -1;
+        }
+        memcpy(out_root, EMPTY_ROOT_H, 32);
+        return 0;
+    }
+
+    DBuf err(ctx), acct_roots(ctx), roots(ctx);
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    HIP_CHECK(ctx, acct_roots.alloc(na * 32));
+    HIP_CHECK(ctx, roots.alloc(32));
+    hipLaunchKernelGGL(k_fill_empty_roots, dim3(grid_for(na)), dim3(BLOCK),
+                       0, ctx->stream, acct_roots.as<uint8_t>(), na);
+    HIP_CHECK(ctx, hipGetLastError());
+    pass_out po;
+    if (ns) {
+        // segment the resident storage; seed supplied roots / flag rebuilds
+        DBuf flags(ctx), seg_id(ctx), lcp(ctx);
+        HIP_CHECK(ctx, flags.alloc(ns * 4));
+        HIP_CHECK(ctx, seg_id.alloc(ns * 4));
+        HIP_CHECK(ctx, lcp.alloc(ns + 1));
+        hipLaunchKernelGGL(k_seg_flags_lcp, dim3(grid_for(ns + 1)),
+                           dim3(BLOCK), 0, ctx->stream, ctx->d_st, ns,
+                           flags.as<uint32_t>(), lcp.as<int8_t>(),
+                           err.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        uint32_t n_seg = 0;
+        if (scan_u32(ctx, flags.as<uint32_t>(), seg_id.as<uint32_t>(), ns,
+                     &n_seg))
+            return -1;
+        hipLaunchKernelGGL(k_seg_fix, dim3(grid_for(ns)), dim3(BLOCK), 0,
+                           ctx->stream, flags.as<uint32_t>(),
+                           seg_id.as<uint32_t>(), ns);
+        DBuf seg_start(ctx), seg_acct(ctx), need(ctx), dkv(ctx), dtk(ctx);
+        HIP_CHECK(ctx, seg_start.alloc((uint64_t)n_seg * 4));
+        HIP_CHECK(ctx, seg_acct.alloc((uint64_t)n_seg * 4));
+        HIP_CHECK(ctx, need.alloc((uint64_t)n_seg * 4));
+        hipLaunchKernelGGL(k_seg_starts, dim3(grid_for(ns)), dim3(BLOCK), 0,
+                           ctx->stream, flags.as<uint32_t>(),
+                           seg_id.as<uint32_t>(), ns,
+                           seg_start.as<uint32_t>());
+        hipLaunchKernelGGL(k_seg_acct, dim3(grid_for(n_seg)), dim3(BLOCK), 0,
+                           ctx->stream, ctx->d_st, seg_start.as<uint32_t>(),
+                           n_seg, ctx->d_acct, na, seg_acct.as<uint32_t>(),
+                           err.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        HIP_CHECK(ctx, dkv.alloc(kv.empty() ? 64 : kv.size() * 64));
+        if (!kv.empty())
+            HIP_CHECK(ctx, hipMemcpyAsync(dkv.p, kv.data(), kv.size() * 64,
+                                          hipMemcpyHostToDevice, ctx->stream));
+        HIP_CHECK(ctx, dtk.alloc(tkeys.empty() ? 32 : tkeys.size()));
+        if (!tkeys.empty())
+            HIP_CHECK(ctx, hipMemcpyAsync(dtk.p, tkeys.data(), tkeys.size(),
+                                          hipMemcpyHostToDevice, ctx->stream));
+        hipLaunchKernelGGL(k_seg_need, dim3(grid_for(n_seg)), dim3(BLOCK), 0,
+                           ctx->stream, ctx->d_st, seg_start.as<uint32_t>(),
+                           n_seg, dkv.as<uint8_t>(),
+                           (uint64_t)kv.size(), dtk.as<uint8_t>(),
+                           (uint64_t)(tkeys.size() / 32),
+                           seg_acct.as<uint32_t>(), acct_roots.as<uint8_t>(),
+                           need.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        std::vector<uint32_t> need_h(n_seg), start_h(n_seg);
+        HIP_CHECK(ctx, hipMemcpy(need_h.data(), need.p, 4ull * n_seg,
+                                 hipMemcpyDeviceToHost));
+        HIP_CHECK(ctx, hipMemcpy(start_h.data(), seg_start.p, 4ull * n_seg,
+                                 hipMemcpyDeviceToHost));
+        std::vector<uint32_t> lo, hi, offs;
+        uint64_t total = 0;
+        for (uint32_t s = 0; s < n_seg; ++s)
+            if (need_h[s]) {
+                uint32_t e = s + 1 < n_seg ? start_h[s + 1] : (uint32_t)ns;
+                lo.push_back(start_h[s]);
+                hi.push_back(e);
+                offs.push_back((uint32_t)total);
+                total += e - start_h[s];
+            }
+        offs.push_back((uint32_t)total);
+        if (total) {
+            uint32_t nneed = (uint32_t)lo.size();
+            DBuf dlo(ctx), dhi(ctx), doffs(ctx), compact(ctx);
+            HIP_CHECK(ctx, dlo.alloc(4ull * nneed));
+            HIP_CHECK(ctx, dhi.alloc(4ull * nneed));
+            HIP_CHECK(ctx, doffs.alloc(4ull * (nneed + 1)));
+            HIP_CHECK(ctx, hipMemcpyAsync(dlo.p, lo.data(), 4ull * nneed,
+                                          hipMemcpyHostToDevice, ctx->stream));
+            HIP_CHECK(ctx, hipMemcpyAsync(dhi.p, hi.data(), 4ull * nneed,
+                                          hipMemcpyHostToDevice, ctx->stream));
+            HIP_CHECK(ctx, hipMemcpyAsync(doffs.p, offs.data(),
+                                          4ull * (nneed + 1),
+                                          hipMemcpyHostToDevice, ctx->stream));
+            HIP_CHECK(ctx, compact.alloc(total * sizeof(sre_storage_entry)));
+            hipLaunchKernelGGL(k_gather_touched, dim3(grid_for(total)),
+                               dim3(BLOCK), 0, ctx->stream, ctx->d_st,
+                               dlo.as<uint32_t>(), dhi.as<uint32_t>(),
+                               doffs.as<uint32_t>(), nneed, total,
+                               compact.as<sre_storage_entry>());
+            HIP_CHECK(ctx, hipGetLastError());
+            if (run_storage_pass(ctx, acct_roots.as<uint8_t>(), &po,
+                                 err.as<uint32_t>(), nullptr, 0, nullptr,
+                                 nullptr, 0, nullptr,
+                                 compact.as<sre_storage_entry>(), total))
+                return -1;
+        }
+        if (check_err(ctx, err.as<uint32_t>()))
+            return -1;
+    }
+    if (run_account_pass(ctx, acct_roots.as<uint8_t>(), 0,
+                         roots.as<uint8_t>(), nullptr, nullptr, &po,
+                         err.as<uint32_t>()))
+        return -1;
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    hipEventRecord(t1, ctx->stream);
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    float total_ms = 0;
+    hipEventElapsedTime(&total_ms, t0, t1);
+    hipEventDestroy(t0);
+    hipEventDestroy(t1);
+    ctx->stats.total_ms = total_ms;
+    ctx->stats.leaf_hash_ms = po.leaf_ms;
+    ctx->stats.branch_hash_ms = po.branch_ms;
+    ctx->stats.leaf_count = po.leaf_count;
+    ctx->stats.leaf_blocks = po.leaf_blocks;
+    ctx->stats.branch_count = po.branch_count;
+    ctx->stats.branch_blocks = po.branch_blocks;
+    ctx->stats.levels = po.levels;
+    HIP_CHECK(ctx, hipMemcpy(out_root, roots.p, 32, hipMemcpyDeviceToHost));
     return 0;
 }
 

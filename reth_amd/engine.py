@@ -256,6 +256,28 @@ class StateRootEngine:
             len(acct_delta), _np_ptr(st_delta), len(st_delta), out))
         return bytes(out), self._fetch_updates()
 
+    def root_from_nodes(self, rows: np.ndarray, acct_delta: np.ndarray = None,
+                        st_delta: np.ndarray = None) -> bytes:
+        """state_root_from_nodes / TrieInput equivalent: root of
+        (resident state + delta) with `rows` (UPDATE_DTYPE, e.g. from
+        root_with_updates) seeding untouched accounts' storage roots so
+        their tries are not recomputed (include/sre.h
+        sre_root_from_nodes). Replaces the resident state with the
+        merged result."""
+        assert rows.dtype == UPDATE_DTYPE
+        if acct_delta is None:
+            acct_delta = np.zeros(0, dtype=DELTA_DTYPE)
+        if st_delta is None:
+            st_delta = np.zeros(0, dtype=STORAGE_DTYPE)
+        assert acct_delta.dtype == DELTA_DTYPE
+        assert st_delta.dtype == STORAGE_DTYPE
+        out = (ctypes.c_uint8 * 32)()
+        self._check(self._lib.sre_root_from_nodes(
+            ctypes.c_void_p(self._ctx), _np_ptr(rows), len(rows),
+            _np_ptr(acct_delta), len(acct_delta),
+            _np_ptr(st_delta), len(st_delta), out))
+        return bytes(out)
+
     def root_with_updates(self):
         """State root + TrieUpdates rows (UPDATE_DTYPE), sorted like reth's
         TrieUpdates::into_sorted. Surface of

@@ -197,7 +197,7 @@ def test_incremental_updates_ordering(eng):
     rows, strows = [], []
     for i in rng.choice(len(keys), 30, replace=False):
         k = keys[int(i)]
-        rows.append((k, 9, 9, ke, 0))
+        rows.append((k, 9, 9, accounts[k][2], 0))
         accounts[k][0] = 9
         accounts[k][1] = 9
         nk = bind.keccak256(b"ord" + k[:4])
