@@ -290,9 +290,11 @@ int sre_account_proof(sre_ctx *ctx,
 
 /* Storage multiproof — StorageProof::storage_multiproof
  * (crates/trie/trie/src/proof/mod.rs) for (acct_key, slot_key)
- * pairs (account must be present; the slot may be absent — exclusion
- * semantics as in sre_account_proof; a storage-less account yields
- * EMPTY_ROOT_HASH and an empty list): per target the account's storage
+ * pairs. The slot may be absent (exclusion semantics as in
+ * sre_account_proof); a storage-less OR ABSENT account yields
+ * EMPTY_ROOT_HASH and an empty node list — exactly
+ * StorageMultiProof::empty(), the reference's short-circuit on an empty
+ * storage cursor (proof/mod.rs storage_multiproof): per target the account's storage
  * root (out_roots, 32 B each) and the root-first node list of its storage
  * trie, same output layout as sre_account_proof. Storage tries CAN
  * contain inline (<32 B) nodes; those are embedded in their parents and

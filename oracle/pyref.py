@@ -291,11 +291,14 @@ def account_proof(accounts: dict, hashed_key: bytes):
 
 
 def storage_proof(accounts: dict, acct_key: bytes, slot_key: bytes):
-    """(storage_root, proof-node list root-first) for a slot of a present
-    account — StorageProof::storage_multiproof semantics. The slot may be
-    absent: _collect_proof then returns the lookup-path nodes ending at
-    the proven divergence (an exclusion proof), matching
-    engine.storage_proof / include/sre.h."""
+    """(storage_root, proof-node list root-first) for a slot —
+    StorageProof::storage_multiproof semantics. The slot may be absent:
+    _collect_proof then returns the lookup-path nodes ending at the
+    proven divergence (an exclusion proof). An absent or storage-less
+    account yields (EMPTY_ROOT_HASH, []) = StorageMultiProof::empty(),
+    matching engine.storage_proof / include/sre.h."""
+    if acct_key not in accounts:
+        return keccak256(b"\x80"), []
     slots = accounts[acct_key][3]
     items = sorted((tuple(nibbles_of(k)), rlp_int(v))
                    for k, v in slots.items() if v != 0)

@@ -4251,11 +4251,10 @@ extern "C" int sre_storage_proof(sre_ctx *ctx, const uint8_t *acct_keys,
                              hipMemcpyDeviceToHost));
     HIP_CHECK(ctx, hipMemcpy(presa.data(), dpra.p, 4ull * n_t,
                              hipMemcpyDeviceToHost));
-    for (uint32_t t = 0; t < n_t; ++t)
-        if (!presa[t]) {
-            set_err(ctx, "sre_storage_proof: account not present");
-            return -1;
-        }
+    // absent accounts are NOT an error: storage_multiproof returns
+    // StorageMultiProof::empty() for them (proof/mod.rs storage_multiproof
+    // short-circuits on an empty storage cursor), i.e. EMPTY_ROOT_HASH +
+    // an empty node list — same as a present storage-less account.
     DBuf dti2(ctx);
     HIP_CHECK(ctx, dti2.alloc(4ull * n_t));
     {
@@ -4301,6 +4300,12 @@ extern "C" int sre_storage_proof(sre_ctx *ctx, const uint8_t *acct_keys,
     for (uint32_t t = 0; t < n_t; ++t) {
         const uint8_t *key = slot_keys + 32ull * t;
         uint8_t sroot[32];
+        if (!presa[t]) {
+            // absent account: StorageMultiProof::empty()
+            memcpy(out_roots + 32ull * t, EMPTY_ROOT_H2, 32);
+            out_counts[t] = 0;
+            continue;
+        }
         HIP_CHECK(ctx, hipMemcpy(sroot,
                                  acct_roots.as<uint8_t>() + 32ull * tia[t], 32,
                                  hipMemcpyDeviceToHost));

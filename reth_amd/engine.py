@@ -315,10 +315,11 @@ class StateRootEngine:
         return out
 
     def storage_proof(self, acct_keys, slot_keys):
-        """Storage multiproof for (account, slot) pairs (account must be
-        present; the slot may be absent -> exclusion proof): returns
-        (roots, proofs) — per target the account's storage root and the
-        root-first node-RLP list of its storage trie."""
+        """Storage multiproof for (account, slot) pairs. The slot may be
+        absent (exclusion proof); an absent or storage-less account
+        yields (EMPTY_ROOT_HASH, []) = StorageMultiProof::empty().
+        Returns (roots, proofs) — per target the account's storage root
+        and the root-first node-RLP list of its storage trie."""
         n = len(acct_keys)
         assert len(slot_keys) == n
         aarr = np.ascontiguousarray(

@@ -203,6 +203,28 @@ def test_storage_proof_slotless_account(eng):
     assert proofs[0] == []
 
 
+def test_storage_proof_absent_account(eng):
+    # an ABSENT account behaves like a storage-less one:
+    # StorageMultiProof::empty() — EMPTY_ROOT_HASH + empty node list
+    # (proof/mod.rs storage_multiproof empty-cursor short circuit)
+    ke = bind.keccak256(b"")
+    a1 = bind.keccak256(b"present")
+    sk = bind.keccak256(b"s1")
+    accounts = {a1: (1, 2, ke, {sk: 5})}
+    acct, st = to_arrays(accounts)
+    eng.upload(acct, st)
+    ghost = bind.keccak256(b"ghost-account")
+    roots, proofs = eng.storage_proof([ghost, a1], [sk, sk])
+    want_root, want_nodes = pyref.storage_proof(accounts, ghost, sk)
+    assert roots[0] == want_root
+    assert roots[0] == bytes.fromhex(
+        "56e81f171bcc55a6ff8345e692c0f86e5b48e01b996cadc001622fb5e363b421")
+    assert proofs[0] == [] and want_nodes == []
+    # the present target in the same batch is unaffected
+    want_root1, want_nodes1 = pyref.storage_proof(accounts, a1, sk)
+    assert roots[1] == want_root1 and proofs[1] == want_nodes1
+
+
 def test_storage_proof_parity(eng):
     acct, st = gen.gen_state_numpy(500, 24, bind.keccak256_batch)
     accounts = _dict_of(acct, st)
