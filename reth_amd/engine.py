@@ -253,7 +253,8 @@ class StateRootEngine:
         out = (ctypes.c_uint8 * 32)()
         self._check(self._lib.sre_incremental_root_with_updates(
             ctypes.c_void_p(self._ctx), _np_ptr(acct_delta),
-            len(acct_delta), _np_ptr(st_delta), len(st_delta), out))
+            ctypes.c_uint64(len(acct_delta)), _np_ptr(st_delta),
+            ctypes.c_uint64(len(st_delta)), out))
         return bytes(out), self._fetch_updates()
 
     def root_from_nodes(self, rows: np.ndarray, acct_delta: np.ndarray = None,
@@ -272,10 +273,13 @@ class StateRootEngine:
         assert acct_delta.dtype == DELTA_DTYPE
         assert st_delta.dtype == STORAGE_DTYPE
         out = (ctypes.c_uint8 * 32)()
+        # 8 args: stack-passed lengths MUST be explicit c_uint64 (a bare
+        # int becomes a 32-bit stack slot with a garbage upper half)
         self._check(self._lib.sre_root_from_nodes(
-            ctypes.c_void_p(self._ctx), _np_ptr(rows), len(rows),
-            _np_ptr(acct_delta), len(acct_delta),
-            _np_ptr(st_delta), len(st_delta), out))
+            ctypes.c_void_p(self._ctx), _np_ptr(rows),
+            ctypes.c_uint64(len(rows)),
+            _np_ptr(acct_delta), ctypes.c_uint64(len(acct_delta)),
+            _np_ptr(st_delta), ctypes.c_uint64(len(st_delta)), out))
         return bytes(out)
 
     def root_with_updates(self):
