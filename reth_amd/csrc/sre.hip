@@ -731,7 +731,9 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
     uint32_t *__restrict__ hist, uint8_t *__restrict__ roots,
     uint8_t *__restrict__ child_refs, uint8_t *__restrict__ child_lens,
     const uint8_t *__restrict__ cell_dirty /* incremental: null = all */,
-    const uint8_t *__restrict__ covered /* incremental: seeded positions */)
+    const uint8_t *__restrict__ covered /* incremental: seeded positions */,
+    const uint32_t *__restrict__ pos_list /* sparse launch: recompute set */,
+    uint64_t n_list)
 {
     __shared__ __align__(16) uint8_t lds[BLOCK * SLOT_ACC + 66 * 4];
     uint32_t *hist_l = (uint32_t *)(lds + BLOCK * SLOT_ACC);
@@ -740,14 +742,21 @@ __global__ void __launch_bounds__(BLOCK) k_leaf_account(
     __syncthreads();
 
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    bool active = i < na;
-    if (active && cell_dirty) {
-        uint32_t cell = ((uint32_t)acct[i].key[0] << 12) |
-                        ((uint32_t)acct[i].key[1] << 4) |
-                        ((uint32_t)acct[i].key[2] >> 4);
-        // recompute leaves of dirty cells and of positions no seeded
-        // cell-top interval covers; clean covered cells keep their seeds
-        active = cell_dirty[cell] != 0 || covered[i] == 0;
+    bool active;
+    if (pos_list) { // sparse launch: the position list IS the recompute set
+        active = i < n_list;
+        if (active)
+            i = pos_list[i];
+    } else {
+        active = i < na;
+        if (active && cell_dirty) {
+            uint32_t cell = ((uint32_t)acct[i].key[0] << 12) |
+                            ((uint32_t)acct[i].key[1] << 4) |
+                            ((uint32_t)acct[i].key[2] >> 4);
+            // recompute leaves of dirty cells and of positions no seeded
+            // cell-top interval covers; clean covered cells keep their seeds
+            active = cell_dirty[cell] != 0 || covered[i] == 0;
+        }
     }
     if (active) {
         uint8_t *slot = lds + (uint64_t)threadIdx.x * SLOT_ACC;
@@ -2277,6 +2286,174 @@ __global__ void k_ovl_scatter_st(const sre_storage_entry *__restrict__ base,
 }
 
 // ---------------------------------------------------------------------------
+// small-delta account merge (dirty-path incremental, accounts-only):
+// ONE pass over the base replaces the flags + two scans + scatter + posmap
+// passes of the general overlay merge. All delta-side rank arrays are tiny
+// (L2-resident); per base row the ranks are closed-form:
+//   p      = lower_bound(delta, base[i].key)
+//   map[i] = i - m_excl[p] + eff_excl[p]
+// where m_excl counts MATCHED delta rows (they drop/replace the base row)
+// and eff_excl counts EFFECTIVE (non-deleted) delta rows (they produce an
+// output entry). Identical to the general bexcl/dexcl formula.
+// ---------------------------------------------------------------------------
+
+__global__ void k_sd_marks(const sre_account_entry *__restrict__ base,
+                           uint64_t nb,
+                           const sre_account_delta *__restrict__ dl,
+                           uint64_t nd, uint32_t *__restrict__ bpos,
+                           uint32_t *__restrict__ match)
+{
+    uint64_t k = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= nd)
+        return;
+    uint64_t p = lb_keys((const uint8_t *)base, sizeof(sre_account_entry), nb,
+                         dl[k].key, 32);
+    bpos[k] = (uint32_t)p;
+    match[k] = (p < nb && cmp_key32(base[p].key, dl[k].key) == 0) ? 1u : 0u;
+}
+
+// merge scatter: survivors copied to their new position; map written for
+// EVERY old index (+ sentinel at nb)
+__global__ void k_sd_scatter(const sre_account_entry *__restrict__ base,
+                             uint64_t nb,
+                             const sre_account_delta *__restrict__ dl,
+                             uint64_t nd,
+                             const uint32_t *__restrict__ m_excl, /* nd+1 */
+                             const uint32_t *__restrict__ eff_excl, /* nd+1 */
+                             sre_account_entry *__restrict__ out,
+                             uint32_t *__restrict__ map)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i > nb)
+        return;
+    if (i == nb) {
+        map[nb] = (uint32_t)(nb - m_excl[nd] + eff_excl[nd]);
+        return;
+    }
+    uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_account_delta), nd,
+                         base[i].key, 32);
+    bool m = p < nd && cmp_key32(dl[p].key, base[i].key) == 0;
+    uint32_t j = (uint32_t)(i - m_excl[p] + eff_excl[p]);
+    map[i] = j;
+    if (!m) {
+        // 13 u64 copies (entries are 8-B aligned at the 104-B stride)
+        const uint64_t *s8 = (const uint64_t *)&base[i];
+        uint64_t *d8 = (uint64_t *)&out[j];
+#pragma unroll
+        for (int w = 0; w < 13; ++w)
+            d8[w] = s8[w];
+    }
+}
+
+// place effective delta rows; newpos[k] = output position of delta row k
+// (for deleted rows: the junction position — the slot where its successor
+// now sits — recorded for the lcp boundary patch)
+__global__ void k_sd_place(const sre_account_delta *__restrict__ dl,
+                           uint64_t nd,
+                           const uint32_t *__restrict__ bpos,
+                           const uint32_t *__restrict__ m_excl,
+                           const uint32_t *__restrict__ eff_excl,
+                           sre_account_entry *__restrict__ out,
+                           uint32_t *__restrict__ newpos)
+{
+    uint64_t k = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= nd)
+        return;
+    uint32_t j = (uint32_t)(bpos[k] - m_excl[k] + eff_excl[k]);
+    newpos[k] = j;
+    if (!dl[k].deleted) {
+        sre_account_entry e;
+        memcpy(e.key, dl[k].key, 32);
+        e.nonce = dl[k].nonce;
+        memcpy(e.balance, dl[k].balance, 32);
+        memcpy(e.code_hash, dl[k].code_hash, 32);
+        out[j] = e;
+    }
+}
+
+// lcp repair: an adjacent old pair (i-1, i) that stays adjacent in the
+// merge keeps its lcp (keys of matched rows are unchanged — a modify
+// replaces the value, not the key). Every other new-pair lcp is a delta
+// boundary and is recomputed exactly by k_sd_lcp_patch.
+__global__ void k_sd_lcp_copy(const uint32_t *__restrict__ map, uint64_t nb,
+                              const int8_t *__restrict__ lcp_old,
+                              int8_t *__restrict__ lcp_new)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i < 1 || i >= nb)
+        return;
+    uint32_t j = map[i];
+    if (j == map[i - 1] + 1)
+        lcp_new[j] = lcp_old[i];
+}
+
+__global__ void k_sd_lcp_patch(const sre_account_entry *__restrict__ out,
+                               uint64_t new_nb,
+                               const uint32_t *__restrict__ patch, uint64_t np,
+                               int8_t *__restrict__ lcp_new,
+                               uint32_t *__restrict__ err)
+{
+    uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (t >= np)
+        return;
+    uint32_t j = patch[t];
+    if (j > new_nb)
+        return;
+    if (j == 0 || j == new_nb) {
+        lcp_new[j] = -1;
+        return;
+    }
+    bool gt;
+    int l = key_lcp(out[j - 1].key, out[j].key, &gt);
+    if (gt || l == 64)
+        atomicOr(err, 1u << E_UNSORTED_ACCT);
+    lcp_new[j] = (int8_t)l;
+}
+
+// sparse-leaf support: compact the positions no seed covers (dirty cells
+// leave covered == 0 too, so this is exactly the recompute set)
+__global__ void k_active_hist(const uint8_t *__restrict__ covered, uint64_t n,
+                              uint32_t *__restrict__ cnts)
+{
+    __shared__ uint32_t c_l;
+    if (threadIdx.x == 0)
+        c_l = 0;
+    __syncthreads();
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    uint64_t m = __ballot(i < n && covered[i] == 0);
+    if ((threadIdx.x & 63) == 0 && m)
+        atomicAdd(&c_l, (uint32_t)__popcll(m));
+    __syncthreads();
+    if (threadIdx.x == 0)
+        cnts[blockIdx.x] = c_l;
+}
+
+__global__ void k_active_scatter(const uint8_t *__restrict__ covered,
+                                 uint64_t n,
+                                 const uint32_t *__restrict__ offs,
+                                 uint32_t *__restrict__ list)
+{
+    __shared__ uint32_t wh[BLOCK / 64];
+    int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    if (threadIdx.x < BLOCK / 64)
+        wh[threadIdx.x] = 0;
+    __syncthreads();
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    bool act = i < n && covered[i] == 0;
+    uint64_t m = __ballot(act);
+    uint32_t rank = __popcll(m & ((1ull << lane) - 1));
+    if (lane == 0)
+        wh[wid] = (uint32_t)__popcll(m);
+    __syncthreads();
+    if (!act)
+        return;
+    uint32_t before = 0;
+    for (int w = 0; w < wid; ++w)
+        before += wh[w];
+    list[offs[blockIdx.x] + before + rank] = (uint32_t)i;
+}
+
+// ---------------------------------------------------------------------------
 // dirty-path incremental: cell-top capture (CELL_NIBBLES-prefix cells)
 // ---------------------------------------------------------------------------
 
@@ -2623,6 +2800,13 @@ struct sre_ctx {
     uint64_t roots_ret_capacity = 0; // in accounts
     void *d_roots_ret2 = nullptr; // ping-pong partner (incremental updates)
     uint64_t roots_ret2_capacity = 0;
+    // retained account-lcp (na+1 i8) for small-delta repair: chained
+    // deltas patch O(delta) boundary lcps instead of recomputing all na
+    void *d_lcp_ret = nullptr;
+    uint64_t lcp_ret_capacity = 0; // bytes
+    void *d_lcp_ret2 = nullptr;
+    uint64_t lcp_ret2_capacity = 0;
+    bool lcp_valid = false; // d_lcp_ret matches the resident accounts
     // size-class buffer pool: the level machinery allocates/frees dozens of
     // transient arrays per level; hipMalloc latency would dominate small
     // jobs. Freed buffers are cached by power-of-2 class and reused (also
@@ -2759,6 +2943,10 @@ extern "C" void sre_destroy(sre_ctx *ctx)
         (void)hipFree(ctx->d_roots_ret);
     if (ctx->d_roots_ret2)
         (void)hipFree(ctx->d_roots_ret2);
+    if (ctx->d_lcp_ret)
+        (void)hipFree(ctx->d_lcp_ret);
+    if (ctx->d_lcp_ret2)
+        (void)hipFree(ctx->d_lcp_ret2);
     for (auto &e : ctx->pool)
         (void)hipFree(e.second);
     ctx->pool.clear();
@@ -2788,6 +2976,7 @@ extern "C" int sre_upload_accounts(sre_ctx *ctx, const sre_account_entry *entrie
         return -1;
     ctx->cells_valid = false;
     ctx->snap_valid = false;
+    ctx->lcp_valid = false;
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     release_acct(ctx);
     void *p = nullptr;
@@ -2809,6 +2998,7 @@ extern "C" int sre_upload_storage(sre_ctx *ctx, const sre_storage_entry *entries
         return -1;
     ctx->cells_valid = false;
     ctx->snap_valid = false;
+    ctx->lcp_valid = false;
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     release_st(ctx);
     void *p = nullptr;
@@ -2830,6 +3020,7 @@ extern "C" int sre_set_accounts_device(sre_ctx *ctx, const void *d_entries, uint
         return -1;
     ctx->cells_valid = false;
     ctx->snap_valid = false;
+    ctx->lcp_valid = false;
     release_acct(ctx);
     ctx->d_acct = (const sre_account_entry *)d_entries;
     ctx->na = n;
@@ -2843,6 +3034,7 @@ extern "C" int sre_set_storage_device(sre_ctx *ctx, const void *d_entries, uint6
         return -1;
     ctx->cells_valid = false;
     ctx->snap_valid = false;
+    ctx->lcp_valid = false;
     release_st(ctx);
     ctx->d_st = (const sre_storage_entry *)d_entries;
     ctx->ns = n;
@@ -3586,7 +3778,7 @@ static int run_account_pass(sre_ctx *ctx, const uint8_t *d_storage_roots, int su
     hipLaunchKernelGGL(k_leaf_account, dim3(grid_for(na)), dim3(BLOCK), 0, ctx->stream,
                        ctx->d_acct, na, d_storage_roots, lcp.as<int8_t>(), subtree,
                        recs.as<node_rec>(), depths.as<uint8_t>(), hist.as<uint32_t>(),
-                       d_roots, d_child_refs, d_child_lens, nullptr, nullptr);
+                       d_roots, d_child_refs, d_child_lens, nullptr, nullptr, nullptr, 0);
     HIP_CHECK(ctx, hipGetLastError());
     hipEventRecord(ev1, ctx->stream);
     HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
@@ -4783,7 +4975,8 @@ extern "C" int sre_apply_delta(sre_ctx *ctx,
                                const sre_storage_entry *st_delta, uint64_t n_st)
 {
     ctx->cells_valid = false;
-    ctx->snap_valid = false; // a plain apply invalidates cell retention
+    ctx->snap_valid = false;
+    ctx->lcp_valid = false; // a plain apply invalidates cell retention
     return apply_delta_impl(ctx, acct_delta, n_acct, st_delta, n_st, nullptr);
 }
 
@@ -4802,12 +4995,136 @@ static int ensure_roots_ret(sre_ctx *ctx, uint64_t want_accounts)
     return 0;
 }
 
+static int ensure_lcp_ret(sre_ctx *ctx, void **slot, uint64_t *cap,
+                          uint64_t want_bytes)
+{
+    if (*cap >= want_bytes)
+        return 0;
+    if (*slot)
+        (void)hipFree(*slot);
+    *slot = nullptr;
+    *cap = 0;
+    HIP_CHECK(ctx, hipMalloc(slot, want_bytes));
+    *cap = want_bytes;
+    return 0;
+}
+
+// Small-delta account merge + lcp repair (accounts-only dirty path): one
+// pass over the base (closed-form ranks from tiny L2-resident delta
+// arrays) replaces the general overlay merge's flags + scans + posmap
+// passes, and the retained lcp is repaired at O(delta) boundary positions
+// instead of recomputed over all na. Preconditions: ns == 0, lcp_valid.
+static int apply_delta_small(sre_ctx *ctx, const sre_account_delta *acct_delta,
+                             uint64_t nd, DBuf *map_out)
+{
+    uint64_t nb = ctx->na;
+    DBuf dl(ctx), bpos(ctx), match(ctx), mex(ctx), eex(ctx), npos(ctx),
+        err(ctx);
+    HIP_CHECK(ctx, dl.alloc(nd * sizeof(sre_account_delta)));
+    HIP_CHECK(ctx, hipMemcpyAsync(dl.p, acct_delta,
+                                  nd * sizeof(sre_account_delta),
+                                  hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHECK(ctx, bpos.alloc(nd * 4));
+    HIP_CHECK(ctx, match.alloc(nd * 4));
+    hipLaunchKernelGGL(k_sd_marks, dim3(grid_for(nd)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_acct, nb,
+                       dl.as<sre_account_delta>(), nd, bpos.as<uint32_t>(),
+                       match.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    std::vector<uint32_t> bpos_h(nd), match_h(nd);
+    HIP_CHECK(ctx, hipMemcpy(bpos_h.data(), bpos.p, 4 * nd,
+                             hipMemcpyDeviceToHost));
+    HIP_CHECK(ctx, hipMemcpy(match_h.data(), match.p, 4 * nd,
+                             hipMemcpyDeviceToHost));
+    std::vector<uint32_t> mex_h(nd + 1), eex_h(nd + 1);
+    mex_h[0] = eex_h[0] = 0;
+    for (uint64_t k = 0; k < nd; ++k) {
+        mex_h[k + 1] = mex_h[k] + (match_h[k] ? 1 : 0);
+        eex_h[k + 1] = eex_h[k] + (acct_delta[k].deleted ? 0 : 1);
+    }
+    uint64_t new_na = nb - mex_h[nd] + eex_h[nd];
+    HIP_CHECK(ctx, mex.alloc(4 * (nd + 1)));
+    HIP_CHECK(ctx, eex.alloc(4 * (nd + 1)));
+    HIP_CHECK(ctx, hipMemcpyAsync(mex.p, mex_h.data(), 4 * (nd + 1),
+                                  hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHECK(ctx, hipMemcpyAsync(eex.p, eex_h.data(), 4 * (nd + 1),
+                                  hipMemcpyHostToDevice, ctx->stream));
+    size_t acct_bytes = (new_na ? new_na : 1) * sizeof(sre_account_entry);
+    void *out = pool_get(ctx, acct_bytes);
+    if (!out) {
+        set_err(ctx, "apply_delta_small: out of memory");
+        return -1;
+    }
+    HIP_CHECK(ctx, map_out->alloc((nb + 1) * 4));
+    hipLaunchKernelGGL(k_sd_scatter, dim3(grid_for(nb + 1)), dim3(BLOCK), 0,
+                       ctx->stream, ctx->d_acct, nb,
+                       dl.as<sre_account_delta>(), nd, mex.as<uint32_t>(),
+                       eex.as<uint32_t>(), (sre_account_entry *)out,
+                       map_out->as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    HIP_CHECK(ctx, npos.alloc(nd * 4));
+    hipLaunchKernelGGL(k_sd_place, dim3(grid_for(nd)), dim3(BLOCK), 0,
+                       ctx->stream, dl.as<sre_account_delta>(), nd,
+                       bpos.as<uint32_t>(), mex.as<uint32_t>(),
+                       eex.as<uint32_t>(), (sre_account_entry *)out,
+                       npos.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    // lcp repair into the ping-pong partner
+    if (ensure_lcp_ret(ctx, &ctx->d_lcp_ret2, &ctx->lcp_ret2_capacity,
+                       new_na + 1))
+        return -1;
+    hipLaunchKernelGGL(k_sd_lcp_copy, dim3(grid_for(nb)), dim3(BLOCK), 0,
+                       ctx->stream, map_out->as<uint32_t>(), nb,
+                       (const int8_t *)ctx->d_lcp_ret,
+                       (int8_t *)ctx->d_lcp_ret2);
+    HIP_CHECK(ctx, hipGetLastError());
+    std::vector<uint32_t> npos_h(nd);
+    HIP_CHECK(ctx, hipMemcpy(npos_h.data(), npos.p, 4 * nd,
+                             hipMemcpyDeviceToHost));
+    std::vector<uint32_t> patch;
+    patch.reserve(2 * nd + 2);
+    patch.push_back(0);
+    patch.push_back((uint32_t)new_na);
+    for (uint64_t k = 0; k < nd; ++k) {
+        patch.push_back(npos_h[k]); // insert/replace pos; delete junction
+        if (!acct_delta[k].deleted)
+            patch.push_back(npos_h[k] + 1);
+    }
+    std::sort(patch.begin(), patch.end());
+    patch.erase(std::unique(patch.begin(), patch.end()), patch.end());
+    DBuf dpatch(ctx);
+    HIP_CHECK(ctx, dpatch.alloc(4 * patch.size()));
+    HIP_CHECK(ctx, hipMemcpyAsync(dpatch.p, patch.data(), 4 * patch.size(),
+                                  hipMemcpyHostToDevice, ctx->stream));
+    HIP_CHECK(ctx, err.alloc(4));
+    HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
+    hipLaunchKernelGGL(k_sd_lcp_patch, dim3(grid_for(patch.size())),
+                       dim3(BLOCK), 0, ctx->stream,
+                       (const sre_account_entry *)out, new_na,
+                       dpatch.as<uint32_t>(), patch.size(),
+                       (int8_t *)ctx->d_lcp_ret2, err.as<uint32_t>());
+    HIP_CHECK(ctx, hipGetLastError());
+    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    if (check_err(ctx, err.as<uint32_t>()))
+        return -1;
+    release_acct(ctx);
+    ctx->d_acct = (const sre_account_entry *)out;
+    ctx->na = new_na;
+    ctx->own_acct = true;
+    ctx->acct_pool_bytes = acct_bytes;
+    std::swap(ctx->d_lcp_ret, ctx->d_lcp_ret2);
+    std::swap(ctx->lcp_ret_capacity, ctx->lcp_ret2_capacity);
+    ctx->lcp_valid = true;
+    return 0;
+}
+
 extern "C" int sre_root_retaining(sre_ctx *ctx, uint8_t out_root[32])
 {
     HIP_CHECK(ctx, hipSetDevice(ctx->device));
     memset(&ctx->stats, 0, sizeof(ctx->stats));
     ctx->cells_valid = false;
     ctx->snap_valid = false;
+    ctx->lcp_valid = false;
     uint64_t want_cap = 2 * (ctx->na < (uint64_t)N_CELLS ? ctx->na
                                                          : (uint64_t)N_CELLS) +
                         8192;
@@ -4912,25 +5229,38 @@ static int incremental_root_impl(sre_ctx *ctx,
     }
 
     DBuf map(ctx);
-    // carry only when the pre-delta state HAS storage: while ns stays 0
-    // every retained root is EMPTY_ROOT (and accounts-only fast-path
-    // deltas may have drifted d_roots_ret's indexing — harmless, since it
-    // is then never read)
-    if (apply_delta_impl(ctx, acct_delta, n_acct, st_delta, n_st, &map,
-                         (new_roots && ctx->ns > 0)
-                             ? (const uint8_t *)ctx->d_roots_ret
-                             : nullptr,
-                         (uint8_t *)new_roots))
-        return -1;
+    // small accounts-only deltas take the one-pass merge with lcp repair
+    // (closed-form ranks; O(delta) boundary lcps patched) — the scan-floor
+    // fix for the configs[4] shape. Everything else keeps the general
+    // overlay merge.
+    bool small = no_storage && n_acct > 0 && ctx->lcp_valid &&
+                 n_acct <= (ctx->na >> 6);
+    if (small) {
+        if (apply_delta_small(ctx, acct_delta, n_acct, &map))
+            return -1;
+    } else {
+        // carry only when the pre-delta state HAS storage: while ns stays 0
+        // every retained root is EMPTY_ROOT (and accounts-only fast-path
+        // deltas may have drifted d_roots_ret's indexing — harmless, since
+        // it is then never read)
+        if (apply_delta_impl(ctx, acct_delta, n_acct, st_delta, n_st, &map,
+                             (new_roots && ctx->ns > 0)
+                                 ? (const uint8_t *)ctx->d_roots_ret
+                                 : nullptr,
+                             (uint8_t *)new_roots))
+            return -1;
+        ctx->lcp_valid = false; // recomputed below
+    }
     uint64_t na = ctx->na;
     if (na == 0) {
         ctx->cells_valid = false;
     ctx->snap_valid = false;
+    ctx->lcp_valid = false;
         memcpy(out_root, EMPTY_ROOT_H, 32);
         return 0;
     }
 
-    DBuf err(ctx), bitmap(ctx), dl(ctx), dls(ctx), lcp(ctx), recs(ctx),
+    DBuf err(ctx), bitmap(ctx), dl(ctx), dls(ctx), recs(ctx),
         depths(ctx), hist(ctx), roots(ctx), capcnt(ctx);
     HIP_CHECK(ctx, err.alloc(4));
     HIP_CHECK(ctx, hipMemsetAsync(err.p, 0, 4, ctx->stream));
@@ -5015,11 +5345,20 @@ static int incremental_root_impl(sre_ctx *ctx,
         if (check_err(ctx, err.as<uint32_t>()))
             return -1;
     }
-    HIP_CHECK(ctx, lcp.alloc(na + 1));
-    hipLaunchKernelGGL(k_lcp_account, dim3(grid_for(na + 1)), dim3(BLOCK), 0,
-                       ctx->stream, ctx->d_acct, na, 0, lcp.as<int8_t>(),
-                       err.as<uint32_t>());
-    HIP_CHECK(ctx, hipGetLastError());
+    // lcp lives in the retained ping-pong buffer: the small path repaired
+    // it already; the general path recomputes it here (arming repair for
+    // the next delta)
+    if (!small) {
+        if (ensure_lcp_ret(ctx, &ctx->d_lcp_ret, &ctx->lcp_ret_capacity,
+                           na + 1))
+            return -1;
+        hipLaunchKernelGGL(k_lcp_account, dim3(grid_for(na + 1)), dim3(BLOCK),
+                           0, ctx->stream, ctx->d_acct, na, 0,
+                           (int8_t *)ctx->d_lcp_ret, err.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        ctx->lcp_valid = true;
+    }
+    int8_t *lcp_p = (int8_t *)ctx->d_lcp_ret;
     DBuf covered(ctx);
     HIP_CHECK(ctx, recs.alloc(na * sizeof(node_rec)));
     HIP_CHECK(ctx, depths.alloc(na));
@@ -5037,7 +5376,7 @@ static int incremental_root_impl(sre_ctx *ctx,
         hipLaunchKernelGGL(k_revalidate_rows, dim3(grid_for(ctx->cap_count)),
                            dim3(BLOCK), 0, ctx->stream,
                            (const cap_row *)ctx->d_cap_rows, ctx->cap_count,
-                           map.as<uint32_t>(), lcp.as<int8_t>(),
+                           map.as<uint32_t>(), lcp_p,
                            bitmap.as<uint8_t>(), recs.as<node_rec>(),
                            depths.as<uint8_t>(), covered.as<uint8_t>(),
                            hist.as<uint32_t>(),
@@ -5049,13 +5388,39 @@ static int incremental_root_impl(sre_ctx *ctx,
     hipEventCreate(&ev0);
     hipEventCreate(&ev1);
     hipEventRecord(ev0, ctx->stream);
-    hipLaunchKernelGGL(k_leaf_account, dim3(grid_for(na)), dim3(BLOCK), 0,
-                       ctx->stream, ctx->d_acct, na,
-                       (const uint8_t *)new_roots /* null => EMPTY_ROOT */,
-                       lcp.as<int8_t>(),
-                       0, recs.as<node_rec>(), depths.as<uint8_t>(),
-                       hist.as<uint32_t>(), roots.as<uint8_t>(), nullptr,
-                       nullptr, bitmap.as<uint8_t>(), covered.as<uint8_t>());
+    // compact the recompute set (covered == 0: dirty cells leave their
+    // positions uncovered, so this is exactly dirty ∪ uncovered) and
+    // launch the leaf kernel over it instead of sweeping all na lanes
+    uint32_t n_active = 0;
+    DBuf alist(ctx);
+    {
+        uint32_t nblk = grid_for(na);
+        DBuf ac(ctx), ao(ctx);
+        HIP_CHECK(ctx, ac.alloc((uint64_t)nblk * 4));
+        HIP_CHECK(ctx, ao.alloc((uint64_t)nblk * 4));
+        hipLaunchKernelGGL(k_active_hist, dim3(nblk), dim3(BLOCK), 0,
+                           ctx->stream, covered.as<uint8_t>(), na,
+                           ac.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        if (scan_u32(ctx, ac.as<uint32_t>(), ao.as<uint32_t>(), nblk,
+                     &n_active))
+            return -1;
+        HIP_CHECK(ctx, alist.alloc(((uint64_t)n_active ? n_active : 1) * 4));
+        hipLaunchKernelGGL(k_active_scatter, dim3(nblk), dim3(BLOCK), 0,
+                           ctx->stream, covered.as<uint8_t>(), na,
+                           ao.as<uint32_t>(), alist.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+    }
+    if (n_active)
+        hipLaunchKernelGGL(k_leaf_account, dim3(grid_for(n_active)),
+                           dim3(BLOCK), 0,
+                           ctx->stream, ctx->d_acct, na,
+                           (const uint8_t *)new_roots /* null=>EMPTY_ROOT */,
+                           lcp_p,
+                           0, recs.as<node_rec>(), depths.as<uint8_t>(),
+                           hist.as<uint32_t>(), roots.as<uint8_t>(), nullptr,
+                           nullptr, nullptr, nullptr,
+                           alist.as<uint32_t>(), n_active);
     HIP_CHECK(ctx, hipGetLastError());
     hipEventRecord(ev1, ctx->stream);
     HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
@@ -5092,7 +5457,7 @@ static int incremental_root_impl(sre_ctx *ctx,
         (const uint8_t *)ctx->d_acct + offsetof(sre_account_entry, key);
     size_t upd_start = ctx->updates.size();
     if (run_levels(ctx, na, recs.as<node_rec>(), depths.as<uint8_t>(),
-                   lcp.as<int8_t>(), keys, sizeof(sre_account_entry), hist_host,
+                   lcp_p, keys, sizeof(sre_account_entry), hist_host,
                    0, roots.as<uint8_t>(), nullptr, nullptr, err.as<uint32_t>(),
                    &po, with_updates ? 0 : -1,
                    with_updates ? abhash.as<uint8_t>() : nullptr,
@@ -5316,6 +5681,7 @@ extern "C" int sre_root_from_nodes(sre_ctx *ctx,
     memset(&ctx->stats, 0, sizeof(ctx->stats));
     ctx->cells_valid = false;
     ctx->snap_valid = false;
+    ctx->lcp_valid = false;
     hipEvent_t t0, t1;
     hipEventCreate(&t0);
     hipEventCreate(&t1);
