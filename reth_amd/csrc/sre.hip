@@ -1651,29 +1651,48 @@ __global__ void __launch_bounds__(BLOCK) k_branch_fused1(
     if (g < n_groups) {
         uint32_t gg = perm[g];
         uint64_t j = gs[gg], jend = gs[gg + 1];
+        int nmem = (int)(jend - j);
+        bool sized_ok = nmem >= 2 && nmem <= 3;
+        // validation words of all members issued up front (2 u32 each:
+        // ref_len word + pad word). This both validates and WARMS the
+        // records' cache lines, so the emit pass re-reads them from L1 —
+        // caching the full 48-B records in registers measured as scratch
+        // spills instead.
+        uint32_t w3[3], w11[3], w0[3], w1[3], w2[3];
+#pragma unroll
+        for (int m = 0; m < 3; ++m)
+            if (sized_ok && m < nmem) {
+                const uint32_t *r32 = (const uint32_t *)&L[j + m];
+                w0[m] = r32[0];
+                w1[m] = r32[1];
+                w2[m] = r32[2];
+                w3[m] = r32[3];
+                w11[m] = r32[11];
+            }
         br_meta mt;
-        mt.s = L[j].s;
-        mt.e = L[jend - 1].e;
-        mt.seg = L[j].seg;
+        mt.s = sized_ok ? w0[0] : L[j].s;
+        mt.e = sized_ok ? w1[nmem - 1] : L[jend - 1].e;
+        mt.seg = sized_ok ? w2[0] : L[j].seg;
         mt.d = (uint8_t)d;
         int8_t pl = lcp[mt.s], pr = lcp[mt.e];
         mt.P = pl > pr ? pl : pr;
-        int nmem = (int)(jend - j);
         uint64_t nibs = 0;
         int payload = 1 + (16 - nmem);
-        bool order_ok = nmem >= 2 && nmem <= 3;
+        bool order_ok = sized_ok;
         bool stored = false;
-        {
+        if (sized_ok) {
             int prev = -1;
-            for (uint64_t mm = j; mm < jend && order_ok; ++mm) {
-                uint8_t pb = L[mm].pad_;
-                int nbm = pb & 0xF;
-                order_ok &= nbm > prev;
-                prev = nbm;
-                nibs |= (uint64_t)nbm << (4 * (mm - j));
-                payload += L[mm].ref_len;
-                stored |= (pb & 0x20) != 0;
-            }
+#pragma unroll
+            for (int m = 0; m < 3; ++m)
+                if (m < nmem) {
+                    uint8_t pb = (uint8_t)(w11[m] >> 24);
+                    int nbm = pb & 0xF;
+                    order_ok &= nbm > prev;
+                    prev = nbm;
+                    nibs |= (uint64_t)nbm << (4 * m);
+                    payload += (w3[m] >> 8) & 0xFF;
+                    stored |= (pb & 0x20) != 0;
+                }
         }
         mt.flags = stored ? 1 : 0;
         node_rec *r = &out[gg];
@@ -1699,32 +1718,35 @@ __global__ void __launch_bounds__(BLOCK) k_branch_fused1(
                 ap.put((uint8_t)payload);
             }
             {
-                uint64_t m = j;
-                for (int b = 0; b < 16; ++b) {
-                    if (m < jend &&
-                        (int)((nibs >> (4 * (m - j))) & 0xf) == b) {
-                        int rl = L[m].ref_len;
-                        const uint32_t *rec32 = (const uint32_t *)&L[m];
-                        uint32_t w[9];
+                int prevnib = -1;
+#pragma unroll
+                for (int m = 0; m < 3; ++m)
+                    if (m < nmem) {
+                        int nbm = (int)((nibs >> (4 * m)) & 0xF);
+                        for (int b = prevnib + 1; b < nbm; ++b)
+                            ap.put(0x80);
+                        prevnib = nbm;
+                        int rl = (w3[m] >> 8) & 0xFF;
+                        const uint32_t *rec32 =
+                            (const uint32_t *)&L[j + m]; // L1-hot re-read
+                        uint32_t v[9];
 #pragma unroll
                         for (int k = 0; k < 9; ++k)
-                            w[k] = rec32[3 + k];
+                            v[k] = rec32[3 + k];
                         uint64_t R[5];
-                        R[0] = ((uint64_t)w[0] >> 16) | ((uint64_t)w[1] << 16) |
-                               ((uint64_t)w[2] << 48);
-                        R[1] = ((uint64_t)w[2] >> 16) | ((uint64_t)w[3] << 16) |
-                               ((uint64_t)w[4] << 48);
-                        R[2] = ((uint64_t)w[4] >> 16) | ((uint64_t)w[5] << 16) |
-                               ((uint64_t)w[6] << 48);
-                        R[3] = ((uint64_t)w[6] >> 16) | ((uint64_t)w[7] << 16) |
-                               ((uint64_t)w[8] << 48);
-                        R[4] = ((uint64_t)w[8] >> 16) & 0xFF;
+                        R[0] = ((uint64_t)v[0] >> 16) | ((uint64_t)v[1] << 16) |
+                               ((uint64_t)v[2] << 48);
+                        R[1] = ((uint64_t)v[2] >> 16) | ((uint64_t)v[3] << 16) |
+                               ((uint64_t)v[4] << 48);
+                        R[2] = ((uint64_t)v[4] >> 16) | ((uint64_t)v[5] << 16) |
+                               ((uint64_t)v[6] << 48);
+                        R[3] = ((uint64_t)v[6] >> 16) | ((uint64_t)v[7] << 16) |
+                               ((uint64_t)v[8] << 48);
+                        R[4] = ((uint64_t)v[8] >> 16) & 0xFF;
                         ap.put_bytes33(R, rl);
-                        m++;
-                    } else {
-                        ap.put(0x80);
                     }
-                }
+                for (int b = prevnib + 1; b < 16; ++b)
+                    ap.put(0x80);
                 ap.put(0x80); // empty value item
             }
             ap.put(0x01); // keccak pad start
