@@ -2432,6 +2432,149 @@ __global__ void k_sd_lcp_patch(const sre_account_entry *__restrict__ out,
     lcp_new[j] = (int8_t)l;
 }
 
+// ---- closed-form STORAGE merge for small deltas (dirty-path): the same
+// rank trick as the account side plus destroyed-account WIPE RANGES —
+// each destroyed account's base rows form one contiguous (acct-key
+// prefixed) range; removed-rows-before-i is then a prefix sum over at
+// most a few thousand ranges, all L1/L2-resident. Replaces the O(ns)
+// flags + scans + posmap of the general storage merge; what remains is
+// the irreducible array rewrite.
+
+__global__ void k_sds_marks(const sre_storage_entry *__restrict__ base,
+                            uint64_t ns,
+                            const sre_storage_entry *__restrict__ dl,
+                            uint64_t nst, uint32_t *__restrict__ bpos,
+                            uint32_t *__restrict__ match)
+{
+    uint64_t k = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= nst)
+        return;
+    uint64_t p = lb_keys((const uint8_t *)base, sizeof(sre_storage_entry), ns,
+                         (const uint8_t *)&dl[k], 64);
+    bpos[k] = (uint32_t)p;
+    match[k] = (p < ns && cmp_key64((const uint8_t *)&base[p],
+                                    (const uint8_t *)&dl[k]) == 0)
+                   ? 1u
+                   : 0u;
+}
+
+// wipe range of each destroyed account: [first row with this acct_key,
+// first row past it)
+__global__ void k_sds_wipes(const sre_storage_entry *__restrict__ base,
+                            uint64_t ns, const uint8_t *__restrict__ sdel,
+                            uint32_t ndel, uint32_t *__restrict__ wlo,
+                            uint32_t *__restrict__ whi)
+{
+    uint32_t a = blockIdx.x * blockDim.x + threadIdx.x;
+    if (a >= ndel)
+        return;
+    const uint8_t *key = sdel + 32ull * a;
+    uint64_t lo = 0, hi = ns;
+    while (lo < hi) { // first i with acct_key >= key
+        uint64_t mid = (lo + hi) / 2;
+        if (cmp_key32(base[mid].acct_key, key) < 0)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    wlo[a] = (uint32_t)lo;
+    uint64_t lo2 = lo;
+    hi = ns;
+    while (lo2 < hi) { // first i with acct_key > key
+        uint64_t mid = (lo2 + hi) / 2;
+        if (cmp_key32(base[mid].acct_key, key) <= 0)
+            lo2 = mid + 1;
+        else
+            hi = mid;
+    }
+    whi[a] = (uint32_t)lo2;
+}
+
+__device__ __forceinline__ uint64_t wipes_before(const uint32_t *wlo,
+                                                 const uint32_t *whi,
+                                                 const uint32_t *wsum,
+                                                 uint32_t ndel, uint64_t i,
+                                                 bool *wiped)
+{
+    // last range with wlo <= i
+    int lo = 0, hi = (int)ndel;
+    while (lo < hi) {
+        int mid = (lo + hi) / 2;
+        if ((uint64_t)wlo[mid] <= i)
+            lo = mid + 1;
+        else
+            hi = mid;
+    }
+    int r = lo - 1;
+    *wiped = false;
+    if (r < 0)
+        return 0;
+    uint32_t clip = i < whi[r] ? (uint32_t)i : whi[r];
+    *wiped = i < whi[r];
+    return wsum[r] + (clip > wlo[r] ? clip - wlo[r] : 0);
+}
+
+__global__ void k_sds_scatter(const sre_storage_entry *__restrict__ base,
+                              uint64_t ns,
+                              const sre_storage_entry *__restrict__ dl,
+                              uint64_t nst,
+                              const uint32_t *__restrict__ m_excl, /* nst+1 */
+                              const uint32_t *__restrict__ eff_excl,
+                              const uint32_t *__restrict__ wlo,
+                              const uint32_t *__restrict__ whi,
+                              const uint32_t *__restrict__ wsum, /* ndel+1 */
+                              uint32_t ndel,
+                              sre_storage_entry *__restrict__ out)
+{
+    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= ns)
+        return;
+    bool wiped;
+    uint64_t wb = wipes_before(wlo, whi, wsum, ndel, i, &wiped);
+    uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_storage_entry), nst,
+                         (const uint8_t *)&base[i], 64);
+    bool matched = p < nst && cmp_key64((const uint8_t *)&dl[p],
+                                        (const uint8_t *)&base[i]) == 0;
+    if (matched || wiped)
+        return;
+    uint64_t j = i - m_excl[p] - wb + eff_excl[p];
+    const uint64_t *s8 = (const uint64_t *)&base[i];
+    uint64_t *d8 = (uint64_t *)&out[j];
+#pragma unroll
+    for (int w = 0; w < 12; ++w) // 96 B
+        d8[w] = s8[w];
+}
+
+__global__ void k_sds_place(const sre_storage_entry *__restrict__ dl,
+                            uint64_t nst, const uint32_t *__restrict__ bpos,
+                            const uint32_t *__restrict__ m_excl,
+                            const uint32_t *__restrict__ eff_excl,
+                            const uint32_t *__restrict__ wlo,
+                            const uint32_t *__restrict__ whi,
+                            const uint32_t *__restrict__ wsum, uint32_t ndel,
+                            sre_storage_entry *__restrict__ out)
+{
+    uint64_t k = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    if (k >= nst)
+        return;
+    // zero value = slot deletion: no output row
+    const uint8_t *v = dl[k].value;
+    bool nz = false;
+#pragma unroll
+    for (int q = 0; q < 4; ++q)
+        nz |= ((const uint64_t *)v)[q] != 0;
+    if (!nz)
+        return;
+    bool wiped;
+    uint64_t wb = wipes_before(wlo, whi, wsum, ndel, bpos[k], &wiped);
+    uint64_t j = bpos[k] - m_excl[k] - wb + eff_excl[k];
+    const uint64_t *s8 = (const uint64_t *)&dl[k];
+    uint64_t *d8 = (uint64_t *)&out[j];
+#pragma unroll
+    for (int w = 0; w < 12; ++w)
+        d8[w] = s8[w];
+}
+
 // sparse-leaf support: compact the positions no seed covers (dirty cells
 // leave covered == 0 too, so this is exactly the recompute set)
 __global__ void k_active_hist(const uint8_t *__restrict__ covered, uint64_t n,
@@ -4929,61 +5072,197 @@ static int apply_delta_impl(sre_ctx *ctx, const sre_account_delta *acct_delta,
     HIP_CHECK(ctx, hipGetLastError());
 
     // ---- storage ----
-    HIP_CHECK(ctx, sa.alloc((ns + 1) * 4));
-    HIP_CHECK(ctx, sd.alloc((n_st + 1) * 4));
-    hipLaunchKernelGGL(k_ovl_base_st_flags, dim3(grid_for(ns + 1)), dim3(BLOCK), 0,
-                       ctx->stream, ctx->d_st, ns, dl_s.as<sre_storage_entry>(),
-                       n_st, sdel.as<uint8_t>(), (uint64_t)Ndel,
-                       sa.as<uint32_t>());
-    hipLaunchKernelGGL(k_ovl_delta_st_flags, dim3(grid_for(n_st + 1)), dim3(BLOCK),
-                       0, ctx->stream, dl_s.as<sre_storage_entry>(), n_st,
-                       sdel.as<uint8_t>(), (uint64_t)Ndel, sd.as<uint32_t>(),
-                       err.as<uint32_t>());
-    HIP_CHECK(ctx, hipGetLastError());
-    uint32_t Sk = 0, Tk = 0;
-    DBuf esa(ctx), esd(ctx);
-    HIP_CHECK(ctx, esa.alloc((ns + 1) * 4));
-    HIP_CHECK(ctx, esd.alloc((n_st + 1) * 4));
-    if (scan_u32(ctx, sa.as<uint32_t>(), esa.as<uint32_t>(), ns + 1, &Sk))
-        return -1;
-    if (scan_u32(ctx, sd.as<uint32_t>(), esd.as<uint32_t>(), n_st + 1, &Tk))
-        return -1;
-    if (check_err(ctx, err.as<uint32_t>())) {
-        pool_put(ctx, acct_bytes, new_acct);
-        return -1;
-    }
-    uint64_t new_ns = Sk + Tk;
-    size_t st_bytes = (new_ns ? new_ns : 1) * sizeof(sre_storage_entry);
-    void *new_st = pool_get(ctx, st_bytes);
-    if (!new_st) {
-        pool_put(ctx, acct_bytes, new_acct);
-        set_err(ctx, "apply_delta: out of memory (storage)");
-        return -1;
-    }
-    gmax = ns > n_st ? ns : n_st;
-    if (gmax)
-        hipLaunchKernelGGL(k_ovl_scatter_st, dim3(grid_for(gmax)), dim3(BLOCK), 0,
-                           ctx->stream, ctx->d_st, ns, esa.as<uint32_t>(),
-                           dl_s.as<sre_storage_entry>(), n_st, esd.as<uint32_t>(),
+    uint64_t new_ns = 0;
+    size_t st_bytes = 0;
+    void *new_st = nullptr;
+    bool st_done = false, keep_st = false;
+    if (ns && n_st == 0 && Ndel == 0 && ctx->own_st) {
+        // accounts-only delta over a storage-bearing state: the storage
+        // array is untouched — keep it in place (engine-owned only: a
+        // borrowed array may be released by the caller after the delta)
+        new_ns = ns;
+        st_done = keep_st = true;
+    } else if (ns > (1ull << 20) && n_st <= (ns >> 8) &&
+               (uint64_t)Ndel <= 4096) {
+        // closed-form small-delta storage merge: delta ranks + wipe-range
+        // prefix sums instead of O(ns) flags/scans/posmap. Host-side
+        // validation of what k_ovl_delta_st_flags checked on device.
+        bool ok = true;
+        std::vector<std::array<uint8_t, 32>> dead;
+        for (uint64_t i = 0; i < n_acct; ++i)
+            if (acct_delta[i].deleted) {
+                std::array<uint8_t, 32> k;
+                memcpy(k.data(), acct_delta[i].key, 32);
+                dead.push_back(k);
+            }
+        for (uint64_t i = 0; i < n_st && ok; ++i) {
+            if (i && memcmp(&st_delta[i - 1], &st_delta[i], 64) >= 0)
+                ok = false; // unsorted or duplicate (acct,slot)
+            auto it = std::lower_bound(
+                dead.begin(), dead.end(), st_delta[i].acct_key,
+                [](const std::array<uint8_t, 32> &a, const uint8_t *b) {
+                    return memcmp(a.data(), b, 32) < 0;
+                });
+            if (it != dead.end() &&
+                memcmp(it->data(), st_delta[i].acct_key, 32) == 0)
+                ok = false; // storage row for a destroyed account
+        }
+        if (!ok) {
+            pool_put(ctx, acct_bytes, new_acct);
+            set_err(ctx, "apply_delta: storage delta unsorted/duplicate or "
+                         "row for a destroyed account");
+            return -1;
+        }
+        DBuf bpos(ctx), match(ctx), mex(ctx), eex(ctx), dwlo(ctx), dwhi(ctx),
+            dwsum(ctx);
+        HIP_CHECK(ctx, bpos.alloc((n_st ? n_st : 1) * 4));
+        HIP_CHECK(ctx, match.alloc((n_st ? n_st : 1) * 4));
+        if (n_st) {
+            hipLaunchKernelGGL(k_sds_marks, dim3(grid_for(n_st)), dim3(BLOCK),
+                               0, ctx->stream, ctx->d_st, ns,
+                               dl_s.as<sre_storage_entry>(), n_st,
+                               bpos.as<uint32_t>(), match.as<uint32_t>());
+            HIP_CHECK(ctx, hipGetLastError());
+        }
+        HIP_CHECK(ctx, dwlo.alloc((Ndel ? Ndel : 1) * 4));
+        HIP_CHECK(ctx, dwhi.alloc((Ndel ? Ndel : 1) * 4));
+        HIP_CHECK(ctx, dwsum.alloc(((uint64_t)Ndel + 1) * 4));
+        if (Ndel) {
+            hipLaunchKernelGGL(k_sds_wipes, dim3(grid_for(Ndel)), dim3(BLOCK),
+                               0, ctx->stream, ctx->d_st, ns,
+                               sdel.as<uint8_t>(), Ndel, dwlo.as<uint32_t>(),
+                               dwhi.as<uint32_t>());
+            HIP_CHECK(ctx, hipGetLastError());
+        }
+        std::vector<uint32_t> bpos_h(n_st), match_h(n_st), wlo_h(Ndel),
+            whi_h(Ndel), wsum_h(Ndel + 1);
+        if (n_st) {
+            HIP_CHECK(ctx, hipMemcpy(bpos_h.data(), bpos.p, 4 * n_st,
+                                     hipMemcpyDeviceToHost));
+            HIP_CHECK(ctx, hipMemcpy(match_h.data(), match.p, 4 * n_st,
+                                     hipMemcpyDeviceToHost));
+        }
+        if (Ndel) {
+            HIP_CHECK(ctx, hipMemcpy(wlo_h.data(), dwlo.p, 4ull * Ndel,
+                                     hipMemcpyDeviceToHost));
+            HIP_CHECK(ctx, hipMemcpy(whi_h.data(), dwhi.p, 4ull * Ndel,
+                                     hipMemcpyDeviceToHost));
+        }
+        wsum_h[0] = 0;
+        for (uint32_t a = 0; a < Ndel; ++a)
+            wsum_h[a + 1] = wsum_h[a] + (whi_h[a] - wlo_h[a]);
+        std::vector<uint32_t> mex_h(n_st + 1), eex_h(n_st + 1);
+        mex_h[0] = eex_h[0] = 0;
+        for (uint64_t k = 0; k < n_st; ++k) {
+            bool nz = false;
+            for (int q = 0; q < 32; ++q)
+                nz |= st_delta[k].value[q] != 0;
+            mex_h[k + 1] = mex_h[k] + (match_h[k] ? 1 : 0);
+            eex_h[k + 1] = eex_h[k] + (nz ? 1 : 0);
+        }
+        new_ns = ns - mex_h[n_st] - wsum_h[Ndel] + eex_h[n_st];
+        HIP_CHECK(ctx, mex.alloc(4 * (n_st + 1)));
+        HIP_CHECK(ctx, eex.alloc(4 * (n_st + 1)));
+        HIP_CHECK(ctx, hipMemcpyAsync(mex.p, mex_h.data(), 4 * (n_st + 1),
+                                      hipMemcpyHostToDevice, ctx->stream));
+        HIP_CHECK(ctx, hipMemcpyAsync(eex.p, eex_h.data(), 4 * (n_st + 1),
+                                      hipMemcpyHostToDevice, ctx->stream));
+        HIP_CHECK(ctx, hipMemcpyAsync(dwsum.p, wsum_h.data(),
+                                      4ull * (Ndel + 1),
+                                      hipMemcpyHostToDevice, ctx->stream));
+        st_bytes = (new_ns ? new_ns : 1) * sizeof(sre_storage_entry);
+        new_st = pool_get(ctx, st_bytes);
+        if (!new_st) {
+            pool_put(ctx, acct_bytes, new_acct);
+            set_err(ctx, "apply_delta: out of memory (storage)");
+            return -1;
+        }
+        hipLaunchKernelGGL(k_sds_scatter, dim3(grid_for(ns)), dim3(BLOCK), 0,
+                           ctx->stream, ctx->d_st, ns,
+                           dl_s.as<sre_storage_entry>(), n_st,
+                           mex.as<uint32_t>(), eex.as<uint32_t>(),
+                           dwlo.as<uint32_t>(), dwhi.as<uint32_t>(),
+                           dwsum.as<uint32_t>(), Ndel,
                            (sre_storage_entry *)new_st);
-    HIP_CHECK(ctx, hipGetLastError());
-    HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+        HIP_CHECK(ctx, hipGetLastError());
+        if (n_st) {
+            hipLaunchKernelGGL(k_sds_place, dim3(grid_for(n_st)), dim3(BLOCK),
+                               0, ctx->stream, dl_s.as<sre_storage_entry>(),
+                               n_st, bpos.as<uint32_t>(), mex.as<uint32_t>(),
+                               eex.as<uint32_t>(), dwlo.as<uint32_t>(),
+                               dwhi.as<uint32_t>(), dwsum.as<uint32_t>(),
+                               Ndel, (sre_storage_entry *)new_st);
+            HIP_CHECK(ctx, hipGetLastError());
+        }
+        HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+        st_done = true;
+    }
+    if (!st_done) {
+        HIP_CHECK(ctx, sa.alloc((ns + 1) * 4));
+        HIP_CHECK(ctx, sd.alloc((n_st + 1) * 4));
+        hipLaunchKernelGGL(k_ovl_base_st_flags, dim3(grid_for(ns + 1)),
+                           dim3(BLOCK), 0,
+                           ctx->stream, ctx->d_st, ns,
+                           dl_s.as<sre_storage_entry>(),
+                           n_st, sdel.as<uint8_t>(), (uint64_t)Ndel,
+                           sa.as<uint32_t>());
+        hipLaunchKernelGGL(k_ovl_delta_st_flags, dim3(grid_for(n_st + 1)),
+                           dim3(BLOCK),
+                           0, ctx->stream, dl_s.as<sre_storage_entry>(), n_st,
+                           sdel.as<uint8_t>(), (uint64_t)Ndel,
+                           sd.as<uint32_t>(),
+                           err.as<uint32_t>());
+        HIP_CHECK(ctx, hipGetLastError());
+        uint32_t Sk = 0, Tk = 0;
+        DBuf esa(ctx), esd(ctx);
+        HIP_CHECK(ctx, esa.alloc((ns + 1) * 4));
+        HIP_CHECK(ctx, esd.alloc((n_st + 1) * 4));
+        if (scan_u32(ctx, sa.as<uint32_t>(), esa.as<uint32_t>(), ns + 1, &Sk))
+            return -1;
+        if (scan_u32(ctx, sd.as<uint32_t>(), esd.as<uint32_t>(), n_st + 1,
+                     &Tk))
+            return -1;
+        if (check_err(ctx, err.as<uint32_t>())) {
+            pool_put(ctx, acct_bytes, new_acct);
+            return -1;
+        }
+        new_ns = Sk + Tk;
+        st_bytes = (new_ns ? new_ns : 1) * sizeof(sre_storage_entry);
+        new_st = pool_get(ctx, st_bytes);
+        if (!new_st) {
+            pool_put(ctx, acct_bytes, new_acct);
+            set_err(ctx, "apply_delta: out of memory (storage)");
+            return -1;
+        }
+        gmax = ns > n_st ? ns : n_st;
+        if (gmax)
+            hipLaunchKernelGGL(k_ovl_scatter_st, dim3(grid_for(gmax)),
+                               dim3(BLOCK), 0,
+                               ctx->stream, ctx->d_st, ns, esa.as<uint32_t>(),
+                               dl_s.as<sre_storage_entry>(), n_st,
+                               esd.as<uint32_t>(),
+                               (sre_storage_entry *)new_st);
+        HIP_CHECK(ctx, hipGetLastError());
+        HIP_CHECK(ctx, hipStreamSynchronize(ctx->stream));
+    }
 
     // adopt the merged state (pool-backed; reused across repeated deltas)
     release_acct(ctx);
-    release_st(ctx);
     ctx->d_acct = (const sre_account_entry *)new_acct;
     ctx->na = new_na;
     ctx->own_acct = true;
     ctx->acct_pool_bytes = acct_bytes;
-    ctx->d_st = (const sre_storage_entry *)new_st;
-    ctx->ns = new_ns;
-    ctx->own_st = true;
-    ctx->st_pool_bytes = st_bytes;
+    if (!keep_st) { // keep_st: untouched storage array stays in place
+        release_st(ctx);
+        ctx->d_st = (const sre_storage_entry *)new_st;
+        ctx->ns = new_ns;
+        ctx->own_st = true;
+        ctx->st_pool_bytes = st_bytes;
+    }
     // prewarm the ping-pong partner buffers: the NEXT apply_delta would
     // otherwise pay a multi-hundred-ms first-time hipMalloc of this size
     // class inside the caller's timed region
-    for (size_t b : {acct_bytes, st_bytes}) {
+    for (size_t b : {acct_bytes, keep_st ? acct_bytes : st_bytes}) {
         void *spare = pool_get(ctx, b);
         if (spare)
             pool_put(ctx, b, spare);
