@@ -2345,15 +2345,45 @@ __global__ void k_sd_scatter(const sre_account_entry *__restrict__ base,
                              sre_account_entry *__restrict__ out,
                              uint32_t *__restrict__ map)
 {
-    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    // the delta rank p is monotone in the base key, so the whole block
+    // shares a tiny search window computed once from its first/last row
+    // (usually width 0-2: 5k delta positions spread over 200M rows) —
+    // the full log2(nd) probe loop per row was half the scatter's cost
+    __shared__ uint32_t plo_s, phi_s;
+    uint64_t i0 = (uint64_t)blockIdx.x * blockDim.x;
+    if (threadIdx.x == 0) {
+        uint64_t a = i0 < nb ? i0 : (nb ? nb - 1 : 0);
+        plo_s = nb ? (uint32_t)lb_keys((const uint8_t *)dl,
+                                       sizeof(sre_account_delta), nd,
+                                       base[a].key, 32)
+                   : 0;
+        uint64_t b = i0 + blockDim.x - 1;
+        if (b >= nb)
+            b = nb ? nb - 1 : 0;
+        phi_s = nb ? (uint32_t)lb_keys((const uint8_t *)dl,
+                                       sizeof(sre_account_delta), nd,
+                                       base[b].key, 32)
+                   : 0;
+    }
+    __syncthreads();
+    uint64_t i = i0 + threadIdx.x;
     if (i > nb)
         return;
     if (i == nb) {
         map[nb] = (uint32_t)(nb - m_excl[nd] + eff_excl[nd]);
         return;
     }
-    uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_account_delta), nd,
-                         base[i].key, 32);
+    uint64_t lo = plo_s, hi = phi_s;
+    // lb within [lo, hi+1): first delta key >= base key
+    uint64_t hh = hi < nd ? hi + 1 : nd;
+    while (lo < hh) {
+        uint64_t mid = (lo + hh) / 2;
+        if (cmp_key32(dl[mid].key, base[i].key) < 0)
+            lo = mid + 1;
+        else
+            hh = mid;
+    }
+    uint64_t p = lo;
     bool m = p < nd && cmp_key32(dl[p].key, base[i].key) == 0;
     uint32_t j = (uint32_t)(i - m_excl[p] + eff_excl[p]);
     map[i] = j;
@@ -2526,13 +2556,37 @@ __global__ void k_sds_scatter(const sre_storage_entry *__restrict__ base,
                               uint32_t ndel,
                               sre_storage_entry *__restrict__ out)
 {
-    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    // block-shared rank window (see k_sd_scatter)
+    __shared__ uint32_t plo_s, phi_s;
+    uint64_t i0 = (uint64_t)blockIdx.x * blockDim.x;
+    if (threadIdx.x == 0) {
+        uint64_t a = i0 < ns ? i0 : ns - 1;
+        plo_s = (uint32_t)lb_keys((const uint8_t *)dl,
+                                  sizeof(sre_storage_entry), nst,
+                                  (const uint8_t *)&base[a], 64);
+        uint64_t b = i0 + blockDim.x - 1;
+        if (b >= ns)
+            b = ns - 1;
+        phi_s = (uint32_t)lb_keys((const uint8_t *)dl,
+                                  sizeof(sre_storage_entry), nst,
+                                  (const uint8_t *)&base[b], 64);
+    }
+    __syncthreads();
+    uint64_t i = i0 + threadIdx.x;
     if (i >= ns)
         return;
     bool wiped;
     uint64_t wb = wipes_before(wlo, whi, wsum, ndel, i, &wiped);
-    uint64_t p = lb_keys((const uint8_t *)dl, sizeof(sre_storage_entry), nst,
-                         (const uint8_t *)&base[i], 64);
+    uint64_t lo = plo_s, hh = phi_s < nst ? (uint64_t)phi_s + 1 : nst;
+    while (lo < hh) {
+        uint64_t mid = (lo + hh) / 2;
+        if (cmp_key64((const uint8_t *)&dl[mid],
+                      (const uint8_t *)&base[i]) < 0)
+            lo = mid + 1;
+        else
+            hh = mid;
+    }
+    uint64_t p = lo;
     bool matched = p < nst && cmp_key64((const uint8_t *)&dl[p],
                                         (const uint8_t *)&base[i]) == 0;
     if (matched || wiped)
@@ -4983,6 +5037,16 @@ static int snap_cmp(const snap_row &a, const snap_row &b)
     return (int)a.path_len - (int)b.path_len;
 }
 
+// minimum resident ns for the closed-form small-delta storage merge;
+// SRE_ST_SMALL_MIN overrides (tests lower it to cover the path on small
+// states — the default keeps tiny states on the trivially-fast general
+// path)
+static uint64_t sds_min_ns()
+{
+    const char *e = getenv("SRE_ST_SMALL_MIN");
+    return e ? (uint64_t)atoll(e) : (1ull << 20);
+}
+
 static int apply_delta_impl(sre_ctx *ctx, const sre_account_delta *acct_delta,
                             uint64_t n_acct, const sre_storage_entry *st_delta,
                             uint64_t n_st, DBuf *map_out /* optional: old->new
@@ -5082,7 +5146,7 @@ static int apply_delta_impl(sre_ctx *ctx, const sre_account_delta *acct_delta,
         // borrowed array may be released by the caller after the delta)
         new_ns = ns;
         st_done = keep_st = true;
-    } else if (ns > (1ull << 20) && n_st <= (ns >> 8) &&
+    } else if (ns > sds_min_ns() && n_st <= (ns >> 8) &&
                (uint64_t)Ndel <= 4096) {
         // closed-form small-delta storage merge: delta ranks + wipe-range
         // prefix sums instead of O(ns) flags/scans/posmap. Host-side
