@@ -2367,6 +2367,25 @@ __global__ void k_sd_scatter(const sre_account_entry *__restrict__ base,
     }
     __syncthreads();
     uint64_t i = i0 + threadIdx.x;
+    // fast path: a full block whose span contains no delta key shifts by
+    // a constant — copy flat u64s, fully coalesced on both sides (at a
+    // 5k delta over 200M rows this is ~99.99% of blocks)
+    if (plo_s == phi_s && i0 + blockDim.x <= nb) {
+        uint32_t p = plo_s;
+        bool lastmatch =
+            p < nd &&
+            cmp_key32(dl[p].key, base[i0 + blockDim.x - 1].key) == 0;
+        if (!lastmatch) {
+            uint64_t j0 = i0 - m_excl[p] + eff_excl[p];
+            const uint64_t *s8 = (const uint64_t *)&base[i0];
+            uint64_t *d8 = (uint64_t *)&out[j0];
+            uint32_t n8 = blockDim.x * 13;
+            for (uint32_t k = threadIdx.x; k < n8; k += blockDim.x)
+                d8[k] = s8[k];
+            map[i] = (uint32_t)(i - m_excl[p] + eff_excl[p]);
+            return;
+        }
+    }
     if (i > nb)
         return;
     if (i == nb) {
@@ -2573,6 +2592,27 @@ __global__ void k_sds_scatter(const sre_storage_entry *__restrict__ base,
     }
     __syncthreads();
     uint64_t i = i0 + threadIdx.x;
+    // fast path: full block, no delta key in span, no wipe-range overlap
+    // -> constant shift, flat coalesced copy
+    if (plo_s == phi_s && i0 + blockDim.x <= ns) {
+        uint32_t p = plo_s;
+        uint64_t last = i0 + blockDim.x - 1;
+        bool w0, w1;
+        uint64_t wb0 = wipes_before(wlo, whi, wsum, ndel, i0, &w0);
+        uint64_t wb1 = wipes_before(wlo, whi, wsum, ndel, last, &w1);
+        bool lastmatch = p < nst &&
+                         cmp_key64((const uint8_t *)&dl[p],
+                                   (const uint8_t *)&base[last]) == 0;
+        if (!lastmatch && !w0 && !w1 && wb0 == wb1) {
+            uint64_t j0 = i0 - m_excl[p] - wb0 + eff_excl[p];
+            const uint64_t *s8 = (const uint64_t *)&base[i0];
+            uint64_t *d8 = (uint64_t *)&out[j0];
+            uint32_t n8 = blockDim.x * 12;
+            for (uint32_t k = threadIdx.x; k < n8; k += blockDim.x)
+                d8[k] = s8[k];
+            return;
+        }
+    }
     if (i >= ns)
         return;
     bool wiped;
