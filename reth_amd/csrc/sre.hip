@@ -900,7 +900,10 @@ __global__ void k_depth_scatter66(const uint8_t *__restrict__ depths,
                                   const node_rec *__restrict__ recs, uint64_t n,
                                   uint32_t nblk,
                                   const uint32_t *__restrict__ offs,
-                                  node_rec *__restrict__ out)
+                                  node_rec *__restrict__ out,
+                                  uint32_t *__restrict__ out_keys /* parallel
+                                  .s array: merge probes touch 4 B/rec
+                                  instead of 48-B-stride records */)
 {
     __shared__ uint32_t wh[66][BLOCK / 64]; // per-wave per-depth counts
     int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -928,6 +931,7 @@ __global__ void k_depth_scatter66(const uint8_t *__restrict__ depths,
         before += wh[key][w];
     uint32_t pos = offs[(uint64_t)key * nblk + blockIdx.x] + before + rank_in_wave;
     copy_rec(&out[pos], &recs[i]);
+    out_keys[pos] = recs[i].s;
 }
 
 
@@ -954,7 +958,8 @@ __global__ void k_depth_hist66_rec(const node_rec *__restrict__ recs,
 __global__ void k_depth_scatter66_rec(const node_rec *__restrict__ recs,
                                       uint64_t n, uint32_t nblk,
                                       const uint32_t *__restrict__ offs,
-                                      node_rec *__restrict__ out)
+                                      node_rec *__restrict__ out,
+                                      uint32_t *__restrict__ out_keys)
 {
     __shared__ uint32_t wh[66][BLOCK / 64];
     int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
@@ -981,15 +986,19 @@ __global__ void k_depth_scatter66_rec(const node_rec *__restrict__ recs,
         before += wh[key][w];
     uint32_t pos = offs[(uint64_t)key * nblk + blockIdx.x] + before + rank_in_wave;
     copy_rec(&out[pos], &recs[i]);
+    out_keys[pos] = recs[i].s;
 }
 
 
 
 // merge two record arrays sorted by .s (distinct keys)
 
+// Bkeys: compact 4-B .s array of B (12x denser probes than searching the
+// 48-B-stride records). Null -> probe the records directly.
 __global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
                           const node_rec *__restrict__ B, uint64_t nB,
-                          node_rec *__restrict__ out)
+                          node_rec *__restrict__ out,
+                          const uint32_t *__restrict__ Bkeys)
 {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= nA)
@@ -998,7 +1007,8 @@ __global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
     uint64_t lo = 0, hi = nB;
     while (lo < hi) {
         uint64_t mid = (lo + hi) / 2;
-        if (B[mid].s < key)
+        uint32_t bk = Bkeys ? Bkeys[mid] : B[mid].s;
+        if (bk < key)
             lo = mid + 1;
         else
             hi = mid;
@@ -1013,13 +1023,15 @@ __global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
 #define KWAY_MAX 12
 struct kway_desc {
     const node_rec *run[KWAY_MAX];
+    const uint32_t *keys[KWAY_MAX]; // compact .s arrays (may be null)
     uint64_t cnt[KWAY_MAX];
     uint64_t acc[KWAY_MAX + 1]; // exclusive prefix of cnt
     int nruns;
 };
 
 __global__ void k_merge_kway(kway_desc kd, uint64_t total,
-                             node_rec *__restrict__ out)
+                             node_rec *__restrict__ out,
+                             uint32_t *__restrict__ out_keys /* nullable */)
 {
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= total)
@@ -1038,10 +1050,12 @@ __global__ void k_merge_kway(kway_desc kd, uint64_t total,
         if (k >= kd.nruns || k == r)
             continue;
         const node_rec *B = kd.run[k];
+        const uint32_t *BK = kd.keys[k];
         uint64_t lo = 0, hi = kd.cnt[k];
         while (lo < hi) {
             uint64_t mid = (lo + hi) / 2;
-            if (B[mid].s < key)
+            uint32_t bk = BK ? BK[mid] : B[mid].s;
+            if (bk < key)
                 lo = mid + 1;
             else
                 hi = mid;
@@ -1049,11 +1063,14 @@ __global__ void k_merge_kway(kway_desc kd, uint64_t total,
         pos += lo;
     }
     copy_rec(&out[pos], me);
+    if (out_keys)
+        out_keys[pos] = key;
 }
 
 __global__ void k_merge_b(const node_rec *__restrict__ A, uint64_t nA,
                           const node_rec *__restrict__ B, uint64_t nB,
-                          node_rec *__restrict__ out)
+                          node_rec *__restrict__ out,
+                          const uint32_t *__restrict__ Akeys)
 {
     uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     if (j >= nB)
@@ -1062,7 +1079,8 @@ __global__ void k_merge_b(const node_rec *__restrict__ A, uint64_t nA,
     uint64_t lo = 0, hi = nA;
     while (lo < hi) {
         uint64_t mid = (lo + hi) / 2;
-        if (A[mid].s < key)
+        uint32_t ak = Akeys ? Akeys[mid] : A[mid].s;
+        if (ak < key)
             lo = mid + 1;
         else
             hi = mid;
@@ -3446,13 +3464,14 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
     // buffers stay alive (pool-backed) until the pass ends.
     struct run_ref {
         const node_rec *p;
+        const uint32_t *keys; // compact .s array (parallel to p)
         uint64_t n;
     };
     std::vector<run_ref> druns[64];
     uint64_t drun_total[64] = {0};
     std::vector<std::unique_ptr<DBuf>> live;
 
-    DBuf Lbuf(ctx), Cbuf(ctx), newn(ctx);
+    DBuf Lbuf(ctx), Cbuf(ctx), Ckeys(ctx), newn(ctx);
     DBuf flags(ctx), gidx(ctx), pend(ctx);
     DBuf gs(ctx), scratch(ctx), meta(ctx), urows(ctx), urow_cnt(ctx),
         urowidx(ctx);
@@ -3467,10 +3486,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 
     // 1. (once) stable depth-major bucket of the leaf records: doff[v] =
     // start of the depth-v slice; every level's fresh input is then a slice.
-    DBuf dsorted(ctx);
+    DBuf dsorted(ctx), dskeys(ctx);
     uint64_t doff[67];
     {
         HIP_CHECK(ctx, dsorted.alloc(n * sizeof(node_rec)));
+        HIP_CHECK(ctx, dskeys.alloc(n * 4));
         uint32_t nblk = (uint32_t)((n + BLOCK - 1) / BLOCK);
         DBuf dc(ctx), do_(ctx);
         HIP_CHECK(ctx, dc.alloc((uint64_t)66 * nblk * 4));
@@ -3484,7 +3504,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
             return -1;
         hipLaunchKernelGGL(k_depth_scatter66, dim3(nblk), dim3(BLOCK), 0,
                            ctx->stream, d_depths, d_recs, n, nblk,
-                           do_.as<uint32_t>(), dsorted.as<node_rec>());
+                           do_.as<uint32_t>(), dsorted.as<node_rec>(),
+                           dskeys.as<uint32_t>());
         HIP_CHECK(ctx, hipGetLastError());
         doff[0] = 0;
         for (int v = 0; v < 66; ++v)
@@ -3501,6 +3522,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
 
         // 1. this level's fresh leaves: the depth-(d+1) slice of dsorted
         node_rec *Lslice = dsorted.as<node_rec>() + doff[d + 1];
+        const uint32_t *Lkeys = dskeys.as<uint32_t>() + doff[d + 1];
         // 2. merge carried runs (lazily, ONCE, k-way — they are small
         // relative to the fresh slice), then 2-way with the fresh slice so
         // the bulk elements do exactly one positioning search
@@ -3515,6 +3537,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
             for (const run_ref &rr : druns[d]) {
                 if (nr < KWAY_MAX) {
                     kd.run[nr] = rr.p;
+                    kd.keys[nr] = rr.keys;
                     kd.cnt[nr] = rr.n;
                     nr++;
                 } else {
@@ -3537,33 +3560,39 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                     hipLaunchKernelGGL(k_merge_a, dim3(grid_for(kd.cnt[a])),
                                        dim3(BLOCK), 0, ctx->stream, kd.run[a],
                                        kd.cnt[a], kd.run[b], kd.cnt[b],
-                                       mb->as<node_rec>());
+                                       mb->as<node_rec>(), kd.keys[b]);
                     hipLaunchKernelGGL(k_merge_b, dim3(grid_for(kd.cnt[b])),
                                        dim3(BLOCK), 0, ctx->stream, kd.run[a],
                                        kd.cnt[a], kd.run[b], kd.cnt[b],
-                                       mb->as<node_rec>());
+                                       mb->as<node_rec>(), kd.keys[a]);
                     HIP_CHECK(ctx, hipGetLastError());
                     kd.run[a] = mb->as<node_rec>();
+                    kd.keys[a] = nullptr;
                     kd.cnt[a] = tot;
                     live.push_back(std::move(mb));
                     kd.run[b] = rr.p;
+                    kd.keys[b] = rr.keys;
                     kd.cnt[b] = rr.n;
                 }
             }
             const node_rec *carry;
+            const uint32_t *carry_keys;
             if (nr == 1) {
                 carry = kd.run[0];
+                carry_keys = kd.keys[0];
             } else {
                 kd.nruns = nr;
                 kd.acc[0] = 0;
                 for (int k = 0; k < nr; ++k)
                     kd.acc[k + 1] = kd.acc[k] + kd.cnt[k];
                 HIP_CHECK(ctx, Cbuf.alloc(nB * sizeof(node_rec)));
+                HIP_CHECK(ctx, Ckeys.alloc(nB * 4));
                 hipLaunchKernelGGL(k_merge_kway, dim3(grid_for(nB)),
                                    dim3(BLOCK), 0, ctx->stream, kd, nB,
-                                   Cbuf.as<node_rec>());
+                                   Cbuf.as<node_rec>(), Ckeys.as<uint32_t>());
                 HIP_CHECK(ctx, hipGetLastError());
                 carry = Cbuf.as<node_rec>();
+                carry_keys = Ckeys.as<uint32_t>();
             }
             if (nA == 0) {
                 L = (node_rec *)carry;
@@ -3571,10 +3600,10 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                 HIP_CHECK(ctx, Lbuf.alloc(n_level * sizeof(node_rec)));
                 hipLaunchKernelGGL(k_merge_a, dim3(grid_for(nA)), dim3(BLOCK),
                                    0, ctx->stream, Lslice, nA, carry, nB,
-                                   Lbuf.as<node_rec>());
+                                   Lbuf.as<node_rec>(), carry_keys);
                 hipLaunchKernelGGL(k_merge_b, dim3(grid_for(nB)), dim3(BLOCK),
                                    0, ctx->stream, Lslice, nA, carry, nB,
-                                   Lbuf.as<node_rec>());
+                                   Lbuf.as<node_rec>(), Lkeys);
                 HIP_CHECK(ctx, hipGetLastError());
                 L = Lbuf.as<node_rec>();
             }
@@ -3823,7 +3852,9 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         }
         if (any_fresh) {
             auto nb = std::make_unique<DBuf>(ctx);
+            auto nbk = std::make_unique<DBuf>(ctx);
             HIP_CHECK(ctx, nb->alloc((uint64_t)n_groups * sizeof(node_rec)));
+            HIP_CHECK(ctx, nbk->alloc((uint64_t)n_groups * 4));
             uint32_t nblk = (uint32_t)((n_groups + BLOCK - 1) / BLOCK);
             DBuf dc2(ctx), do2(ctx);
             HIP_CHECK(ctx, dc2.alloc((uint64_t)66 * nblk * 4));
@@ -3838,17 +3869,21 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                 return -1;
             hipLaunchKernelGGL(k_depth_scatter66_rec, dim3(nblk), dim3(BLOCK),
                                0, ctx->stream, newn.as<node_rec>(), n_groups,
-                               nblk, do2.as<uint32_t>(), nb->as<node_rec>());
+                               nblk, do2.as<uint32_t>(), nb->as<node_rec>(),
+                               nbk->as<uint32_t>());
             HIP_CHECK(ctx, hipGetLastError());
             // record the per-depth slices as carried runs, in place
             node_rec *basep = nb->as<node_rec>();
+            uint32_t *keyp = nbk->as<uint32_t>();
             for (int p = 0; p < d; ++p)
                 if (fresh_cnt[p]) {
                     druns[p].push_back({basep + slice_off[p + 1],
+                                        keyp + slice_off[p + 1],
                                         fresh_cnt[p]});
                     drun_total[p] += fresh_cnt[p];
                 }
             live.push_back(std::move(nb));
+            live.push_back(std::move(nbk));
         }
     }
     hipEventDestroy(ev0);
