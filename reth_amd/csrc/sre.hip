@@ -2397,9 +2397,14 @@ __global__ void k_sd_scatter(const sre_account_entry *__restrict__ base,
             uint64_t j0 = i0 - m_excl[p] + eff_excl[p];
             const uint64_t *s8 = (const uint64_t *)&base[i0];
             uint64_t *d8 = (uint64_t *)&out[j0];
-            uint32_t n8 = blockDim.x * 13;
-            for (uint32_t k = threadIdx.x; k < n8; k += blockDim.x)
-                d8[k] = s8[k];
+            // static trip count: issue all 13 loads before any store
+            uint64_t v[13];
+#pragma unroll
+            for (int w = 0; w < 13; ++w)
+                v[w] = s8[threadIdx.x + (uint32_t)w * BLOCK];
+#pragma unroll
+            for (int w = 0; w < 13; ++w)
+                d8[threadIdx.x + (uint32_t)w * BLOCK] = v[w];
             map[i] = (uint32_t)(i - m_excl[p] + eff_excl[p]);
             return;
         }
@@ -2625,9 +2630,13 @@ __global__ void k_sds_scatter(const sre_storage_entry *__restrict__ base,
             uint64_t j0 = i0 - m_excl[p] - wb0 + eff_excl[p];
             const uint64_t *s8 = (const uint64_t *)&base[i0];
             uint64_t *d8 = (uint64_t *)&out[j0];
-            uint32_t n8 = blockDim.x * 12;
-            for (uint32_t k = threadIdx.x; k < n8; k += blockDim.x)
-                d8[k] = s8[k];
+            uint64_t v[12];
+#pragma unroll
+            for (int w = 0; w < 12; ++w)
+                v[w] = s8[threadIdx.x + (uint32_t)w * BLOCK];
+#pragma unroll
+            for (int w = 0; w < 12; ++w)
+                d8[threadIdx.x + (uint32_t)w * BLOCK] = v[w];
             return;
         }
     }
