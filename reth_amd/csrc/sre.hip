@@ -915,14 +915,17 @@ __global__ void k_depth_scatter66(const uint8_t *__restrict__ depths,
     uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
     int key = (i < n) ? depths[i] : 255;
     uint32_t rank_in_wave = 0;
-    // 66 uniform ballots: each lane keeps the rank for its own key
-    for (int k = 0; k < 66; ++k) {
-        uint64_t m = __ballot(key == k);
-        if (key == k)
-            rank_in_wave = __popcll(m & ((1ull << lane) - 1));
-        if (lane == 0 && m)
-            wh[k][wid] = (uint32_t)__popcll(m);
-    }
+    // 66 uniform ballots: each lane keeps the rank for its own key.
+    // Fully-dead waves (all keys > 65 — the common case in incremental
+    // mode, where most positions are seeded or dead) skip the loop.
+    if (__ballot(key <= 65))
+        for (int k = 0; k < 66; ++k) {
+            uint64_t m = __ballot(key == k);
+            if (key == k)
+                rank_in_wave = __popcll(m & ((1ull << lane) - 1));
+            if (lane == 0 && m)
+                wh[k][wid] = (uint32_t)__popcll(m);
+        }
     __syncthreads();
     if (key > 65)
         return;
@@ -3733,6 +3736,14 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
             // split assemble (stream2) / hash (main), overlapped
             uint32_t c0 = use_fused ? c0s[chunk_i] : 0;
             uint32_t gsplit = gc - c0;
+            // split the fused class-0 work across both streams so it
+            // overlaps the split hash instead of serializing on main:
+            // main runs fused[0,c0m) then (after the assemble event) the
+            // split hash; stream2 runs assemble then fused[c0m,c0)
+            uint32_t c0m = (pipe2 && gsplit) ? (uint32_t)((uint64_t)c0 * 55 /
+                                                          100)
+                                             : c0;
+            uint32_t c0t = c0 - c0m;
             uint32_t *d_perm = use_cls ? perm.as<uint32_t>() + g0 + c0
                                        : nullptr;
             hipStream_t s_asm = pipe2 ? ctx->stream2 : ctx->stream;
@@ -3746,10 +3757,13 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                    d_lcp, d_keys, key_stride, d, scr,
                                    chunk, mt, d_perm, d_err);
             HIP_CHECK(ctx, hipGetLastError());
-            if (c0) {
-                hipLaunchKernelGGL(k_branch_fused1, dim3(grid_for(c0)),
+            if (pipe2) {
+                hipEventRecord(ev_asm[buf], s_asm);
+            }
+            if (c0m) {
+                hipLaunchKernelGGL(k_branch_fused1, dim3(grid_for(c0m)),
                                    dim3(BLOCK), 0, ctx->stream, L,
-                                   gs.as<uint32_t>() + g0, c0, d_lcp, d_keys,
+                                   gs.as<uint32_t>() + g0, c0m, d_lcp, d_keys,
                                    key_stride, d, subtree,
                                    newn.as<node_rec>() + g0,
                                    perm.as<uint32_t>() + g0, d_seg_roots,
@@ -3757,8 +3771,18 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                    pend.as<uint32_t>(), d_err);
                 HIP_CHECK(ctx, hipGetLastError());
             }
+            if (c0t) {
+                hipLaunchKernelGGL(k_branch_fused1, dim3(grid_for(c0t)),
+                                   dim3(BLOCK), 0, s_asm, L,
+                                   gs.as<uint32_t>() + g0, c0t, d_lcp, d_keys,
+                                   key_stride, d, subtree,
+                                   newn.as<node_rec>() + g0,
+                                   perm.as<uint32_t>() + g0 + c0m, d_seg_roots,
+                                   d_child_refs, d_child_lens,
+                                   pend.as<uint32_t>(), d_err);
+                HIP_CHECK(ctx, hipGetLastError());
+            }
             if (pipe2) {
-                hipEventRecord(ev_asm[buf], s_asm);
                 hipStreamWaitEvent(ctx->stream, ev_asm[buf], 0);
             }
             if (n_pt) { // multiproof: copy path-node RLPs out of scratch
@@ -3815,6 +3839,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                              hipMemcpyDeviceToHost));
                 }
             }
+        }
+        if (pipe2) { // trailing stream2 fused work must land before the
+                     // level's outputs are consumed
+            hipEventRecord(ev_asm[0], ctx->stream2);
+            hipStreamWaitEvent(ctx->stream, ev_asm[0], 0);
         }
         for (int b = 0; b < 2; ++b) {
             hipEventDestroy(ev_asm[b]);
