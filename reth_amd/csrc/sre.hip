@@ -3671,7 +3671,8 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
         // the main stream (classes 1-3 keep the split assemble/hash
         // pipeline); meta/scratch are produced inside the fused kernel, so
         // updates/proof modes (which read them across kernels) disable it
-        bool use_fused = use_cls && n_pt == 0 && updates_kind < 0;
+        bool use_fused = use_cls && n_pt == 0 && updates_kind < 0 &&
+                         getenv("SRE_NO_FUSED") == nullptr;
         std::vector<uint32_t> c0s;
         DBuf cinv(ctx);
         if (use_cls && n_pt)
