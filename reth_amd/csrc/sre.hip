@@ -1645,37 +1645,6 @@ __global__ void __launch_bounds__(BLOCK) k_branch_hash(
 }
 
 
-// Compact the class-0 groups' members into dense fixed-stride rows
-// (3 records per group) so the fused kernel STREAMS its members instead
-// of gathering scattered 48-B records (PMC measured the gathers at
-// ~3.5x algorithmic fetch). Reads L sequentially; writes follow the
-// stable class permutation, so they are near-sequential too.
-__global__ void k_compact0(const node_rec *__restrict__ L, uint64_t n,
-                           const uint32_t *__restrict__ flags,
-                           const uint32_t *__restrict__ gidx,
-                           const uint32_t *__restrict__ gs,
-                           const uint32_t *__restrict__ inv,
-                           uint64_t chunk,
-                           const uint32_t *__restrict__ base0,
-                           const uint32_t *__restrict__ c0s,
-                           node_rec *__restrict__ dense,
-                           uint8_t *__restrict__ dcnt)
-{
-    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (j >= n)
-        return;
-    uint32_t grp = gidx[j] + flags[j] - 1;
-    uint32_t ci = (uint32_t)(grp / chunk);
-    uint32_t slot = inv[grp]; // chunk-local class-partition slot
-    if (slot >= c0s[ci])
-        return; // not class 0
-    uint64_t di = (uint64_t)base0[ci] + slot;
-    uint32_t m = (uint32_t)(j - gs[grp]);
-    copy_rec(&dense[di * 3 + m], &L[j]);
-    if (m == 0)
-        dcnt[di] = (uint8_t)(gs[grp + 1] - gs[grp]);
-}
-
 // Fused assemble+hash for the 1-block branch class (nmem <= 3, the
 // majority of groups at uniform keys): the branch RLP is built in a
 // per-lane 136-B LDS slot and hashed in the same kernel — no global
@@ -1685,10 +1654,8 @@ __global__ void k_compact0(const node_rec *__restrict__ L, uint64_t n,
 // scratch across kernels).
 #define SLOT_F1 136
 __global__ void __launch_bounds__(BLOCK) k_branch_fused1(
-    const node_rec *__restrict__ dense /* this slice's 3-record member
-    rows (k_compact0) */,
-    const uint8_t *__restrict__ dcnt /* member counts, same slice */,
-    uint32_t n_groups /* class-0 groups of this slice */,
+    const node_rec *__restrict__ L, const uint32_t *__restrict__ gs,
+    uint32_t n_groups /* class-0 groups of this chunk */,
     const int8_t *__restrict__ lcp, const uint8_t *__restrict__ keys,
     uint64_t key_stride, int d, int subtree, node_rec *__restrict__ out,
     const uint32_t *__restrict__ perm /* class-0 slice */,
@@ -1704,16 +1671,19 @@ __global__ void __launch_bounds__(BLOCK) k_branch_fused1(
     uint32_t g = blockIdx.x * blockDim.x + threadIdx.x;
     if (g < n_groups) {
         uint32_t gg = perm[g];
-        const node_rec *mem = dense + (uint64_t)g * 3;
-        int nmem = dcnt[g];
+        uint64_t j = gs[gg], jend = gs[gg + 1];
+        int nmem = (int)(jend - j);
         bool sized_ok = nmem >= 2 && nmem <= 3;
         // validation words of all members issued up front (2 u32 each:
-        // ref_len word + pad word); the emit pass re-reads the rows L1-hot
+        // ref_len word + pad word). This both validates and WARMS the
+        // records' cache lines, so the emit pass re-reads them from L1 —
+        // caching the full 48-B records in registers measured as scratch
+        // spills instead.
         uint32_t w3[3], w11[3], w0[3], w1[3], w2[3];
 #pragma unroll
         for (int m = 0; m < 3; ++m)
             if (sized_ok && m < nmem) {
-                const uint32_t *r32 = (const uint32_t *)&mem[m];
+                const uint32_t *r32 = (const uint32_t *)&L[j + m];
                 w0[m] = r32[0];
                 w1[m] = r32[1];
                 w2[m] = r32[2];
@@ -1721,9 +1691,9 @@ __global__ void __launch_bounds__(BLOCK) k_branch_fused1(
                 w11[m] = r32[11];
             }
         br_meta mt;
-        mt.s = sized_ok ? w0[0] : mem[0].s;
-        mt.e = sized_ok ? w1[nmem > 0 ? nmem - 1 : 0] : mem[0].e;
-        mt.seg = sized_ok ? w2[0] : mem[0].seg;
+        mt.s = sized_ok ? w0[0] : L[j].s;
+        mt.e = sized_ok ? w1[nmem - 1] : L[jend - 1].e;
+        mt.seg = sized_ok ? w2[0] : L[j].seg;
         mt.d = (uint8_t)d;
         int8_t pl = lcp[mt.s], pr = lcp[mt.e];
         mt.P = pl > pr ? pl : pr;
@@ -1779,7 +1749,7 @@ __global__ void __launch_bounds__(BLOCK) k_branch_fused1(
                         prevnib = nbm;
                         int rl = (w3[m] >> 8) & 0xFF;
                         const uint32_t *rec32 =
-                            (const uint32_t *)&mem[m]; // L1-hot re-read
+                            (const uint32_t *)&L[j + m]; // L1-hot re-read
                         uint32_t v[9];
 #pragma unroll
                         for (int k = 0; k < 9; ++k)
@@ -3705,7 +3675,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                          getenv("SRE_NO_FUSED") == nullptr;
         std::vector<uint32_t> c0s;
         DBuf cinv(ctx);
-        if (use_cls && (n_pt || use_fused))
+        if (use_cls && n_pt)
             HIP_CHECK(ctx, cinv.alloc((uint64_t)n_groups * 4));
         if (use_cls) {
             HIP_CHECK(ctx, perm.alloc((uint64_t)n_groups * 4));
@@ -3728,9 +3698,7 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                    0, ctx->stream, gs.as<uint32_t>() + g0, gc,
                                    nblk, coff.as<uint32_t>(),
                                    perm.as<uint32_t>() + g0,
-                                   (n_pt || use_fused)
-                                       ? cinv.as<uint32_t>() + g0
-                                       : nullptr);
+                                   n_pt ? cinv.as<uint32_t>() + g0 : nullptr);
                 HIP_CHECK(ctx, hipGetLastError());
                 if (use_fused) { // class-0 count = start offset of class 1
                     uint32_t c0 = 0;
@@ -3738,39 +3706,6 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
                                              4, hipMemcpyDeviceToHost));
                     c0s.push_back(c0);
                 }
-            }
-        }
-        // dense class-0 member rows for the fused kernel (k_compact0):
-        // sequential L reads, near-sequential writes along the stable
-        // class permutation
-        DBuf dense(ctx), dcnt(ctx), dbase0(ctx), dc0s(ctx);
-        std::vector<uint32_t> base0(c0s.size() + 1, 0);
-        uint64_t total_c0 = 0;
-        if (use_fused) {
-            for (size_t q = 0; q < c0s.size(); ++q)
-                base0[q + 1] = base0[q] + c0s[q];
-            total_c0 = base0.back();
-            if (total_c0) {
-                HIP_CHECK(ctx, dense.alloc(total_c0 * 3 * sizeof(node_rec)));
-                HIP_CHECK(ctx, dcnt.alloc(total_c0));
-                HIP_CHECK(ctx, dbase0.alloc(4 * base0.size()));
-                HIP_CHECK(ctx, dc0s.alloc(4 * c0s.size()));
-                HIP_CHECK(ctx, hipMemcpyAsync(dbase0.p, base0.data(),
-                                              4 * base0.size(),
-                                              hipMemcpyHostToDevice,
-                                              ctx->stream));
-                HIP_CHECK(ctx, hipMemcpyAsync(dc0s.p, c0s.data(),
-                                              4 * c0s.size(),
-                                              hipMemcpyHostToDevice,
-                                              ctx->stream));
-                hipLaunchKernelGGL(k_compact0, dim3(grid_for(n_level)),
-                                   dim3(BLOCK), 0, ctx->stream, L, n_level,
-                                   flags.as<uint32_t>(), gidx.as<uint32_t>(),
-                                   gs.as<uint32_t>(), cinv.as<uint32_t>(),
-                                   chunk, dbase0.as<uint32_t>(),
-                                   dc0s.as<uint32_t>(),
-                                   dense.as<node_rec>(), dcnt.as<uint8_t>());
-                HIP_CHECK(ctx, hipGetLastError());
             }
         }
         bool pipe2 = n_groups > chunk; // >1 chunk: overlap pays for 2nd buf
@@ -3826,13 +3761,11 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
             if (pipe2) {
                 hipEventRecord(ev_asm[buf], s_asm);
             }
-            uint64_t db = use_fused ? base0[chunk_i] : 0;
             if (c0m) {
                 hipLaunchKernelGGL(k_branch_fused1, dim3(grid_for(c0m)),
-                                   dim3(BLOCK), 0, ctx->stream,
-                                   dense.as<node_rec>() + db * 3,
-                                   dcnt.as<uint8_t>() + db, c0m, d_lcp,
-                                   d_keys, key_stride, d, subtree,
+                                   dim3(BLOCK), 0, ctx->stream, L,
+                                   gs.as<uint32_t>() + g0, c0m, d_lcp, d_keys,
+                                   key_stride, d, subtree,
                                    newn.as<node_rec>() + g0,
                                    perm.as<uint32_t>() + g0, d_seg_roots,
                                    d_child_refs, d_child_lens,
@@ -3841,10 +3774,9 @@ static int run_levels(sre_ctx *ctx, uint64_t n, node_rec *d_recs, uint8_t *d_dep
             }
             if (c0t) {
                 hipLaunchKernelGGL(k_branch_fused1, dim3(grid_for(c0t)),
-                                   dim3(BLOCK), 0, s_asm,
-                                   dense.as<node_rec>() + (db + c0m) * 3,
-                                   dcnt.as<uint8_t>() + db + c0m, c0t, d_lcp,
-                                   d_keys, key_stride, d, subtree,
+                                   dim3(BLOCK), 0, s_asm, L,
+                                   gs.as<uint32_t>() + g0, c0t, d_lcp, d_keys,
+                                   key_stride, d, subtree,
                                    newn.as<node_rec>() + g0,
                                    perm.as<uint32_t>() + g0 + c0m, d_seg_roots,
                                    d_child_refs, d_child_lens,
