@@ -997,17 +997,14 @@ __global__ void k_depth_scatter66_rec(const node_rec *__restrict__ recs,
 // merge two record arrays sorted by .s (distinct keys)
 
 // Bkeys: compact 4-B .s array of B (12x denser probes than searching the
-// 48-B-stride records). Null -> probe the records directly.
-__global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
-                          const node_rec *__restrict__ B, uint64_t nB,
-                          node_rec *__restrict__ out,
-                          const uint32_t *__restrict__ Bkeys)
+// 48-B-stride records). Null -> probe the records directly. The block
+// computes a shared probe window from its first/last key (consecutive
+// elements' ranks are near-monotone), so per-element searches touch a
+// handful of shared cache lines instead of log2(nB) scattered ones.
+__device__ __forceinline__ uint64_t lb_s(const node_rec *B,
+                                         const uint32_t *Bkeys, uint64_t lo,
+                                         uint64_t hi, uint32_t key)
 {
-    uint64_t i = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= nA)
-        return;
-    uint32_t key = A[i].s;
-    uint64_t lo = 0, hi = nB;
     while (lo < hi) {
         uint64_t mid = (lo + hi) / 2;
         uint32_t bk = Bkeys ? Bkeys[mid] : B[mid].s;
@@ -1016,6 +1013,29 @@ __global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
         else
             hi = mid;
     }
+    return lo;
+}
+
+__global__ void k_merge_a(const node_rec *__restrict__ A, uint64_t nA,
+                          const node_rec *__restrict__ B, uint64_t nB,
+                          node_rec *__restrict__ out,
+                          const uint32_t *__restrict__ Bkeys)
+{
+    __shared__ uint64_t wlo_s, whi_s;
+    uint64_t i0 = (uint64_t)blockIdx.x * blockDim.x;
+    if (threadIdx.x == 0) {
+        uint64_t a = i0 < nA ? i0 : (nA ? nA - 1 : 0);
+        uint64_t b = i0 + blockDim.x - 1;
+        if (b >= nA)
+            b = nA ? nA - 1 : 0;
+        wlo_s = nA ? lb_s(B, Bkeys, 0, nB, A[a].s) : 0;
+        whi_s = nA ? lb_s(B, Bkeys, wlo_s, nB, A[b].s + 1) : 0;
+    }
+    __syncthreads();
+    uint64_t i = i0 + threadIdx.x;
+    if (i >= nA)
+        return;
+    uint64_t lo = lb_s(B, Bkeys, wlo_s, whi_s, A[i].s);
     copy_rec(&out[i + lo], &A[i]);
 }
 
@@ -1075,19 +1095,21 @@ __global__ void k_merge_b(const node_rec *__restrict__ A, uint64_t nA,
                           node_rec *__restrict__ out,
                           const uint32_t *__restrict__ Akeys)
 {
-    uint64_t j = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    __shared__ uint64_t wlo_s, whi_s;
+    uint64_t j0 = (uint64_t)blockIdx.x * blockDim.x;
+    if (threadIdx.x == 0) {
+        uint64_t a = j0 < nB ? j0 : (nB ? nB - 1 : 0);
+        uint64_t b = j0 + blockDim.x - 1;
+        if (b >= nB)
+            b = nB ? nB - 1 : 0;
+        wlo_s = nB ? lb_s(A, Akeys, 0, nA, B[a].s) : 0;
+        whi_s = nB ? lb_s(A, Akeys, wlo_s, nA, B[b].s + 1) : 0;
+    }
+    __syncthreads();
+    uint64_t j = j0 + threadIdx.x;
     if (j >= nB)
         return;
-    uint32_t key = B[j].s;
-    uint64_t lo = 0, hi = nA;
-    while (lo < hi) {
-        uint64_t mid = (lo + hi) / 2;
-        uint32_t ak = Akeys ? Akeys[mid] : A[mid].s;
-        if (ak < key)
-            lo = mid + 1;
-        else
-            hi = mid;
-    }
+    uint64_t lo = lb_s(A, Akeys, wlo_s, whi_s, B[j].s);
     copy_rec(&out[j + lo], &B[j]);
 }
 
