@@ -5519,6 +5519,13 @@ static int apply_delta_small(sre_ctx *ctx, const sre_account_delta *acct_delta,
                              uint64_t nd, DBuf *map_out)
 {
     uint64_t nb = ctx->na;
+    // the general path's k_ovl_delta_acct_flags validates this on device;
+    // closed-form ranks would silently mis-merge on unsorted input
+    for (uint64_t k = 1; k < nd; ++k)
+        if (memcmp(acct_delta[k - 1].key, acct_delta[k].key, 32) >= 0) {
+            set_err(ctx, "apply_delta: account delta unsorted or duplicate");
+            return -1;
+        }
     DBuf dl(ctx), bpos(ctx), match(ctx), mex(ctx), eex(ctx), npos(ctx),
         err(ctx);
     HIP_CHECK(ctx, dl.alloc(nd * sizeof(sre_account_delta)));
@@ -5735,7 +5742,8 @@ static int incremental_root_impl(sre_ctx *ctx,
     // fix for the configs[4] shape. Everything else keeps the general
     // overlay merge.
     bool small = no_storage && n_acct > 0 && ctx->lcp_valid &&
-                 n_acct <= (ctx->na >> 6);
+                 (n_acct <= (ctx->na >> 6) ||
+                  (getenv("SRE_SD_FORCE") && n_acct <= ctx->na));
     if (small) {
         if (apply_delta_small(ctx, acct_delta, n_acct, &map))
             return -1;

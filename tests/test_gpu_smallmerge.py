@@ -119,6 +119,115 @@ def test_smallmerge_dirty_path_with_storage(eng):
             bind.state_root(*_arrays_of(accounts)), f"step {step}"
 
 
+@pytest.fixture()
+def force_small():
+    # SRE_SD_FORCE lowers the account small-path gate (nd <= na instead
+    # of nd <= na/64) so test-sized deltas take apply_delta_small
+    os.environ["SRE_SD_FORCE"] = "1"
+    yield
+    os.environ.pop("SRE_SD_FORCE", None)
+
+
+def test_smallmerge_forced_account_path_chained(eng, force_small):
+    rng = np.random.default_rng(777)
+    acct, _ = gen.gen_state_numpy(1500, 0, bind.keccak256_batch)
+    accounts = {bytes(a["key"]): [int(a["nonce"]),
+                int.from_bytes(bytes(a["balance"]), "big"),
+                bytes(a["code_hash"]), {}] for a in acct}
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    assert eng.root_retaining() == bind.state_root(*_arrays_of(accounts))
+    ke = bind.keccak256(b"")
+    for step in range(4):
+        keys = sorted(accounts)
+        rows = []
+        for i in rng.choice(len(keys), 60, replace=False):
+            k = keys[int(i)]
+            v = accounts[k]
+            rows.append((k, v[0] + 1, v[1] + 3, v[2], 0))
+        for i in rng.choice(len(keys), 20, replace=False):
+            k = keys[int(i)]
+            if any(r[0] == k for r in rows):
+                continue
+            rows.append((k, 0, 0, ke, 1))
+        for i in range(25):
+            rows.append((bind.keccak256(b"fs" + bytes([step, i])),
+                         1, 100 + i, ke, 0))
+        rows = sorted(set(rows))
+        _apply_dict(accounts, rows)
+        d, _ = _mk_delta(rows, [])
+        assert eng.incremental_root(d) == \
+            bind.state_root(*_arrays_of(accounts)), f"step {step}"
+
+
+def test_smallmerge_forced_with_updates(eng, force_small):
+    # the small merge + incremental TrieUpdates diff combination
+    from tests.test_gpu_incremental_updates import (_rowmap, _apply_diff)
+    rng = np.random.default_rng(31415)
+    acct, _ = gen.gen_state_numpy(2000, 0, bind.keccak256_batch)
+    accounts = {bytes(a["key"]): [int(a["nonce"]),
+                int.from_bytes(bytes(a["balance"]), "big"),
+                bytes(a["code_hash"]), {}] for a in acct}
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    root0, rows0 = eng.root_retaining_with_updates()
+    cur = _rowmap(rows0)
+    ke = bind.keccak256(b"")
+    for step in range(3):
+        keys = sorted(accounts)
+        rows = []
+        for i in rng.choice(len(keys), 50, replace=False):
+            k = keys[int(i)]
+            v = accounts[k]
+            rows.append((k, v[0] + 2, v[1] + 9, v[2], 0))
+        for i in rng.choice(len(keys), 15, replace=False):
+            k = keys[int(i)]
+            if any(r[0] == k for r in rows):
+                continue
+            rows.append((k, 0, 0, ke, 1))
+        rows.append((bind.keccak256(b"fw" + bytes([step])), 1, 7, ke, 0))
+        rows = sorted(set(rows))
+        _apply_dict(accounts, rows)
+        d, _ = _mk_delta(rows, [])
+        root, diff = eng.incremental_root_with_updates(d)
+        oroot, orows = bind.state_root_with_updates(*_arrays_of(accounts))
+        assert root == oroot, f"step {step}"
+        cur = _apply_diff(cur, diff)
+        assert cur == _rowmap(orows), f"step {step}"
+
+
+def test_smallmerge_forced_delete_all(eng, force_small):
+    ke = bind.keccak256(b"")
+    acct, _ = gen.gen_state_numpy(300, 0, bind.keccak256_batch)
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    eng.root_retaining()
+    # one small delta first (arms the lcp repair chain), then delete all
+    keys = sorted(bytes(a["key"]) for a in acct)
+    d1, _ = _mk_delta([(keys[0], 9, 9, ke, 0)], [])
+    eng.incremental_root(d1)
+    rows = [(k, 0, 0, ke, 1) for k in keys]
+    d, _ = _mk_delta(sorted(rows), [])
+    empty_root = bytes.fromhex(
+        "56e81f171bcc55a6ff8345e692c0f86e5b48e01b996cadc001622fb5e363b421")
+    assert eng.incremental_root(d) == empty_root
+
+
+def test_smallmerge_rejects_unsorted_account_delta(eng, force_small):
+    acct, _ = gen.gen_state_numpy(200, 0, bind.keccak256_batch)
+    eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
+    eng.root_retaining()
+    ke = bind.keccak256(b"")
+    # warm delta arms the retained lcp so the NEXT call takes the small path
+    warm, _ = _mk_delta([(bytes(acct[0]["key"]), 5, 5, ke, 0)], [])
+    eng.incremental_root(warm)
+    k1, k2 = sorted([bind.keccak256(b"u1"), bind.keccak256(b"u2")])
+    d = np.zeros(2, DELTA_DTYPE)
+    for i, k in enumerate([k2, k1]):  # wrong order
+        d[i]["key"] = np.frombuffer(k, np.uint8)
+        d[i]["nonce"] = 1
+        d[i]["code_hash"] = np.frombuffer(ke, np.uint8)
+    with pytest.raises(RuntimeError):
+        eng.incremental_root(d)
+
+
 def test_smallmerge_rejects_bad_delta(eng):
     acct, st = gen.gen_state_numpy(200, 2, bind.keccak256_batch)
     eng.upload(acct, st)
