@@ -210,6 +210,50 @@ def test_smallmerge_forced_delete_all(eng, force_small):
     assert eng.incremental_root(d) == empty_root
 
 
+def test_smallmerge_storage_with_updates_diff(eng):
+    # closed-form STORAGE merge (gate lowered module-wide) under the
+    # incremental TrieUpdates diff: wipe ranges + slot churn must yield a
+    # net diff that reproduces the oracle's full-rebuild row set
+    from tests.test_gpu_incremental_updates import (_rowmap, _apply_diff)
+    rng = np.random.default_rng(60221023)
+    acct, st = gen.gen_state_numpy(1200, 5, bind.keccak256_batch)
+    accounts = _dict_of(acct, st)
+    eng.upload(acct, st)
+    root0, rows0 = eng.root_retaining_with_updates()
+    oroot, orows = bind.state_root_with_updates(*_arrays_of(accounts))
+    assert root0 == oroot
+    cur = _rowmap(rows0)
+    assert cur == _rowmap(orows)
+    ke = bind.keccak256(b"")
+    for step in range(3):
+        keys = sorted(accounts)
+        rows, strows = [], []
+        for i in rng.choice(len(keys), 10, replace=False):
+            k = keys[int(i)]
+            slots = accounts[k][3]
+            if slots and rng.random() < 0.5:
+                dead = sorted(slots)[0]
+                strows.append((k, dead, 0))
+                del slots[dead]
+            nk = bind.keccak256(b"su" + bytes([step]) + k[:4])
+            strows.append((k, nk, 777 + step))
+            slots[nk] = 777 + step
+        victim = next(k for k in keys if accounts[k][3]
+                      and all(r[0] != k for r in strows))
+        rows.append((victim, 0, 0, ke, 1))
+        del accounts[victim]
+        rows = sorted(set(rows))
+        strows = sorted(set(strows))
+        d, s = _mk_delta(rows, strows)
+        root, diff = eng.incremental_root_with_updates(d, s)
+        oroot, orows = bind.state_root_with_updates(*_arrays_of(accounts))
+        assert root == oroot, f"step {step}"
+        assert bytes(victim) in {bytes(r["acct_key"])
+                                 for r in diff[diff["removed"] == 2]}
+        cur = _apply_diff(cur, diff)
+        assert cur == _rowmap(orows), f"step {step}"
+
+
 def test_smallmerge_rejects_unsorted_account_delta(eng, force_small):
     acct, _ = gen.gen_state_numpy(200, 0, bind.keccak256_batch)
     eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
