@@ -6102,6 +6102,10 @@ extern "C" int sre_incremental_root_with_updates(
         if (s.kind == 0) {
             if (s.path_len <= 4)
                 return true;
+            // empty bitmap: the delta emptied the state (the impl returns
+            // before the account pass) — everything is rebuilt region
+            if (bitmap.empty())
+                return true;
             uint32_t cell = ((uint32_t)s.path[0] << 12) |
                             ((uint32_t)s.path[1] << 4) |
                             ((uint32_t)s.path[2] >> 4);

@@ -175,6 +175,26 @@ def test_incremental_updates_suppresses_unchanged(eng):
     assert root == root0 and len(diff) == 0
 
 
+def test_incremental_updates_delete_everything(eng):
+    # a delta that empties the state: every stored row becomes a removal
+    # (plus whole-trie markers for storage-bearing destroyed accounts)
+    acct, st = gen.gen_state_numpy(400, 3, bind.keccak256_batch)
+    accounts = _dict_of(acct, st)
+    eng.upload(acct, st)
+    root0, rows0 = eng.root_retaining_with_updates()
+    cur = _rowmap(rows0)
+    ke = bind.keccak256(b"")
+    rows = [(k, 0, 0, ke, 1) for k in sorted(accounts)]
+    d, _ = _mk_delta(rows, [])
+    root, diff = eng.incremental_root_with_updates(d)
+    empty_root = bytes.fromhex(
+        "56e81f171bcc55a6ff8345e692c0f86e5b48e01b996cadc001622fb5e363b421")
+    assert root == empty_root
+    cur = _apply_diff(cur, diff)
+    assert cur == {}, "all rows must be removed"
+    assert not any(int(r["removed"]) == 0 for r in diff), "no upserts"
+
+
 def test_incremental_updates_requires_arming(eng):
     acct, _ = gen.gen_state_numpy(100, 0, bind.keccak256_batch)
     eng.upload(acct, np.zeros(0, bind.STORAGE_DTYPE))
